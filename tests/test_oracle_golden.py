@@ -132,7 +132,8 @@ def test_timestamps_with_versions():
     assert t2 == ts
     assert v2 == vers
     # irregular timestamps -> DeltaOfDelta family
-    ts = sorted(1_700_000_000_000_000_000 + random.Random(3).randint(0, 10 ** 10) for _ in range(64))
+    rng = random.Random(3)
+    ts = sorted(1_700_000_000_000_000_000 + rng.randint(0, 10 ** 10) for _ in range(64))
     m = o.timestamps_encode(ts, [1] * 64)
     assert m["enc"] == 8  # DeltaOfDeltaWithVersion
     t2, _ = o.timestamps_decode(m, 64)
