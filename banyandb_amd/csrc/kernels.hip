@@ -1,0 +1,867 @@
+// banyandb_amd/csrc/kernels.hip — MI355X (gfx950) decode+fold kernels and
+// the C-ABI session around them.
+//
+// Hot path replaced (apache/skywalking-banyandb):
+//   blockCursor.loadData -> block.mustReadFrom (banyand/measure/block.go:818,
+//   324) -> BytesToInt64List (pkg/encoding/int_list.go:57) ->
+//   BatchAggregation.Consume fold (pkg/query/vectorized/measure/
+//   aggregation.go:310-351,464-486; pkg/query/aggregation/function.go).
+//
+// Design (CDNA4-first, not a translation):
+//  * one 64-lane WAVE per (series,block) — 1.2M blocks at the benchmark
+//    configs is ~150 blocks per wave slot at full occupancy; no intra-block
+//    sync, no LDS;
+//  * the varint stream (zigzag, 7-bit groups LSB-first, terminator = high
+//    bit clear — pkg/encoding/int.go:81-148) is decoded 64 bytes per step:
+//    each lane loads one byte, a 64-bit __ballot of "high bit clear" marks
+//    value ends, each terminator lane reconstructs its value (1-byte fast
+//    path covers the dominant case);
+//  * SUM/COUNT need no prefix reconstruction at all: for delta streams
+//    sum(v_i, i=r0..r1) = nsel*first + sum_j w_j * d_j with closed-form
+//    weights, so the kernel does a weighted fold of raw deltas (exact mod
+//    2^64 — Go int64 addition wraps identically);
+//  * MIN/MAX (and float64 restore) reconstruct values with a wave-wide
+//    inclusive scan (__shfl_up, 6 steps) — the reference's serial carry
+//    chain (delta.go:45-117) becomes a parallel prefix sum;
+//  * float64 columns stay in the decimal-int domain in-kernel (monotone
+//    restore, float.go:69-102 applied at finalize); the mantissa sum is
+//    carried per block into a double;
+//  * per-block partials land in dense per-group accumulators with one
+//    device-scope atomic set per wave per block (Guideline 12).
+#include "../../include/bydb_gpu.h"
+
+#include <hip/hip_runtime.h>
+
+#include <cstdio>
+#include <cstdlib>
+#include <cstring>
+#include <string>
+#include <vector>
+
+#define WAVE 64
+
+// ---------------- device helpers ----------------
+
+__device__ __forceinline__ int64_t zz_dec(uint64_t u) {
+    return (int64_t)(u >> 1) ^ -(int64_t)(u & 1);
+}
+
+__device__ __forceinline__ uint64_t lanemask_lt(int lane) {
+    return (lane == 0) ? 0ull : (~0ull >> (64 - lane));
+}
+
+// wave-wide inclusive scan of a u64 (wrapping adds)
+__device__ __forceinline__ uint64_t wave_incl_scan(uint64_t v, int lane) {
+#pragma unroll
+    for (int off = 1; off < 64; off <<= 1) {
+        uint64_t t = (uint64_t)__shfl_up((long long)v, off);
+        if (lane >= off) v += t;
+    }
+    return v;
+}
+
+__device__ __forceinline__ uint64_t wave_reduce_add(uint64_t v) {
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) v += (uint64_t)__shfl_xor((long long)v, off);
+    return v;
+}
+
+__device__ __forceinline__ int64_t wave_reduce_min(int64_t v) {
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) {
+        int64_t t = (int64_t)__shfl_xor((long long)v, off);
+        v = t < v ? t : v;
+    }
+    return v;
+}
+
+__device__ __forceinline__ int64_t wave_reduce_max(int64_t v) {
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) {
+        int64_t t = (int64_t)__shfl_xor((long long)v, off);
+        v = t > v ? t : v;
+    }
+    return v;
+}
+
+__device__ __forceinline__ double wave_reduce_addf(double v) {
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) v += __shfl_xor(v, off);
+    return v;
+}
+
+// Decode one varint serially starting at p (used for wave-uniform headers:
+// the DeltaConst stride and the delta-of-delta d1 — every lane runs the
+// same loop on the same bytes, so the result is wave-uniform).
+__device__ __forceinline__ int64_t decode_one_varint(const uint8_t *p, int *len) {
+    uint8_t c = p[0];
+    if (c < 0x80) { *len = 1; return zz_dec(c); }
+    uint64_t u = (uint64_t)(c & 0x7f);
+    int i = 1;
+    unsigned shift = 0;
+    while (c >= 0x80 && i <= 10) {
+        c = p[i++];
+        shift += 7;
+        u |= (uint64_t)(c & 0x7f) << shift;
+    }
+    *len = i;
+    return zz_dec(u);
+}
+
+// Triangular number T(m) = m(m+1)/2 for m <= 8192 (fits u32 domain widely).
+__device__ __forceinline__ uint64_t tri(uint64_t m) { return m * (m + 1) / 2; }
+
+struct DevErr {
+    unsigned code;      // first error code seen (atomicCAS)
+    unsigned block_lo;  // low 32 bits of the offending block index
+};
+
+enum {
+    DERR_NONE = 0,
+    DERR_BAD_STREAM = 1,    // no terminator in a 64-byte window
+    DERR_BAD_ENC = 2,       // unsupported encode type in kernel
+    DERR_DESC_TS = 3,       // descending timestamps (reference never writes them)
+};
+
+__device__ __forceinline__ void dev_set_err(DevErr *e, unsigned code, uint64_t bi) {
+    if (atomicCAS(&e->code, DERR_NONE, code) == DERR_NONE)
+        e->block_lo = (unsigned)bi;
+}
+
+// ---------------- per-block decode primitives ----------------
+
+// Weighted delta fold (fast path, no value reconstruction):
+//   returns sum_{j=jlo..jhi} d_j * w(j) over the varint stream,
+//   w(j) = nsel for j <= r0, else (r1 - j + 1); deltas are 1-indexed.
+// Used for EncodeTypeDelta sum. jend = min(n-1, r1).
+__device__ uint64_t fold_delta_weighted(const uint8_t *stream, int64_t n_deltas,
+                                        int64_t r0, int64_t r1, int lane,
+                                        DevErr *derr, uint64_t bi) {
+    uint64_t acc = 0;
+    uint64_t pos = 0;
+    int64_t j = 1;
+    int64_t jend = r1 < n_deltas ? r1 : n_deltas;
+    uint64_t nsel = (uint64_t)(r1 - r0 + 1);
+    while (j <= jend) {
+        uint8_t b = stream[pos + (uint64_t)lane];
+        uint64_t emask = __ballot(b < 0x80);
+        if (emask == 0) { dev_set_err(derr, DERR_BAD_STREAM, bi); return acc; }
+        int rank = __popcll(emask & lanemask_lt(lane));
+        int64_t myj = j + rank;
+        bool is_term = (b < 0x80) && (myj <= jend);
+        int64_t d;
+        if (emask == ~0ull) {
+            d = zz_dec(b);
+        } else if (is_term) {
+            uint64_t below = emask & lanemask_lt(lane);
+            int start = below ? (64 - __clzll(below)) : 0;
+            uint64_t u = 0;
+            unsigned sh = 0;
+            for (int i = start; i < lane; ++i) {
+                u |= (uint64_t)(stream[pos + (uint64_t)i] & 0x7f) << sh;
+                sh += 7;
+            }
+            u |= (uint64_t)b << sh;
+            d = zz_dec(u);
+        } else {
+            d = 0;
+        }
+        if (is_term) {
+            uint64_t w = myj <= r0 ? nsel : (uint64_t)(r1 - myj + 1);
+            acc += (uint64_t)d * w;
+        }
+        int nterm = __popcll(emask);
+        if (j + nterm > jend) break;
+        j += nterm;
+        // advance past the last terminator in this window
+        pos += (uint64_t)(64 - __clzll(emask));
+    }
+    return acc;
+}
+
+// Weighted delta-of-delta fold: sum_{i=r0..r1} v_i with
+//   v_i = first + i*d1 + sum_{k=2..i} d2_k * (i-k+1)
+// => contribution of d2_k (k in [2, r1]):  W_k = T(r1-k+1) - T(max(k,r0)-k).
+__device__ uint64_t fold_dod_weighted(const uint8_t *stream, int64_t n_deltas,
+                                      int64_t r0, int64_t r1, int lane,
+                                      DevErr *derr, uint64_t bi) {
+    uint64_t acc = 0;
+    uint64_t pos = 0;
+    int64_t k = 2;                       // d2 indices run 2..n-1
+    int64_t kend = r1 < n_deltas ? r1 : n_deltas;
+    while (k <= kend) {
+        uint8_t b = stream[pos + (uint64_t)lane];
+        uint64_t emask = __ballot(b < 0x80);
+        if (emask == 0) { dev_set_err(derr, DERR_BAD_STREAM, bi); return acc; }
+        int rank = __popcll(emask & lanemask_lt(lane));
+        int64_t myk = k + rank;
+        bool is_term = (b < 0x80) && (myk <= kend);
+        int64_t d;
+        if (emask == ~0ull) {
+            d = zz_dec(b);
+        } else if (is_term) {
+            uint64_t below = emask & lanemask_lt(lane);
+            int start = below ? (64 - __clzll(below)) : 0;
+            uint64_t u = 0;
+            unsigned sh = 0;
+            for (int i = start; i < lane; ++i) {
+                u |= (uint64_t)(stream[pos + (uint64_t)i] & 0x7f) << sh;
+                sh += 7;
+            }
+            u |= (uint64_t)b << sh;
+            d = zz_dec(u);
+        } else {
+            d = 0;
+        }
+        if (is_term) {
+            int64_t a = myk > r0 ? myk : r0;
+            uint64_t w = tri((uint64_t)(r1 - myk + 1)) - tri((uint64_t)(a - myk));
+            acc += (uint64_t)d * w;
+        }
+        int nterm = __popcll(emask);
+        if (k + nterm > kend) break;
+        k += nterm;
+        pos += (uint64_t)(64 - __clzll(emask));
+    }
+    return acc;
+}
+
+// Full value reconstruction over a delta or delta-of-delta stream with a
+// per-value callback encoded as flags (fold min/max/sum/sumf, or count
+// ts-range bounds).  dod=false: values are rows j=1..n-1 (row 0 = first,
+// handled by the caller).  dod=true: stream holds d2 for rows 2..n-1 and
+// d1_init has been parsed; rows 0,1 handled by the caller.
+struct ScanFold {
+    uint64_t sum;       // wrapping sum of selected values
+    int64_t mn, mx;     // min/max of selected values
+    uint64_t nsel;      // selected row count
+    uint64_t n_lo;      // rows with v < lo_bound (ts clamp use)
+    uint64_t n_hi;      // rows with v > hi_bound
+};
+
+__device__ void scan_stream(const uint8_t *stream, int64_t n_deltas, bool dod,
+                            int64_t first_plus /* v at row (dod?1:0) */,
+                            int64_t d1_init, int64_t r0, int64_t r1,
+                            int64_t lo_bound, int64_t hi_bound, int lane,
+                            ScanFold *f, DevErr *derr, uint64_t bi) {
+    uint64_t pos = 0;
+    int64_t j = dod ? 2 : 1;
+    int64_t jmax = n_deltas;            // always scan the whole stream
+    uint64_t v_carry = (uint64_t)first_plus;
+    uint64_t d1_carry = (uint64_t)d1_init;
+    uint64_t l_sum = 0, l_nsel = 0, l_nlo = 0, l_nhi = 0;
+    int64_t l_mn = INT64_MAX, l_mx = INT64_MIN;
+    while (j <= jmax) {
+        uint8_t b = stream[pos + (uint64_t)lane];
+        uint64_t emask = __ballot(b < 0x80);
+        if (emask == 0) { dev_set_err(derr, DERR_BAD_STREAM, bi); break; }
+        int rank = __popcll(emask & lanemask_lt(lane));
+        int64_t myj = j + rank;
+        bool is_term = (b < 0x80) && (myj <= jmax);
+        uint64_t d = 0;
+        if (emask == ~0ull) {
+            d = (uint64_t)zz_dec(b);
+        } else if (is_term) {
+            uint64_t below = emask & lanemask_lt(lane);
+            int start = below ? (64 - __clzll(below)) : 0;
+            uint64_t u = 0;
+            unsigned sh = 0;
+            for (int i = start; i < lane; ++i) {
+                u |= (uint64_t)(stream[pos + (uint64_t)i] & 0x7f) << sh;
+                sh += 7;
+            }
+            u |= (uint64_t)b << sh;
+            d = (uint64_t)zz_dec(u);
+        }
+        if (!is_term) d = 0;
+        uint64_t val;
+        uint64_t tot;
+        if (dod) {
+            // first scan: d1_j = d1_carry + incl_scan(d2)
+            uint64_t s1 = wave_incl_scan(d, lane);
+            uint64_t d1j = is_term ? (d1_carry + s1) : 0;
+            uint64_t s1_tot = (uint64_t)__shfl((long long)s1, 63);
+            // second scan: v_j = v_carry + incl_scan(d1_j over values)
+            uint64_t s2 = wave_incl_scan(d1j, lane);
+            val = v_carry + s2;
+            // totals: number of values in window and their d1 sum
+            int nterm_all = __popcll(emask);
+            int64_t nvals = (jmax - j + 1) < (int64_t)nterm_all
+                                ? (jmax - j + 1) : (int64_t)nterm_all;
+            // d1 total over consumed values = s2 at last consumed terminator;
+            // easier: v_carry advances by s2 at the last consumed value lane.
+            // find lane of the nvals-th terminator:
+            uint64_t mm = emask;
+            int last_lane = 0;
+            for (int t = 0; t < nvals; t++) {   // nvals <= 64; wave-uniform loop
+                last_lane = __ffsll((unsigned long long)mm) - 1;
+                mm &= mm - 1;
+            }
+            tot = (uint64_t)__shfl((long long)s2, last_lane);
+            v_carry += tot;
+            // d1_carry advances by scan1 at last consumed value lane
+            uint64_t d1tot = (uint64_t)__shfl((long long)s1, last_lane);
+            d1_carry += d1tot;
+            (void)s1_tot;
+        } else {
+            uint64_t s = wave_incl_scan(d, lane);
+            val = v_carry + s;
+            int nterm_all = __popcll(emask);
+            int64_t nvals = (jmax - j + 1) < (int64_t)nterm_all
+                                ? (jmax - j + 1) : (int64_t)nterm_all;
+            uint64_t mm = emask;
+            int last_lane = 0;
+            for (int t = 0; t < nvals; t++) {
+                last_lane = __ffsll((unsigned long long)mm) - 1;
+                mm &= mm - 1;
+            }
+            tot = (uint64_t)__shfl((long long)s, last_lane);
+            v_carry += tot;
+        }
+        if (is_term) {
+            int64_t sv = (int64_t)val;
+            if (myj >= r0 && myj <= r1) {
+                l_sum += val;
+                l_nsel++;
+                l_mn = sv < l_mn ? sv : l_mn;
+                l_mx = sv > l_mx ? sv : l_mx;
+            }
+            if (sv < lo_bound) l_nlo++;
+            if (sv > hi_bound) l_nhi++;
+        }
+        int nterm = __popcll(emask);
+        if (j + nterm > jmax) break;
+        j += nterm;
+        pos += (uint64_t)(64 - __clzll(emask));
+    }
+    f->sum = l_sum;
+    f->mn = l_mn;
+    f->mx = l_mx;
+    f->nsel = l_nsel;
+    f->n_lo = l_nlo;
+    f->n_hi = l_nhi;
+}
+
+// ---------------- the scan+aggregate kernel ----------------
+
+enum {
+    KF_NEED_VALUES = 1,  // min/max requested -> reconstruct values
+    KF_FLOAT = 2,        // float64 field: also accumulate mantissa sum as double
+};
+
+__global__ __launch_bounds__(256) void k_scan_agg(
+    const uint8_t *__restrict__ payload, const bydb_block_desc *__restrict__ blocks,
+    int64_t n_blocks, int64_t min_ts, int64_t max_ts, int flags,
+    bydb_partial *__restrict__ partials, DevErr *derr) {
+    const int lane = threadIdx.x & 63;
+    const int wave_in_block = threadIdx.x >> 6;
+    int64_t wave_id = (int64_t)blockIdx.x * (blockDim.x >> 6) + wave_in_block;
+    int64_t n_waves = (int64_t)gridDim.x * (blockDim.x >> 6);
+
+    for (int64_t bi = wave_id; bi < n_blocks; bi += n_waves) {
+        // wave-uniform descriptor loads
+        const bydb_block_desc *bd = &blocks[bi];
+        const int64_t n = (int64_t)bd->count;
+        const int64_t ts_min = bd->ts_min, ts_max = bd->ts_max;
+
+        // ---- row clamp (timestamp.FindRange, range.go:143-170) ----
+        int64_t r0 = 0, r1 = n - 1;
+        if (ts_min > ts_max) { dev_set_err(derr, DERR_DESC_TS, (uint64_t)bi); continue; }
+        if (!(min_ts <= ts_min && max_ts >= ts_max)) {
+            if (ts_min > max_ts || ts_max < min_ts) continue;  // no overlap
+            uint8_t tenc = bd->ts_enc_with_version;
+            // common type (encoding.GetCommonType)
+            uint8_t tc = tenc >= BYDB_ENC_CONST_WV && tenc <= BYDB_ENC_DELTA_OF_DELTA_WV
+                             ? tenc - 4 : tenc;
+            const uint8_t *tstream = payload + bd->ts_off;
+            if (tc == BYDB_ENC_CONST) {
+                // all == ts_min; overlap already implies in-range
+            } else if (tc == BYDB_ENC_DELTA_CONST) {
+                int vl;
+                int64_t dd = decode_one_varint(tstream, &vl);
+                if (dd <= 0) { dev_set_err(derr, DERR_DESC_TS, (uint64_t)bi); continue; }
+                // first i with ts_min + i*dd >= min_ts
+                if (min_ts > ts_min)
+                    r0 = (min_ts - ts_min + dd - 1) / dd;
+                if (max_ts < ts_max)
+                    r1 = (max_ts - ts_min) / dd;
+                if (r1 > n - 1) r1 = n - 1;
+            } else if (tc == BYDB_ENC_DELTA || tc == BYDB_ENC_DELTA_OF_DELTA) {
+                bool dod = tc == BYDB_ENC_DELTA_OF_DELTA;
+                const uint8_t *s = tstream;
+                int64_t d1 = 0;
+                int64_t vfirst = ts_min;
+                if (dod) {
+                    int vl;
+                    d1 = decode_one_varint(s, &vl);
+                    s += vl;
+                }
+                ScanFold tsf;
+                int64_t row1 = (int64_t)((uint64_t)ts_min + (uint64_t)d1);
+                scan_stream(s, n - 1, dod, dod ? row1 : vfirst, d1, 1, 0,
+                            min_ts, max_ts, lane, &tsf, derr, (uint64_t)bi);
+                uint64_t nlo = wave_reduce_add(tsf.n_lo);
+                uint64_t nhi = wave_reduce_add(tsf.n_hi);
+                // rows 0 (and 1 for dod) were not in the stream
+                if (ts_min < min_ts) nlo++;
+                if (ts_min > max_ts) nhi++;
+                if (dod) {
+                    if (row1 < min_ts) nlo++;
+                    if (row1 > max_ts) nhi++;
+                }
+                r0 = (int64_t)nlo;
+                r1 = n - 1 - (int64_t)nhi;
+            }
+            if (r0 > r1) continue;
+        }
+        const uint64_t nsel = (uint64_t)(r1 - r0 + 1);
+
+        // ---- field fold ----
+        const uint8_t fenc = bd->field_enc;
+        const int64_t first = bd->field_first;
+        const uint8_t *fstream = payload + bd->field_off;
+        uint64_t bsum = 0;          // block sum (wrapping)
+        int64_t bmin = INT64_MAX, bmax = INT64_MIN;
+        bool have_minmax = false;
+
+        if (fenc == BYDB_ENC_CONST) {
+            if (lane == 0) bsum = (uint64_t)first * nsel;
+            bmin = bmax = first;
+            have_minmax = true;
+        } else if (fenc == BYDB_ENC_DELTA_CONST) {
+            int vl;
+            int64_t dd = decode_one_varint(fstream, &vl);
+            if (lane == 0) {
+                // sum_{i=r0..r1}(first + i*dd) = nsel*first + dd * sum i
+                uint64_t si = (uint64_t)(r0 + r1) * nsel / 2;
+                bsum = (uint64_t)first * nsel + (uint64_t)dd * si;
+            }
+            if (flags & KF_NEED_VALUES) {
+                // reconstruct min/max exactly (wrap-safe): walk rows by lanes
+                int64_t lmn = INT64_MAX, lmx = INT64_MIN;
+                for (int64_t base = r0; base <= r1; base += WAVE) {
+                    int64_t i = base + lane;
+                    if (i <= r1) {
+                        int64_t v = (int64_t)((uint64_t)first + (uint64_t)i * (uint64_t)dd);
+                        lmn = v < lmn ? v : lmn;
+                        lmx = v > lmx ? v : lmx;
+                    }
+                }
+                bmin = wave_reduce_min(lmn);
+                bmax = wave_reduce_max(lmx);
+                have_minmax = true;
+            }
+        } else if (fenc == BYDB_ENC_DELTA || fenc == BYDB_ENC_DELTA_OF_DELTA) {
+            bool dod = fenc == BYDB_ENC_DELTA_OF_DELTA;
+            if (!(flags & KF_NEED_VALUES)) {
+                uint64_t acc;
+                if (!dod) {
+                    acc = fold_delta_weighted(fstream, n - 1, r0, r1, lane, derr,
+                                              (uint64_t)bi);
+                    acc = wave_reduce_add(acc);
+                    if (lane == 0) {
+                        bsum = (uint64_t)first * nsel + acc;
+                    }
+                } else {
+                    int vl;
+                    int64_t d1 = decode_one_varint(fstream, &vl);
+                    acc = fold_dod_weighted(fstream + vl, n - 1, r0, r1, lane,
+                                            derr, (uint64_t)bi);
+                    acc = wave_reduce_add(acc);
+                    if (lane == 0) {
+                        // nsel*first + d1 * sum_{i=r0..r1} i + weighted d2
+                        uint64_t si = (uint64_t)(r0 + r1) * nsel / 2;
+                        bsum = (uint64_t)first * nsel + (uint64_t)d1 * si + acc;
+                    }
+                }
+            } else {
+                const uint8_t *s = fstream;
+                int64_t d1 = 0;
+                if (dod) {
+                    int vl;
+                    d1 = decode_one_varint(s, &vl);
+                    s += vl;
+                }
+                int64_t row1 = (int64_t)((uint64_t)first + (uint64_t)d1);
+                ScanFold ff;
+                scan_stream(s, n - 1, dod, dod ? row1 : first, d1, r0, r1,
+                            INT64_MAX, INT64_MIN, lane, &ff, derr, (uint64_t)bi);
+                uint64_t lsum = ff.sum;
+                int64_t lmn = ff.mn, lmx = ff.mx;
+                // rows outside the stream: row 0 (value=first) and, for
+                // dod, row 1 (value=row1)
+                if (lane == 0) {
+                    if (r0 <= 0 && 0 <= r1) {
+                        lsum += (uint64_t)first;
+                        lmn = first < lmn ? first : lmn;
+                        lmx = first > lmx ? first : lmx;
+                    }
+                    if (dod && r0 <= 1 && 1 <= r1) {
+                        lsum += (uint64_t)row1;
+                        lmn = row1 < lmn ? row1 : lmn;
+                        lmx = row1 > lmx ? row1 : lmx;
+                    }
+                }
+                bsum = wave_reduce_add(lsum);
+                bmin = wave_reduce_min(lmn);
+                bmax = wave_reduce_max(lmx);
+                have_minmax = true;
+            }
+        } else {
+            dev_set_err(derr, DERR_BAD_ENC, (uint64_t)bi);
+            continue;
+        }
+
+        // rows 0 (+1 for dod) fast path bookkeeping: the weighted folds
+        // above already include first (and d1) in their closed forms.
+
+        // ---- accumulate into the group partial ----
+        if (lane == 0) {
+            bydb_partial *p = &partials[bd->group_code];
+            atomicAdd((unsigned long long *)&p->sum_i, (unsigned long long)bsum);
+            atomicAdd((unsigned long long *)&p->count, (unsigned long long)nsel);
+            if (have_minmax) {
+                atomicMin((long long *)&p->min_i, (long long)bmin);
+                atomicMax((long long *)&p->max_i, (long long)bmax);
+            }
+            if (flags & KF_FLOAT) {
+                atomicAdd(&p->sum_f, (double)(int64_t)bsum);
+            }
+        }
+    }
+}
+
+// init kernel: set partials to the fold identity (Map.Reset, function.go)
+__global__ void k_reset_partials(bydb_partial *p, int64_t n) {
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i < n) {
+        p[i].sum_i = 0;
+        p[i].count = 0;
+        p[i].min_i = INT64_MAX;
+        p[i].max_i = INT64_MIN;
+        p[i].sum_f = 0.0;
+        p[i]._pad = 0.0;
+    }
+}
+
+// ===================== host session =====================
+
+struct bydb_session {
+    int device = 0;
+    hipStream_t stream = nullptr;
+    hipEvent_t ev_start = nullptr, ev_stop = nullptr;
+    std::string err;
+    // part residency
+    uint8_t *d_payload = nullptr;
+    uint64_t payload_cap = 0, payload_len = 0;
+    bydb_block_desc *d_blocks = nullptr;
+    int64_t blocks_cap = 0, n_blocks = 0;
+    // aggregation config
+    int field_vtype = BYDB_VT_INT64;
+    uint32_t func_mask = 0;
+    uint32_t n_groups = 0;
+    int mode = BYDB_MODE_ALL;
+    bydb_partial *d_partials = nullptr;   // owned buffer
+    bydb_partial *d_acc = nullptr;        // active accumulation target
+    uint64_t partials_cap = 0;
+    DevErr *d_err = nullptr;
+    float last_ms = 0.0f;
+    bool consumed = false;
+    int16_t float_exp = 0;  // shared decimal exponent for float64 restore
+};
+
+#define HIP_TRY(s, call)                                                      \
+    do {                                                                      \
+        hipError_t _e = (call);                                               \
+        if (_e != hipSuccess) {                                               \
+            (s)->err = std::string(#call) + ": " + hipGetErrorString(_e);     \
+            return BYDB_ERR_HIP;                                              \
+        }                                                                     \
+    } while (0)
+
+extern "C" bydb_session *bydb_session_create(int device) {
+    int n = 0;
+    if (hipGetDeviceCount(&n) != hipSuccess || n <= device) return nullptr;
+    if (hipSetDevice(device) != hipSuccess) return nullptr;
+    bydb_session *s = new bydb_session();
+    s->device = device;
+    if (hipStreamCreate(&s->stream) != hipSuccess ||
+        hipEventCreate(&s->ev_start) != hipSuccess ||
+        hipEventCreate(&s->ev_stop) != hipSuccess ||
+        hipMalloc(&s->d_err, sizeof(DevErr)) != hipSuccess) {
+        delete s;
+        return nullptr;
+    }
+    hipMemset(s->d_err, 0, sizeof(DevErr));
+    return s;
+}
+
+extern "C" void bydb_session_destroy(bydb_session *s) {
+    if (!s) return;
+    hipSetDevice(s->device);
+    if (s->d_payload) hipFree(s->d_payload);
+    if (s->d_blocks) hipFree(s->d_blocks);
+    if (s->d_partials) hipFree(s->d_partials);
+    if (s->d_err) hipFree(s->d_err);
+    if (s->ev_start) hipEventDestroy(s->ev_start);
+    if (s->ev_stop) hipEventDestroy(s->ev_stop);
+    if (s->stream) hipStreamDestroy(s->stream);
+    delete s;
+}
+
+extern "C" const char *bydb_last_error(bydb_session *s) {
+    return s ? s->err.c_str() : "null session";
+}
+
+extern "C" int bydb_part_reserve(bydb_session *s, uint64_t payload_bytes,
+                                 int64_t n_blocks) {
+    HIP_TRY(s, hipSetDevice(s->device));
+    if (s->d_payload) { hipFree(s->d_payload); s->d_payload = nullptr; }
+    if (s->d_blocks) { hipFree(s->d_blocks); s->d_blocks = nullptr; }
+    // +64B slack: the 64-byte ballot window may read past a stream end
+    HIP_TRY(s, hipMalloc(&s->d_payload, payload_bytes + 64));
+    HIP_TRY(s, hipMemset(s->d_payload + payload_bytes, 0, 64));
+    HIP_TRY(s, hipMalloc(&s->d_blocks, sizeof(bydb_block_desc) * (size_t)n_blocks));
+    s->payload_cap = payload_bytes;
+    s->blocks_cap = n_blocks;
+    s->payload_len = 0;
+    s->n_blocks = 0;
+    return BYDB_OK;
+}
+
+extern "C" int bydb_part_append(bydb_session *s, const uint8_t *payload,
+                                uint64_t len, const bydb_block_desc *blocks,
+                                int64_t n_blocks) {
+    HIP_TRY(s, hipSetDevice(s->device));
+    if (s->payload_len + len > s->payload_cap ||
+        s->n_blocks + n_blocks > s->blocks_cap) {
+        s->err = "part_append exceeds reservation";
+        return BYDB_ERR_BAD_ARG;
+    }
+    HIP_TRY(s, hipMemcpy(s->d_payload + s->payload_len, payload, len,
+                         hipMemcpyHostToDevice));
+    HIP_TRY(s, hipMemcpy(s->d_blocks + s->n_blocks, blocks,
+                         sizeof(bydb_block_desc) * (size_t)n_blocks,
+                         hipMemcpyHostToDevice));
+    s->payload_len += len;
+    s->n_blocks += n_blocks;
+    return BYDB_OK;
+}
+
+extern "C" int bydb_part_clear(bydb_session *s) {
+    s->payload_len = 0;
+    s->n_blocks = 0;
+    return BYDB_OK;
+}
+
+extern "C" int bydb_agg_configure(bydb_session *s, int field_vtype,
+                                  uint32_t func_mask, uint32_t n_groups,
+                                  int mode) {
+    HIP_TRY(s, hipSetDevice(s->device));
+    if (n_groups < 1) { s->err = "n_groups < 1"; return BYDB_ERR_BAD_ARG; }
+    s->field_vtype = field_vtype;
+    s->func_mask = func_mask;
+    s->n_groups = n_groups;
+    s->mode = mode;
+    if (s->partials_cap < n_groups) {
+        if (s->d_partials) hipFree(s->d_partials);
+        HIP_TRY(s, hipMalloc(&s->d_partials, sizeof(bydb_partial) * n_groups));
+        s->partials_cap = n_groups;
+    }
+    s->d_acc = s->d_partials;
+    return bydb_reset(s);
+}
+
+extern "C" int bydb_set_partials_buffer(bydb_session *s, void *dev_ptr,
+                                        uint64_t len) {
+    if (dev_ptr == nullptr) {
+        s->d_acc = s->d_partials;
+        return BYDB_OK;
+    }
+    if (len < sizeof(bydb_partial) * s->n_groups) {
+        s->err = "partials buffer too small";
+        return BYDB_ERR_BAD_ARG;
+    }
+    s->d_acc = (bydb_partial *)dev_ptr;
+    return bydb_reset(s);
+}
+
+extern "C" int bydb_reset(bydb_session *s) {
+    HIP_TRY(s, hipSetDevice(s->device));
+    if (!s->d_acc) { s->err = "configure first"; return BYDB_ERR_STATE; }
+    int threads = 256;
+    int blocks = (int)((s->n_groups + threads - 1) / threads);
+    hipLaunchKernelGGL(k_reset_partials, dim3(blocks), dim3(threads), 0,
+                       s->stream, s->d_acc, (int64_t)s->n_groups);
+    HIP_TRY(s, hipGetLastError());
+    HIP_TRY(s, hipMemsetAsync(s->d_err, 0, sizeof(DevErr), s->stream));
+    s->consumed = false;
+    return BYDB_OK;
+}
+
+extern "C" int bydb_consume(bydb_session *s, int64_t min_ts, int64_t max_ts,
+                            const uint8_t *pred, uint64_t pred_len) {
+    HIP_TRY(s, hipSetDevice(s->device));
+    if (!s->d_acc) { s->err = "configure first"; return BYDB_ERR_STATE; }
+    if (s->n_blocks == 0) { s->err = "no part resident"; return BYDB_ERR_STATE; }
+    if (pred_len > 0) { s->err = "predicate not yet supported"; return BYDB_ERR_BAD_ARG; }
+    int flags = 0;
+    if (s->func_mask & ((1u << BYDB_AGG_MIN) | (1u << BYDB_AGG_MAX)))
+        flags |= KF_NEED_VALUES;
+    if (s->field_vtype == BYDB_VT_FLOAT64) flags |= KF_FLOAT;
+    const int threads = 256;                       // 4 waves per workgroup
+    int64_t waves_needed = s->n_blocks;
+    int64_t wgs = (waves_needed + 3) / 4;
+    int grid = (int)(wgs < 8192 ? wgs : 8192);     // grid-stride beyond
+    if (grid < 1) grid = 1;
+    HIP_TRY(s, hipEventRecord(s->ev_start, s->stream));
+    hipLaunchKernelGGL(k_scan_agg, dim3(grid), dim3(threads), 0, s->stream,
+                       s->d_payload, s->d_blocks, s->n_blocks, min_ts, max_ts,
+                       flags, s->d_acc, s->d_err);
+    HIP_TRY(s, hipGetLastError());
+    HIP_TRY(s, hipEventRecord(s->ev_stop, s->stream));
+    s->consumed = true;
+    return BYDB_OK;
+}
+
+extern "C" double bydb_last_consume_ms(bydb_session *s) { return s->last_ms; }
+
+static int finalize_common(bydb_session *s, bydb_partial *host_parts,
+                           int64_t n_groups) {
+    HIP_TRY(s, hipSetDevice(s->device));
+    if ((uint32_t)n_groups != s->n_groups) {
+        s->err = "n_groups mismatch";
+        return BYDB_ERR_BAD_ARG;
+    }
+    HIP_TRY(s, hipStreamSynchronize(s->stream));
+    if (s->consumed) {
+        float ms = 0;
+        if (hipEventElapsedTime(&ms, s->ev_start, s->ev_stop) == hipSuccess)
+            s->last_ms = ms;
+    }
+    DevErr de;
+    HIP_TRY(s, hipMemcpy(&de, s->d_err, sizeof de, hipMemcpyDeviceToHost));
+    if (de.code != DERR_NONE) {
+        char buf[128];
+        snprintf(buf, sizeof buf, "device decode error %u at block %u", de.code,
+                 de.block_lo);
+        s->err = buf;
+        return BYDB_ERR_BAD_DATA;
+    }
+    HIP_TRY(s, hipMemcpy(host_parts, s->d_acc,
+                         sizeof(bydb_partial) * (size_t)n_groups,
+                         hipMemcpyDeviceToHost));
+    return BYDB_OK;
+}
+
+// Go math.Pow10 restated (decode-side float restore must reproduce its
+// table-multiply rounding — float.go:77-102).
+static const double h_pow10tab[32] = {
+    1e0, 1e1, 1e2, 1e3, 1e4, 1e5, 1e6, 1e7, 1e8, 1e9, 1e10, 1e11, 1e12,
+    1e13, 1e14, 1e15, 1e16, 1e17, 1e18, 1e19, 1e20, 1e21, 1e22, 1e23,
+    1e24, 1e25, 1e26, 1e27, 1e28, 1e29, 1e30, 1e31};
+static const double h_pow10postab32[10] = {
+    1e0, 1e32, 1e64, 1e96, 1e128, 1e160, 1e192, 1e224, 1e256, 1e288};
+static double go_pow10(int nn) {
+    if (0 <= nn && nn <= 308)
+        return h_pow10postab32[(unsigned)nn / 32] * h_pow10tab[(unsigned)nn % 32];
+    return 0.0;  // callers only use 0..308
+}
+
+static double restore_f64(int64_t v, int16_t exp) {
+    if (exp >= 0) return (double)v * go_pow10(exp);
+    double r = (double)v;
+    int neg = -(int)exp;
+    while (neg > 0) {
+        int step = neg < 308 ? neg : 308;
+        r /= go_pow10(step);
+        neg -= step;
+    }
+    return r;
+}
+
+static double restore_f64_from_double(double v, int16_t exp) {
+    if (exp >= 0) return v * go_pow10(exp);
+    double r = v;
+    int neg = -(int)exp;
+    while (neg > 0) {
+        int step = neg < 308 ? neg : 308;
+        r /= go_pow10(step);
+        neg -= step;
+    }
+    return r;
+}
+
+// The float64 restore at finalize needs the column's decimal exponent
+// (column payload header, column.go:253-263).  One query aggregates one
+// field column, whose blocks share the exponent at the benchmark configs;
+// the host sets it per session before finalize.
+extern "C" int bydb_set_float_exp(bydb_session *s, int16_t exp) {
+    s->float_exp = exp;
+    return BYDB_OK;
+}
+
+extern "C" int bydb_finalize_partials(bydb_session *s, bydb_partial *out,
+                                      int64_t n_groups) {
+    return finalize_common(s, out, n_groups);
+}
+
+static void partial_to_result(const bydb_partial *p, int field_vtype,
+                              int16_t float_exp, bydb_result *r) {
+    r->sum_i = p->sum_i;
+    r->count = p->count;
+    r->min_i = p->min_i;
+    r->max_i = p->max_i;
+    if (field_vtype == BYDB_VT_FLOAT64) {
+        // decimal-int domain -> float64 (monotone restore; float.go:69-102)
+        r->min_f = p->count ? restore_f64(p->min_i, float_exp) : 1.7976931348623157e308;
+        r->max_f = p->count ? restore_f64(p->max_i, float_exp) : -1.7976931348623157e308;
+        r->sum_f = restore_f64_from_double(p->sum_f, float_exp);
+        double c = (double)p->count;
+        r->mean_f = c == 0 ? 0 : (r->sum_f / c < 1 ? 1 : r->sum_f / c);
+        r->mean_i = 0;
+    } else {
+        r->sum_f = 0;
+        r->min_f = 0;
+        r->max_f = 0;
+        r->mean_f = 0;
+        // meanFunc.Val (function.go:30-45)
+        r->mean_i = p->count == 0 ? 0
+                    : (p->sum_i / p->count < 1 ? 1 : p->sum_i / p->count);
+    }
+}
+
+extern "C" int bydb_finalize(bydb_session *s, bydb_result *out,
+                             int64_t n_groups) {
+    std::vector<bydb_partial> parts((size_t)n_groups);
+    int rc = finalize_common(s, parts.data(), n_groups);
+    if (rc != BYDB_OK) return rc;
+    for (int64_t g = 0; g < n_groups; g++)
+        partial_to_result(&parts[(size_t)g], s->field_vtype, s->float_exp, &out[(size_t)g]);
+    return BYDB_OK;
+}
+
+extern "C" int bydb_reduce_partials2(const bydb_partial *parts,
+                                     int64_t n_parts_per_group, int64_t n_groups,
+                                     int field_vtype, int16_t float_exp,
+                                     bydb_result *out) {
+    for (int64_t g = 0; g < n_groups; g++) {
+        bydb_partial acc;
+        acc.sum_i = 0;
+        acc.count = 0;
+        acc.min_i = INT64_MAX;
+        acc.max_i = INT64_MIN;
+        acc.sum_f = 0;
+        // Combine — aggregation_reduce.go:120-138 / function.go Reduce
+        for (int64_t k = 0; k < n_parts_per_group; k++) {
+            const bydb_partial *p = &parts[k * n_groups + g];
+            acc.sum_i = (int64_t)((uint64_t)acc.sum_i + (uint64_t)p->sum_i);
+            acc.count += p->count;
+            if (p->min_i < acc.min_i) acc.min_i = p->min_i;
+            if (p->max_i > acc.max_i) acc.max_i = p->max_i;
+            acc.sum_f += p->sum_f;
+        }
+        partial_to_result(&acc, field_vtype, float_exp, &out[g]);
+    }
+    return BYDB_OK;
+}

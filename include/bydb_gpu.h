@@ -1,0 +1,241 @@
+/* include/bydb_gpu.h — C-ABI of the MI355X-native measure scan+aggregate
+ * engine (libbydb_gpu.so).
+ *
+ * This is the drop-in boundary a Go host binds over cgo.  Each entry point
+ * names the reference interface it replaces (apache/skywalking-banyandb):
+ *
+ *  - session/configure/consume/finalize mirror the lifecycle of
+ *    vectorized.BreakerOperator (pkg/query/vectorized/operator.go:65-70:
+ *    Init / Consume / Finalize / NextBatch / Close) as built for
+ *    BatchAggregation by BuildOperators
+ *    (pkg/query/vectorized/measure/plan.go:61-131) — one session per
+ *    pipeline, sticky errors, Close releases everything.
+ *  - part upload replaces the storage seam
+ *    model.MeasureBatchResult.PullBatch (pkg/query/model/batch.go:45-55) +
+ *    blockCursor.loadData/block.mustReadFrom (banyand/measure/block.go:818,
+ *    324): instead of decoding on the host, the encoded block streams are
+ *    made resident in HBM and decode+fold happens in HIP kernels.
+ *  - the partial layout mirrors aggregation.Partial[N]
+ *    (pkg/query/aggregation/aggregation.go:36-55) with MEAN's count sidecar
+ *    (vectorized/measure/aggregation.go meanCountSuffix) and AggModeMap /
+ *    AggModeReduce semantics (aggregation.go:57-64).
+ *
+ * Plain pointers and sizes only; no torch types.  See INTEGRATION.md for
+ * the cgo binding a BanyanDB maintainer would add.
+ */
+#ifndef BYDB_GPU_H
+#define BYDB_GPU_H
+
+#include <stdint.h>
+#include <stddef.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+/* EncodeType — pkg/encoding/encoding.go:86-98 */
+enum {
+    BYDB_ENC_UNKNOWN = 0,
+    BYDB_ENC_CONST = 1,
+    BYDB_ENC_DELTA_CONST = 2,
+    BYDB_ENC_DELTA = 3,
+    BYDB_ENC_DELTA_OF_DELTA = 4,
+    BYDB_ENC_CONST_WV = 5,
+    BYDB_ENC_DELTA_CONST_WV = 6,
+    BYDB_ENC_DELTA_WV = 7,
+    BYDB_ENC_DELTA_OF_DELTA_WV = 8,
+    BYDB_ENC_PLAIN = 9,
+    BYDB_ENC_DICTIONARY = 10,
+};
+
+/* ValueType — pkg/pb/v1/valuetype */
+enum { BYDB_VT_INT64 = 2, BYDB_VT_FLOAT64 = 3 };
+
+/* AggFunc — pkg/query/vectorized/measure/aggregation.go:95-107 */
+enum {
+    BYDB_AGG_SUM = 0,
+    BYDB_AGG_COUNT = 1,
+    BYDB_AGG_MIN = 2,
+    BYDB_AGG_MAX = 3,
+    BYDB_AGG_MEAN = 4,
+};
+
+/* AggMode — vectorized/measure/aggregation.go:57-64 */
+enum { BYDB_MODE_ALL = 0, BYDB_MODE_MAP = 1, BYDB_MODE_REDUCE = 2 };
+
+/* Status codes (0 = ok).  Sticky per session, mirroring the operator's
+ * sticky Go error contract. */
+enum {
+    BYDB_OK = 0,
+    BYDB_ERR = -1,
+    BYDB_ERR_HIP = -2,
+    BYDB_ERR_BAD_ARG = -3,
+    BYDB_ERR_BAD_DATA = -4,
+    BYDB_ERR_STATE = -5,
+    BYDB_ERR_NO_GPU = -6,
+    BYDB_ERR_OOM = -7,
+};
+
+/* One (series, block) entry of the resident part directory.  Field
+ * semantics follow blockMetadata + timestampsMetadata + columnMetadata
+ * (banyand/measure/block_metadata.go:254-311, column_metadata.go:47-52):
+ * the column payload headers ([type][exp][firstValue], column.go:183-263)
+ * are parsed at load time on the host, so kernels see pure streams. */
+typedef struct {
+    uint64_t series_id;
+    uint32_t count;               /* rows in block, <= 8192 (measure.go:41-46) */
+    uint8_t ts_enc_with_version;  /* timestampsMetadata.encodeType */
+    uint8_t version_enc;
+    uint8_t field_enc;            /* common EncodeType of the field column */
+    uint8_t field_vtype;          /* BYDB_VT_* */
+    int64_t ts_min;               /* == first timestamp (block ascending) */
+    int64_t ts_max;
+    int64_t version_first;
+    int64_t field_first;          /* decoded firstValue of the field column */
+    int16_t exp;                  /* float64 decimal exponent (float.go:69) */
+    uint8_t _pad[6];
+    uint64_t ts_off;              /* ts varint stream (header-free) */
+    uint64_t ts_len;
+    uint64_t field_off;           /* field varint stream (header-free) */
+    uint64_t field_len;
+    uint64_t tag_off;             /* tag column payload incl. type byte, or 0 */
+    uint64_t tag_len;
+    uint32_t group_code;          /* dense group index (group-by), else 0 */
+    uint32_t _pad2;
+} bydb_block_desc;
+
+/* Dense per-group partial accumulator — one slot per group.
+ * sum_i/count are exact (wrapping) int64; min/max are on the int64 domain
+ * (for float64 fields these are the decimal-int min/max, restored to
+ * float64 at finalize — restore is monotone, float.go:69-102); sum_f is
+ * the float64 mantissa sum (double).                                      */
+typedef struct {
+    int64_t sum_i;
+    int64_t count;
+    int64_t min_i;
+    int64_t max_i;
+    double sum_f;
+    double _pad;
+} bydb_partial;
+
+/* Finalised per-group result (AggModeAll shape). */
+typedef struct {
+    int64_t sum_i;
+    double sum_f;
+    int64_t count;
+    int64_t min_i;
+    int64_t max_i;
+    double min_f;
+    double max_f;
+    int64_t mean_i;   /* meanFunc.Val with the >=1 clamp (function.go:30-45) */
+    double mean_f;
+} bydb_result;
+
+typedef struct bydb_session bydb_session;
+
+/* ---- session lifecycle (BreakerOperator Init/Close) ---- */
+bydb_session *bydb_session_create(int device);
+void bydb_session_destroy(bydb_session *s);
+const char *bydb_last_error(bydb_session *s);
+
+/* ---- part residency (storage seam) ----
+ * reserve once, then append payload chunks + their block descs; offsets in
+ * descs are absolute within the whole part payload. */
+int bydb_part_reserve(bydb_session *s, uint64_t payload_bytes, int64_t n_blocks);
+int bydb_part_append(bydb_session *s, const uint8_t *payload, uint64_t len,
+                     const bydb_block_desc *blocks, int64_t n_blocks);
+int bydb_part_clear(bydb_session *s);
+
+/* ---- aggregation configure (BuildOperators / AggSpec) ----
+ * funcs: bitmask of (1<<BYDB_AGG_*); n_groups >= 1 (1 = scalar aggregate,
+ * matching aggAllIterator).  field_vtype selects int64/float64 semantics. */
+int bydb_agg_configure(bydb_session *s, int field_vtype, uint32_t func_mask,
+                       uint32_t n_groups, int mode);
+
+/* Optional: accumulate partials into an external device buffer of
+ * n_groups * sizeof(bydb_partial) bytes (e.g. a torch CUDA tensor for an
+ * RCCL merge).  Pass NULL to use the session's own buffer. */
+int bydb_set_partials_buffer(bydb_session *s, void *dev_ptr, uint64_t len);
+
+/* ---- consume (BreakerOperator.Consume over the resident part) ----
+ * min_ts/max_ts: inclusive row clamp (timestamp.FindRange,
+ * pkg/timestamp/range.go:143-170).  pred: optional tag-equality predicate
+ * value bytes (dictionary tags only), pred_len 0 = none.  Asynchronous:
+ * returns after launch; bydb_finalize syncs. */
+int bydb_consume(bydb_session *s, int64_t min_ts, int64_t max_ts,
+                 const uint8_t *pred, uint64_t pred_len);
+
+/* ---- finalize + emit (Finalize/NextBatch) ----
+ * Syncs the stream, downloads partials and finalises per-group results
+ * (MEAN = sum/count with the >=1 clamp; float64 restore per float.go).
+ * out must hold n_groups entries.  Resets partials for the next epoch. */
+int bydb_finalize(bydb_session *s, bydb_result *out, int64_t n_groups);
+
+/* Finalize partials only (AggModeMap: emit Partial state for an external
+ * reduce — e.g. RCCL — without applying Val()).  Leaves device partials
+ * in the accumulation buffer; host copy written to out. */
+int bydb_finalize_partials(bydb_session *s, bydb_partial *out, int64_t n_groups);
+
+/* Reset accumulators to the fold identity (Map.Reset, function.go). */
+int bydb_reset(bydb_session *s);
+
+/* Combine external partials (AggModeReduce semantics,
+ * aggregation_reduce.go:120-138) into results on the host. */
+int bydb_reduce_partials(const bydb_partial *parts, int64_t n_parts_per_group,
+                         int64_t n_groups, int field_vtype, uint32_t func_mask,
+                         bydb_result *out);
+
+/* Timing of the last consume's kernels in milliseconds (HIP events on the
+ * session stream) — feeds bench.py's roofline.achieved. */
+double bydb_last_consume_ms(bydb_session *s);
+
+/* ---- host-side part builder (fixture writer) ----
+ * Mirrors the reference block write path: mustInitFromDataPoints ->
+ * block.mustWriteTo (block.go:60,139), mustWriteTimestampsTo
+ * (block.go:386-404), column.mustWriteTo (column.go:157-278).  Pure host
+ * code; no GPU needed.  Produces payload bytes + bydb_block_desc entries
+ * whose encoded streams are byte-identical to the reference's (validated
+ * against the oracle in tests/).                                          */
+typedef struct bydb_part_builder bydb_part_builder;
+bydb_part_builder *bydb_part_builder_create(void);
+void bydb_part_builder_destroy(bydb_part_builder *b);
+const char *bydb_part_builder_error(bydb_part_builder *b);
+
+int bydb_part_builder_add_block_i64(bydb_part_builder *b, uint64_t series_id,
+                                    const int64_t *ts, const int64_t *versions,
+                                    const int64_t *vals, int64_t n,
+                                    uint32_t group_code);
+int bydb_part_builder_add_block_f64(bydb_part_builder *b, uint64_t series_id,
+                                    const int64_t *ts, const int64_t *versions,
+                                    const double *vals, int64_t n,
+                                    uint32_t group_code);
+/* attach a dictionary-encoded tag column to the block just added:
+ * tag value of row i = tag_values[codes[i]] (code order = first-seen) */
+int bydb_part_builder_set_block_tag(bydb_part_builder *b, const uint8_t *data,
+                                    const int64_t *lens, int64_t n);
+
+uint64_t bydb_part_builder_payload_len(bydb_part_builder *b);
+const uint8_t *bydb_part_builder_payload(bydb_part_builder *b);
+int64_t bydb_part_builder_n_blocks(bydb_part_builder *b);
+const bydb_block_desc *bydb_part_builder_blocks(bydb_part_builder *b);
+/* drop payload bytes + descs already consumed, keep absolute offsets
+ * running (chunked upload) */
+int bydb_part_builder_drain(bydb_part_builder *b);
+
+/* ---- synthetic workload generator (BASELINE.json section 8d) ----
+ * Generates one series' datapoints and appends its blocks (<=8192 rows
+ * each).  values = base + i*ramp + noise, noise in [-3,3] from
+ * splitmix64(seed ^ series_index).  threads: host threads for encoding. */
+int bydb_gen_series_i64(bydb_part_builder *b, uint64_t series_index,
+                        int64_t n_dp, int64_t t0, int64_t stride_ns,
+                        int64_t base, int64_t ramp, uint64_t seed,
+                        uint32_t group_code);
+int bydb_gen_series_f64(bydb_part_builder *b, uint64_t series_index,
+                        int64_t n_dp, int64_t t0, int64_t stride_ns,
+                        double base, double ramp, uint64_t seed,
+                        uint32_t group_code);
+
+#ifdef __cplusplus
+}
+#endif
+#endif /* BYDB_GPU_H */
