@@ -150,6 +150,16 @@ def _load():
     lib.bydb_gen_series_f64.argtypes = [C.c_void_p, C.c_uint64, C.c_int64, C.c_int64,
                                         C.c_int64, C.c_double, C.c_double, C.c_uint64,
                                         C.c_uint32]
+    lib.bydb_gen_series_bulk_i64.restype = C.c_int
+    lib.bydb_gen_series_bulk_i64.argtypes = [C.c_void_p, C.c_uint64, C.c_int64,
+                                             C.c_int64, C.c_int64, C.c_int64,
+                                             C.c_int64, C.c_int64, C.c_uint64,
+                                             C.c_uint32, C.c_int]
+    lib.bydb_gen_series_bulk_f64.restype = C.c_int
+    lib.bydb_gen_series_bulk_f64.argtypes = [C.c_void_p, C.c_uint64, C.c_int64,
+                                             C.c_int64, C.c_int64, C.c_int64,
+                                             C.c_double, C.c_double, C.c_uint64,
+                                             C.c_uint32, C.c_int]
     return lib
 
 
@@ -204,6 +214,20 @@ class PartBuilder:
                        seed, group_code=0):
         self._ck(_lib.bydb_gen_series_f64(self._h, series_index, n_dp, t0,
                                           stride_ns, base, ramp, seed, group_code))
+
+    def gen_bulk_i64(self, first_index, n_series, n_dp, t0, stride_ns,
+                     base_step, ramp, seed, group_mod=0, threads=None):
+        threads = threads or os.cpu_count() or 8
+        self._ck(_lib.bydb_gen_series_bulk_i64(
+            self._h, first_index, n_series, n_dp, t0, stride_ns, base_step,
+            ramp, seed, group_mod, threads))
+
+    def gen_bulk_f64(self, first_index, n_series, n_dp, t0, stride_ns,
+                     base_step, ramp, seed, group_mod=0, threads=None):
+        threads = threads or os.cpu_count() or 8
+        self._ck(_lib.bydb_gen_series_bulk_f64(
+            self._h, first_index, n_series, n_dp, t0, stride_ns, base_step,
+            ramp, seed, group_mod, threads))
 
     @property
     def payload(self) -> bytes:

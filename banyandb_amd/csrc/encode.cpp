@@ -608,3 +608,82 @@ extern "C" int bydb_gen_series_f64(bydb_part_builder *b, uint64_t series_index,
     }
     return BYDB_OK;
 }
+
+// ===================== threaded bulk generator =====================
+// Generates [first_index, first_index+n_series) series in parallel with
+// per-thread private builders, then splices them into b preserving series
+// order.  base(series) = series_index * base_step; group_code =
+// series_index % group_mod (group_mod 0 -> all group 0).
+#include <thread>
+
+static void splice_builder(bydb_part_builder *dst, bydb_part_builder *src) {
+    uint64_t shift = dst->base_off + dst->payload.size();
+    dst->payload.insert(dst->payload.end(), src->payload.begin(),
+                        src->payload.end());
+    for (bydb_block_desc d : src->blocks) {
+        d.ts_off += shift;
+        d.field_off += shift;
+        if (d.tag_len) d.tag_off += shift;
+        dst->blocks.push_back(d);
+    }
+}
+
+template <typename GenFn>
+static int bulk_gen(bydb_part_builder *b, uint64_t first_index,
+                    int64_t n_series, int n_threads, GenFn gen) {
+    if (n_threads < 1) n_threads = 1;
+    if (n_threads > 64) n_threads = 64;
+    if ((int64_t)n_threads > n_series) n_threads = (int)n_series;
+    std::vector<bydb_part_builder> locals((size_t)n_threads);
+    std::vector<int> rcs((size_t)n_threads, BYDB_OK);
+    std::vector<std::thread> threads;
+    int64_t per = (n_series + n_threads - 1) / n_threads;
+    for (int t = 0; t < n_threads; t++) {
+        threads.emplace_back([&, t]() {
+            int64_t lo = t * per, hi = lo + per < n_series ? lo + per : n_series;
+            for (int64_t si = lo; si < hi; si++) {
+                int rc = gen(&locals[(size_t)t], first_index + (uint64_t)si);
+                if (rc != BYDB_OK) { rcs[(size_t)t] = rc; return; }
+            }
+        });
+    }
+    for (auto &th : threads) th.join();
+    for (int t = 0; t < n_threads; t++) {
+        if (rcs[(size_t)t] != BYDB_OK) {
+            b->err = locals[(size_t)t].err;
+            return rcs[(size_t)t];
+        }
+        splice_builder(b, &locals[(size_t)t]);
+    }
+    return BYDB_OK;
+}
+
+extern "C" int bydb_gen_series_bulk_i64(bydb_part_builder *b,
+                                        uint64_t first_index, int64_t n_series,
+                                        int64_t n_dp, int64_t t0,
+                                        int64_t stride_ns, int64_t base_step,
+                                        int64_t ramp, uint64_t seed,
+                                        uint32_t group_mod, int n_threads) {
+    return bulk_gen(b, first_index, n_series, n_threads,
+                    [&](bydb_part_builder *lb, uint64_t si) {
+                        uint32_t gc = group_mod ? (uint32_t)(si % group_mod) : 0;
+                        return bydb_gen_series_i64(lb, si, n_dp, t0, stride_ns,
+                                                   (int64_t)si * base_step, ramp,
+                                                   seed, gc);
+                    });
+}
+
+extern "C" int bydb_gen_series_bulk_f64(bydb_part_builder *b,
+                                        uint64_t first_index, int64_t n_series,
+                                        int64_t n_dp, int64_t t0,
+                                        int64_t stride_ns, double base_step,
+                                        double ramp, uint64_t seed,
+                                        uint32_t group_mod, int n_threads) {
+    return bulk_gen(b, first_index, n_series, n_threads,
+                    [&](bydb_part_builder *lb, uint64_t si) {
+                        uint32_t gc = group_mod ? (uint32_t)(si % group_mod) : 0;
+                        return bydb_gen_series_f64(lb, si, n_dp, t0, stride_ns,
+                                                   (double)si * base_step, ramp,
+                                                   seed, gc);
+                    });
+}

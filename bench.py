@@ -1,0 +1,239 @@
+#!/usr/bin/env python3
+"""bench.py — BanyanDB measure scan+aggregate throughput on MI355X.
+
+Workload (BASELINE.json configs[1], the headline single-GPU config):
+10k series x 1M int64 datapoints (10^10 dp), sum+count aggregate, no
+predicate, whole time range.  Encoded measure blocks (the reference's own
+on-disk stream format) are resident in HBM before the timed region; one
+step = one pass of the decode+fold hot path over the whole resident part.
+
+N>1 (launched by the driver via torch.distributed.run): weak scaling —
+each rank owns one time-bucket shard of the same shape (SURVEY section 8e,
+storage/tsdb.go:61 segment model); the group-partial exchange is an RCCL
+reduce over the partials tensor.  value = datapoints all ranks processed
+per second.
+
+CPU baseline: the CPU oracle (restated reference Go path, 'port') timed on
+this box's host cores over a bounded sample of the same workload — a
+reported baseline, not the target.
+"""
+import argparse
+import ctypes
+import json
+import os
+import sys
+import time
+
+REPO = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, REPO)
+
+T0 = 1_700_000_000_000_000_000  # T0 = 1.7e18 ns
+STRIDE = 10 ** 6                # 1 ms
+SEED = 0xB4DB
+ALGO_BYTES_PER_DP = 16          # 8B value + 8B timestamp
+HBM_PEAK_GBS = 8000.0           # MI355X_MICROARCH.md: HBM3E 8.0 TB/s spec
+
+
+def log(msg):
+    if int(os.environ.get("RANK", "0")) == 0:
+        print(f"[bench] {msg}", file=sys.stderr, flush=True)
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=3)
+    ap.add_argument("--warmup", type=int, default=1)
+    ap.add_argument("--series", type=int,
+                    default=int(os.environ.get("BYDB_BENCH_SERIES", 10000)))
+    ap.add_argument("--dp", type=int,
+                    default=int(os.environ.get("BYDB_BENCH_DP", 1_000_000)))
+    ap.add_argument("--skip-cpu-baseline", action="store_true",
+                    default=os.environ.get("BYDB_SKIP_CPU_BASELINE") == "1")
+    args = ap.parse_args()
+
+    import torch
+    import banyandb_amd as ba
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    if args.gpus > 1 and world == 1:
+        log("WARNING: --gpus > 1 without torchrun; measuring 1 rank")
+    n_gpus = world if world > 1 else 1
+    dist = None
+    if world > 1:
+        import torch.distributed as dist_mod
+        dist = dist_mod
+        torch.cuda.set_device(local_rank)
+        dist.init_process_group("nccl")
+
+    n_series, n_dp = args.series, args.dp
+    full_config = n_series == 10000 and n_dp == 1_000_000
+    workload = (f"{n_series}_series_x_{n_dp}_int64_sum_count"
+                if not full_config else "10k_series_x_1M_int64_sum_count")
+
+    # ---- build + upload the rank's shard (untimed) ----
+    threads = max(1, (os.cpu_count() or 8) // max(world, 1))
+    t0_r = T0 + rank * n_dp * STRIDE
+    seed_r = SEED ^ (rank << 32)
+    total_dp_rank = n_series * n_dp
+    est_payload = int(total_dp_rank * 1.05) + n_series * 64  # ~1.0 B/dp + block overhead
+    est_blocks = n_series * ((n_dp + 8191) // 8192)
+
+    sess = ba.Session(local_rank)
+    sess.reserve(est_payload, est_blocks)
+
+    gen_t = time.perf_counter()
+    chunk = 500
+    b = ba.PartBuilder()
+    uploaded_dp = 0
+    for s0 in range(0, n_series, chunk):
+        ns = min(chunk, n_series - s0)
+        b.gen_bulk_i64(s0, ns, n_dp, t0_r, STRIDE, 1000, 1, seed_r,
+                       group_mod=0, threads=threads)
+        sess.append(b)
+        uploaded_dp += ns * n_dp
+        b.drain()
+    gen_s = time.perf_counter() - gen_t
+    log(f"rank {rank}: generated+uploaded {uploaded_dp} dp in {gen_s:.1f}s")
+
+    n_groups = 1
+    sess.configure(ba.VT_INT64, [ba.AGG_SUM, ba.AGG_COUNT], n_groups=n_groups)
+
+    # partials live in a torch CUDA tensor so the merge is RCCL over xGMI
+    part_t = None
+    if world > 1:
+        part_t = torch.zeros(n_groups * 6, dtype=torch.int64,
+                             device=f"cuda:{local_rank}")
+        sess.set_partials_buffer(part_t.data_ptr(), part_t.numel() * 8)
+
+    def one_step():
+        sess.reset()
+        sess.consume(min_ts=t0_r, max_ts=t0_r + n_dp * STRIDE)
+        if world > 1:
+            # sync the session stream, then RCCL-merge the partials:
+            # sums+counts reduce with SUM (AggModeReduce Combine semantics)
+            parts = sess.finalize_partials()
+            dist.all_reduce(part_t[0:2], op=dist.ReduceOp.SUM)
+            return parts
+        return sess.finalize_partials()
+
+    # ---- warmup ----
+    for _ in range(args.warmup):
+        one_step()
+    # ---- timed region ----
+    launch_ms = []
+    if world > 1:
+        dist.barrier()
+    torch.cuda.synchronize(local_rank) if torch.cuda.is_available() else None
+    t_start = time.perf_counter()
+    for _ in range(args.steps):
+        one_step()
+        launch_ms.append(sess.last_consume_ms())
+    torch.cuda.synchronize(local_rank) if torch.cuda.is_available() else None
+    if world > 1:
+        dist.barrier()
+    elapsed = time.perf_counter() - t_start
+    if world > 1:
+        t = torch.tensor([elapsed], device=f"cuda:{local_rank}")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    # ---- correctness spot-check (outside the timed region) ----
+    parts = sess.finalize_partials() if world == 1 else None
+    if world == 1:
+        assert parts[0].count == total_dp_rank, \
+            f"count {parts[0].count} != {total_dp_rank}"
+
+    ms_per_step = elapsed * 1000.0 / args.steps
+    total_dp = total_dp_rank * max(world, 1)
+    value = total_dp * args.steps / elapsed
+
+    avg_launch_ms = sum(launch_ms) / len(launch_ms)
+    algo_bytes_per_launch = ALGO_BYTES_PER_DP * total_dp_rank
+    achieved_gbs = algo_bytes_per_launch / (avg_launch_ms / 1e3) / 1e9
+
+    traffic = None
+    tpath = os.path.join(REPO, "profiles", "pmc_traffic.json")
+    if os.path.exists(tpath):
+        try:
+            tj = json.load(open(tpath))
+            if tj.get("workload") == workload:
+                traffic = tj.get("bytes_per_launch")
+        except Exception:
+            pass
+
+    cpu_baseline = None
+    if rank == 0 and world == 1 and not args.skip_cpu_baseline:
+        cpu_baseline = run_cpu_baseline(n_dp)
+
+    if rank == 0:
+        out = {
+            "metric": "datapoints/sec scanned+aggregated",
+            "value": value,
+            "unit": "datapoints/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "int64",
+            "data": "synthetic",
+            "config": {
+                "workload": workload,
+                "series_per_shard": n_series,
+                "datapoints_per_series": n_dp,
+                "agg": "sum+count",
+                "parallelism": f"time-bucket shards x{max(world,1)}",
+            },
+            "roofline": {
+                "bound": "hbm",
+                "achieved": achieved_gbs,
+                "peak": HBM_PEAK_GBS,
+                "unit": "GB/s",
+                "frac": achieved_gbs / HBM_PEAK_GBS,
+                "traffic": traffic,
+            },
+            "cpu_baseline": cpu_baseline,
+        }
+        print(json.dumps(out), flush=True)
+
+    if world > 1:
+        dist.destroy_process_group()
+    sess.close()
+
+
+def run_cpu_baseline(n_dp):
+    """Time the CPU oracle (restated reference Go path) on a bounded sample
+    of the same workload.  Single thread (the reference decode+fold is
+    single-goroutine per cursor chain; we report 1 core)."""
+    sys.path.insert(0, os.path.join(REPO, "oracle"))
+    sys.path.insert(0, os.path.join(REPO, "tests"))
+    import banyandb_amd as ba
+    import oracle as o
+    from helpers import oracle_blocks
+    sample_series = int(os.environ.get("BYDB_CPU_BASELINE_SERIES", 1000))
+    b = ba.PartBuilder()
+    b.gen_bulk_i64(0, sample_series, n_dp, T0, STRIDE, 1000, 1, SEED,
+                   group_mod=0, threads=os.cpu_count() or 8)
+    payload, blocks = oracle_blocks(b)
+    t = time.perf_counter()
+    res = o.scan_agg(payload, blocks, o.VT_INT64)[0]
+    dt = time.perf_counter() - t
+    dp = sample_series * n_dp
+    assert res.count == dp
+    return {
+        "value": dp / dt,
+        "unit": "datapoints/s",
+        "cores": 1,
+        "kind": "port",
+        "sample": f"{sample_series} of 10000 series x {n_dp} dp, one pass, "
+                  f"single thread ({dt:.1f}s)",
+    }
+
+
+if __name__ == "__main__":
+    main()

