@@ -581,14 +581,23 @@ struct bydb_session {
 
 extern "C" bydb_session *bydb_session_create(int device) {
     int n = 0;
-    if (hipGetDeviceCount(&n) != hipSuccess || n <= device) return nullptr;
-    if (hipSetDevice(device) != hipSuccess) return nullptr;
+    hipError_t e = hipGetDeviceCount(&n);
+    if (e != hipSuccess || n <= device) {
+        fprintf(stderr, "bydb_session_create: hipGetDeviceCount=%s n=%d dev=%d\n",
+                hipGetErrorString(e), n, device);
+        return nullptr;
+    }
+    if ((e = hipSetDevice(device)) != hipSuccess) {
+        fprintf(stderr, "bydb_session_create: hipSetDevice=%s\n", hipGetErrorString(e));
+        return nullptr;
+    }
     bydb_session *s = new bydb_session();
     s->device = device;
-    if (hipStreamCreate(&s->stream) != hipSuccess ||
-        hipEventCreate(&s->ev_start) != hipSuccess ||
-        hipEventCreate(&s->ev_stop) != hipSuccess ||
-        hipMalloc(&s->d_err, sizeof(DevErr)) != hipSuccess) {
+    if ((e = hipStreamCreate(&s->stream)) != hipSuccess ||
+        (e = hipEventCreate(&s->ev_start)) != hipSuccess ||
+        (e = hipEventCreate(&s->ev_stop)) != hipSuccess ||
+        (e = hipMalloc(&s->d_err, sizeof(DevErr))) != hipSuccess) {
+        fprintf(stderr, "bydb_session_create: init failed: %s\n", hipGetErrorString(e));
         delete s;
         return nullptr;
     }
