@@ -77,6 +77,15 @@ def _build_if_possible():
 
 
 def _load():
+    # Load torch's bundled HIP runtime FIRST: our .so and torch both carry
+    # libamdhip64 with the same soname; whichever loads first is reused by
+    # the other.  The engine runs fine on torch's runtime, but torch breaks
+    # on the system one — so pin the order here (observed on the MI355X
+    # box: torch-first OK both ways, engine-first torch.cuda dead).
+    try:
+        import torch  # noqa: F401
+    except Exception:
+        pass
     if not os.path.exists(_SO):
         try:
             _build_if_possible()
