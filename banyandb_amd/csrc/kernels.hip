@@ -130,53 +130,113 @@ __device__ __forceinline__ void dev_set_err(DevErr *e, unsigned code, uint64_t b
 
 // ---------------- per-block decode primitives ----------------
 
-// Weighted delta fold (fast path, no value reconstruction):
-//   returns sum_{j=jlo..jhi} d_j * w(j) over the varint stream,
-//   w(j) = nsel for j <= r0, else (r1 - j + 1); deltas are 1-indexed.
-// Used for EncodeTypeDelta sum. jend = min(n-1, r1).
-__device__ uint64_t fold_delta_weighted(const uint8_t *stream, int64_t n_deltas,
-                                        int64_t r0, int64_t r1, int lane,
-                                        DevErr *derr, uint64_t bi) {
-    uint64_t acc = 0;
+// Fast weighted delta fold: 256-byte windows (one dword per lane) with an
+// all-1-byte fast case — the dominant shape for delta-encoded telemetry
+// (zigzag deltas < 64 encode to one byte, int.go:84-88).  For a window
+// fully inside the linear-weight region (j > r0, tail within [.., jend]),
+//   sum d_j * (r1 - j + 1) = (r1+1) * S1 - S2,  S1 = sum d_j, S2 = sum j*d_j
+// so the per-value work is 32-bit byte math folded into two lane
+// accumulators.  Windows with multi-byte varints or boundary weights fall
+// back to one 64-byte ballot step (exact same semantics), then resume.
+// The next window's dwords are loaded before the current one is processed
+// (the fast path's advance is a compile-time +256), hiding HBM latency.
+__device__ uint64_t fold_delta_weighted_fast(const uint8_t *stream,
+                                             int64_t n_deltas, int64_t r0,
+                                             int64_t r1, int lane, DevErr *derr,
+                                             uint64_t bi) {
+    uint64_t gen_acc = 0;   // generic-weight accumulator (boundary/slow)
+    uint64_t s1 = 0;        // sum of d_j  (linear region)
+    uint64_t s2 = 0;        // sum of j*d_j (linear region)
     uint64_t pos = 0;
     int64_t j = 1;
     int64_t jend = r1 < n_deltas ? r1 : n_deltas;
     uint64_t nsel = (uint64_t)(r1 - r0 + 1);
+    // 256-byte windows at 4-byte-aligned absolute addresses; `off` is the
+    // unconsumed-prefix length (bytes before the next value start).
+    const uint32_t *wp =
+        (const uint32_t *)((uintptr_t)(stream + pos) & ~(uintptr_t)3);
+    unsigned off = (unsigned)((uintptr_t)(stream + pos) - (uintptr_t)wp);
+    uint32_t w = wp[lane];
     while (j <= jend) {
-        uint8_t b = stream[pos + (uint64_t)lane];
-        uint64_t emask = __ballot(b < 0x80);
-        if (emask == 0) { dev_set_err(derr, DERR_BAD_STREAM, bi); return acc; }
-        int rank = __popcll(emask & lanemask_lt(lane));
-        int64_t myj = j + rank;
-        bool is_term = (b < 0x80) && (myj <= jend);
-        int64_t d;
-        if (emask == ~0ull) {
-            d = zz_dec(b);
-        } else if (is_term) {
-            uint64_t below = emask & lanemask_lt(lane);
-            int start = below ? (64 - __clzll(below)) : 0;
-            uint64_t u = 0;
-            unsigned sh = 0;
-            for (int i = start; i < lane; ++i) {
-                u |= (uint64_t)(stream[pos + (uint64_t)i] & 0x7f) << sh;
-                sh += 7;
+        uint32_t w_next = wp[64 + lane];               // prefetch next window
+        uint32_t valid = (lane == 0 && off) ? (0xFFFFFFFFu << (8 * off))
+                                            : 0xFFFFFFFFu;
+        uint32_t want = 0x80808080u & valid;
+        bool all_term = ((~w) & want) == want;
+        int64_t nvals_window = 256 - (int64_t)off;
+        bool fast = __all(all_term) && (jend - j + 1) >= nvals_window;
+        if (fast) {
+            bool linear = j > r0;   // whole window past the r0 boundary
+            int32_t sum_d = 0, sum_kd = 0;
+#pragma unroll
+            for (int k = 0; k < 4; k++) {
+                int32_t bidx = 4 * lane + k;
+                uint32_t b = (w >> (8 * k)) & 0xffu;
+                int32_t d = (int32_t)(b >> 1) ^ -(int32_t)(b & 1);
+                if ((unsigned)bidx < off) d = 0;       // lane-0 masked prefix
+                if (linear) {
+                    sum_d += d;
+                    sum_kd += (bidx - (int32_t)off) * d;
+                } else {
+                    int64_t myj = j + (bidx - (int64_t)off);
+                    uint64_t wt = myj <= r0 ? nsel : (uint64_t)(r1 - myj + 1);
+                    if ((unsigned)bidx >= off)
+                        gen_acc += (uint64_t)(int64_t)d * wt;
+                }
             }
-            u |= (uint64_t)b << sh;
-            d = zz_dec(u);
-        } else {
-            d = 0;
+            if (linear) {
+                s1 += (uint64_t)(int64_t)sum_d;
+                s2 += (uint64_t)j * (uint64_t)(int64_t)sum_d +
+                      (uint64_t)(int64_t)sum_kd;
+            }
+            pos += (uint64_t)nvals_window;
+            j += nvals_window;
+            wp += 64;
+            off = 0;
+            w = w_next;
+            continue;
         }
-        if (is_term) {
-            uint64_t w = myj <= r0 ? nsel : (uint64_t)(r1 - myj + 1);
-            acc += (uint64_t)d * w;
+        // slow step: one 64-byte ballot window from pos (multi-byte
+        // varints and/or the stream tail) — identical semantics to v1
+        {
+            uint8_t b = stream[pos + (uint64_t)lane];
+            uint64_t emask = __ballot(b < 0x80);
+            if (emask == 0) { dev_set_err(derr, DERR_BAD_STREAM, bi); break; }
+            int rank = __popcll(emask & lanemask_lt(lane));
+            int64_t myj = j + rank;
+            bool is_term = (b < 0x80) && (myj <= jend);
+            int64_t d;
+            if (emask == ~0ull) {
+                d = zz_dec(b);
+            } else if (is_term) {
+                uint64_t below = emask & lanemask_lt(lane);
+                int start = below ? (64 - __clzll(below)) : 0;
+                uint64_t u = 0;
+                unsigned sh = 0;
+                for (int i = start; i < lane; ++i) {
+                    u |= (uint64_t)(stream[pos + (uint64_t)i] & 0x7f) << sh;
+                    sh += 7;
+                }
+                u |= (uint64_t)b << sh;
+                d = zz_dec(u);
+            } else {
+                d = 0;
+            }
+            if (is_term) {
+                uint64_t wt = myj <= r0 ? nsel : (uint64_t)(r1 - myj + 1);
+                gen_acc += (uint64_t)d * wt;
+            }
+            int nterm = __popcll(emask);
+            if (j + nterm > jend) { j = jend + 1; break; }
+            j += nterm;
+            pos += (uint64_t)(64 - __clzll(emask));
+            wp = (const uint32_t *)((uintptr_t)(stream + pos) & ~(uintptr_t)3);
+            off = (unsigned)((uintptr_t)(stream + pos) - (uintptr_t)wp);
+            w = wp[lane];
         }
-        int nterm = __popcll(emask);
-        if (j + nterm > jend) break;
-        j += nterm;
-        // advance past the last terminator in this window
-        pos += (uint64_t)(64 - __clzll(emask));
     }
-    return acc;
+    // fold the linear region: sum d_j*(r1-j+1) = (r1+1)*S1 - S2 (mod 2^64)
+    return gen_acc + (uint64_t)(r1 + 1) * s1 - s2;
 }
 
 // Weighted delta-of-delta fold: sum_{i=r0..r1} v_i with
@@ -456,8 +516,8 @@ __global__ __launch_bounds__(256) void k_scan_agg(
             if (!(flags & KF_NEED_VALUES)) {
                 uint64_t acc;
                 if (!dod) {
-                    acc = fold_delta_weighted(fstream, n - 1, r0, r1, lane, derr,
-                                              (uint64_t)bi);
+                    acc = fold_delta_weighted_fast(fstream, n - 1, r0, r1,
+                                                   lane, derr, (uint64_t)bi);
                     acc = wave_reduce_add(acc);
                     if (lane == 0) {
                         bsum = (uint64_t)first * nsel + acc;
@@ -627,9 +687,10 @@ extern "C" int bydb_part_reserve(bydb_session *s, uint64_t payload_bytes,
     HIP_TRY(s, hipSetDevice(s->device));
     if (s->d_payload) { hipFree(s->d_payload); s->d_payload = nullptr; }
     if (s->d_blocks) { hipFree(s->d_blocks); s->d_blocks = nullptr; }
-    // +64B slack: the 64-byte ballot window may read past a stream end
-    HIP_TRY(s, hipMalloc(&s->d_payload, payload_bytes + 64));
-    HIP_TRY(s, hipMemset(s->d_payload + payload_bytes, 0, 64));
+    // +1KiB slack: the 256-byte window loop prefetches one window ahead
+    // and may read past the last stream's end
+    HIP_TRY(s, hipMalloc(&s->d_payload, payload_bytes + 1024));
+    HIP_TRY(s, hipMemset(s->d_payload + payload_bytes, 0, 1024));
     HIP_TRY(s, hipMalloc(&s->d_blocks, sizeof(bydb_block_desc) * (size_t)n_blocks));
     s->payload_cap = payload_bytes;
     s->blocks_cap = n_blocks;

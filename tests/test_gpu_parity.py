@@ -47,6 +47,13 @@ def make_builder_styles(seed=11):
         # multi-byte-heavy Delta (large deltas)
         b.add_block_i64(sid, ts, vers,
                         [rng.randint(-2**61, 2**61) for _ in range(n)]); sid += 1
+        # mixed-width Delta: 1-byte runs with periodic multi-byte deltas,
+        # exercising the 256-byte fast window -> 64-byte slow step handoff
+        v, vals = 0, []
+        for i in range(n):
+            v += rng.randint(-3, 3) if i % 37 else rng.randint(-10**9, 10**9)
+            vals.append(v)
+        b.add_block_i64(sid, ts, vers, vals); sid += 1
     return b
 
 
