@@ -130,6 +130,125 @@ __device__ __forceinline__ void dev_set_err(DevErr *e, unsigned code, uint64_t b
 
 // ---------------- per-block decode primitives ----------------
 
+// ---- dense all-1-byte delta fold ----
+// When the stream length equals the delta count, every varint is one byte
+// (int.go:84-88), so the stream IS the zigzag byte array and the weighted
+// fold becomes dense byte math: per dword, SWAR zigzag
+// (d = (b>>1) ^ (-(b&1) as 0x00/0xff)) turns 4 bytes into 4 signed i8
+// lanes, and v_dot4_i32_i8 folds them against {1,1,1,1} / {0,1,2,3}
+// weights.  Per-lane accumulators stay in int32 for a whole block
+// (<= 128 bytes/lane * 63 * 8191 < 2^31).
+__device__ __forceinline__ int32_t dot4_i8(uint32_t a, uint32_t b, int32_t c) {
+#if defined(__gfx950__) || defined(__AMDGCN__)
+    return __builtin_amdgcn_sdot4((int)a, (int)b, c, false);
+#else
+    int32_t r = c;
+    for (int k = 0; k < 4; k++)
+        r += (int32_t)(int8_t)(a >> (8 * k)) * (int32_t)(int8_t)(b >> (8 * k));
+    return r;
+#endif
+}
+
+__device__ __forceinline__ uint32_t swar_zigzag(uint32_t w) {
+    uint32_t mag = (w >> 1) & 0x7f7f7f7fu;
+    uint32_t sgn = w & 0x01010101u;
+    return mag ^ (sgn * 0xffu);  // per-byte signed i8 delta
+}
+
+// Fold Sum(d_j) and Sum(bidx*d_j) over stream bytes bidx in [lo, hi)
+// (delta j = bidx+1).  Interior dwords go through the dense dot4 loop;
+// the <=6 boundary bytes are folded by lane 0.
+__device__ void dense_region(const uint8_t *stream, int64_t lo, int64_t hi,
+                             int lane, int64_t *out_sum_d, int64_t *out_sum_jd) {
+    *out_sum_d = 0;
+    *out_sum_jd = 0;
+    if (hi <= lo) return;
+    // aligned interior: first dword-aligned byte >= lo, last < hi
+    uintptr_t s0 = (uintptr_t)stream;
+    int64_t alo = (int64_t)(((s0 + (uint64_t)lo + 3) & ~(uintptr_t)3) - s0);
+    int64_t ahi = (int64_t)(((s0 + (uint64_t)hi) & ~(uintptr_t)3) - s0);
+    int32_t acc_d = 0;
+    int64_t acc_jd = 0;   // 64-bit: boundary folds may exceed int32 ranges? no — keep 64 for safety on jd only at fold
+    int32_t acc_jd32 = 0;
+    if (ahi > alo) {
+        const uint32_t *p = (const uint32_t *)(s0 + (uint64_t)alo);
+        int64_t ndw = (ahi - alo) / 4;
+        int64_t q = lane;
+        // 4-deep manual batching: issue 4 lane-strided loads, then fold
+        for (; q + 192 < ndw; q += 256) {
+            uint32_t w0 = p[q], w1 = p[q + 64], w2 = p[q + 128], w3 = p[q + 192];
+            uint32_t s_0 = swar_zigzag(w0), s_1 = swar_zigzag(w1),
+                     s_2 = swar_zigzag(w2), s_3 = swar_zigzag(w3);
+            int32_t t0 = dot4_i8(s_0, 0x01010101u, 0);
+            int32_t t1 = dot4_i8(s_1, 0x01010101u, 0);
+            int32_t t2 = dot4_i8(s_2, 0x01010101u, 0);
+            int32_t t3 = dot4_i8(s_3, 0x01010101u, 0);
+            int32_t k0 = dot4_i8(s_0, 0x03020100u, 0);
+            int32_t k1 = dot4_i8(s_1, 0x03020100u, 0);
+            int32_t k2 = dot4_i8(s_2, 0x03020100u, 0);
+            int32_t k3 = dot4_i8(s_3, 0x03020100u, 0);
+            int32_t b0 = (int32_t)(alo + 4 * q);
+            acc_d += t0 + t1 + t2 + t3;
+            acc_jd32 += k0 + k1 + k2 + k3;
+            acc_jd32 += b0 * t0 + (b0 + 256) * t1 + (b0 + 512) * t2 +
+                        (b0 + 768) * t3;
+        }
+        for (; q < ndw; q += 64) {
+            uint32_t w0 = p[q];
+            uint32_t s_0 = swar_zigzag(w0);
+            int32_t t0 = dot4_i8(s_0, 0x01010101u, 0);
+            int32_t k0 = dot4_i8(s_0, 0x03020100u, 0);
+            acc_d += t0;
+            acc_jd32 += k0 + (int32_t)(alo + 4 * q) * t0;
+        }
+    }
+    acc_jd += acc_jd32;
+    // boundary bytes on lane 0: with an interior, the head [lo, alo) and
+    // tail [ahi, hi); without one, the whole [lo, hi)
+    if (lane == 0) {
+        int64_t h_lo = lo, h_hi, t_lo, t_hi;
+        if (ahi > alo) { h_hi = alo; t_lo = ahi; t_hi = hi; }
+        else { h_hi = hi; t_lo = 0; t_hi = 0; }
+        for (int64_t b = h_lo; b < h_hi; b++) {
+            uint32_t c = stream[b];
+            int32_t d = (int32_t)(c >> 1) ^ -(int32_t)(c & 1);
+            acc_d += d;
+            acc_jd += (int64_t)b * d;
+        }
+        for (int64_t b = t_lo; b < t_hi; b++) {
+            uint32_t c = stream[b];
+            int32_t d = (int32_t)(c >> 1) ^ -(int32_t)(c & 1);
+            acc_d += d;
+            acc_jd += (int64_t)b * d;
+        }
+    }
+    *out_sum_d = (int64_t)acc_d;
+    *out_sum_jd = acc_jd;
+}
+
+// Weighted fold over an all-1-byte delta stream:
+//   sum_{j=1..jend} d_j * w(j),  w(j) = nsel (j<=r0) | r1-j+1 (else)
+// via two dense regions (delta j lives at byte j-1):
+//   region1 bytes [0, min(r0,jend)):          contributes nsel * S1d
+//   region2 bytes [min(r0,jend), jend):       contributes (r1+1)*S2d - S2jd'
+// where S2jd' uses delta index j = bidx+1.
+__device__ uint64_t dense_delta_weighted(const uint8_t *stream,
+                                         int64_t n_deltas, int64_t r0,
+                                         int64_t r1, int lane) {
+    int64_t jend = r1 < n_deltas ? r1 : n_deltas;
+    if (jend < 1) return 0;
+    uint64_t nsel = (uint64_t)(r1 - r0 + 1);
+    int64_t b1 = r0 < jend ? r0 : jend;  // deltas 1..b1 have weight nsel
+    int64_t s1d, s1jd, s2d, s2jd;
+    dense_region(stream, 0, b1, lane, &s1d, &s1jd);
+    dense_region(stream, b1, jend, lane, &s2d, &s2jd);
+    // j = bidx + 1  =>  sum j*d = sum bidx*d + sum d
+    uint64_t sum_jd2 = (uint64_t)s2jd + (uint64_t)s2d;
+    uint64_t acc = nsel * (uint64_t)s1d;
+    acc += (uint64_t)(r1 + 1) * (uint64_t)s2d - sum_jd2;
+    return acc;
+}
+
 // Fast weighted delta fold: 256-byte windows (one dword per lane) with an
 // all-1-byte fast case — the dominant shape for delta-encoded telemetry
 // (zigzag deltas < 64 encode to one byte, int.go:84-88).  For a window
@@ -516,8 +635,16 @@ __global__ __launch_bounds__(256) void k_scan_agg(
             if (!(flags & KF_NEED_VALUES)) {
                 uint64_t acc;
                 if (!dod) {
-                    acc = fold_delta_weighted_fast(fstream, n - 1, r0, r1,
-                                                   lane, derr, (uint64_t)bi);
+                    if (bd->field_len == (uint64_t)(n - 1)) {
+                        // stream length == delta count -> every varint is
+                        // one byte: dense dot4 fold
+                        acc = dense_delta_weighted(fstream, n - 1, r0, r1,
+                                                   lane);
+                    } else {
+                        acc = fold_delta_weighted_fast(fstream, n - 1, r0, r1,
+                                                       lane, derr,
+                                                       (uint64_t)bi);
+                    }
                     acc = wave_reduce_add(acc);
                     if (lane == 0) {
                         bsum = (uint64_t)first * nsel + acc;

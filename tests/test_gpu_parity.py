@@ -124,6 +124,36 @@ def test_i64_time_clamp():
         s.close()
 
 
+def test_dense_1byte_clamp_boundaries():
+    """All-1-byte delta streams (the dense dot4 path) with clamps landing
+    at every alignment of the dword regions."""
+    rng = random.Random(21)
+    b = PartBuilder()
+    for sid in range(6):
+        n = 5000
+        ts = [T0 + i * MS for i in range(n)]
+        vals = []
+        v = rng.randint(-10**9, 10**9)
+        for _ in range(n):
+            v += rng.randint(-3, 3)  # deltas always 1 byte
+            vals.append(v)
+        b.add_block_i64(sid + 1, ts, [1] * n, vals)
+    for lo_row, hi_row in [(0, 4999), (1, 4998), (2, 4997), (3, 4996),
+                           (4, 4995), (7, 4993), (63, 4930), (64, 4929),
+                           (255, 4700), (256, 4699), (257, 4698),
+                           (1000, 1000), (4998, 4999), (0, 0)]:
+        lo, hi = T0 + lo_row * MS, T0 + hi_row * MS
+        orc = oracle_scan(b, VT_INT64, min_ts=lo, max_ts=hi)[0]
+        s = Session(0)
+        s.upload_part(b)
+        s.configure(VT_INT64, [AGG_SUM, AGG_COUNT])
+        s.consume(min_ts=lo, max_ts=hi)
+        g = s.finalize()[0]
+        assert g.count == orc.count, (lo_row, hi_row)
+        assert g.sum_i == orc.sum_i, (lo_row, hi_row)
+        s.close()
+
+
 def test_f64_min_max_exact_sum_tolerance():
     b = PartBuilder()
     rng = random.Random(6)
