@@ -528,6 +528,20 @@ enum {
     KF_FLOAT = 2,        // float64 field: also accumulate mantissa sum as double
 };
 
+__device__ __forceinline__ void flush_partial(bydb_partial *partials,
+                                              int64_t group, uint64_t wsum,
+                                              uint64_t wcnt, int64_t wmin,
+                                              int64_t wmax, double wsumf,
+                                              int lane) {
+    if (group < 0 || lane != 0) return;
+    bydb_partial *p = &partials[group];
+    if (wsum) atomicAdd((unsigned long long *)&p->sum_i, (unsigned long long)wsum);
+    if (wcnt) atomicAdd((unsigned long long *)&p->count, (unsigned long long)wcnt);
+    if (wmin != INT64_MAX) atomicMin((long long *)&p->min_i, (long long)wmin);
+    if (wmax != INT64_MIN) atomicMax((long long *)&p->max_i, (long long)wmax);
+    if (wsumf != 0.0) atomicAdd(&p->sum_f, wsumf);
+}
+
 __global__ __launch_bounds__(256) void k_scan_agg(
     const uint8_t *__restrict__ payload, const bydb_block_desc *__restrict__ blocks,
     int64_t n_blocks, int64_t min_ts, int64_t max_ts, int flags,
@@ -536,6 +550,12 @@ __global__ __launch_bounds__(256) void k_scan_agg(
     const int wave_in_block = threadIdx.x >> 6;
     int64_t wave_id = (int64_t)blockIdx.x * (blockDim.x >> 6) + wave_in_block;
     int64_t n_waves = (int64_t)gridDim.x * (blockDim.x >> 6);
+
+    // per-wave partial (flushed on group change / at the end)
+    int64_t cur_group = -1;
+    uint64_t wsum = 0, wcnt = 0;
+    int64_t wmin = INT64_MAX, wmax = INT64_MIN;
+    double wsumf = 0.0;
 
     for (int64_t bi = wave_id; bi < n_blocks; bi += n_waves) {
         // wave-uniform descriptor loads
@@ -702,20 +722,27 @@ __global__ __launch_bounds__(256) void k_scan_agg(
         // rows 0 (+1 for dod) fast path bookkeeping: the weighted folds
         // above already include first (and d1) in their closed forms.
 
-        // ---- accumulate into the group partial ----
-        if (lane == 0) {
-            bydb_partial *p = &partials[bd->group_code];
-            atomicAdd((unsigned long long *)&p->sum_i, (unsigned long long)bsum);
-            atomicAdd((unsigned long long *)&p->count, (unsigned long long)nsel);
-            if (have_minmax) {
-                atomicMin((long long *)&p->min_i, (long long)bmin);
-                atomicMax((long long *)&p->max_i, (long long)bmax);
-            }
-            if (flags & KF_FLOAT) {
-                atomicAdd(&p->sum_f, (double)(int64_t)bsum);
-            }
+        // ---- accumulate into the per-wave register partial ----
+        // One atomic set per BLOCK on a single group serialises on one
+        // cacheline (~88 atomics/us on one word) and dominates the whole
+        // scan; instead fold block totals into per-wave registers and
+        // flush only when the group changes or the wave is done.
+        bsum = (uint64_t)__shfl((long long)bsum, 0);  // lane0 holds closed forms
+        if ((int64_t)bd->group_code != cur_group) {
+            flush_partial(partials, cur_group, wsum, wcnt, wmin, wmax, wsumf,
+                          lane);
+            cur_group = (int64_t)bd->group_code;
+            wsum = 0; wcnt = 0; wmin = INT64_MAX; wmax = INT64_MIN; wsumf = 0;
         }
+        wsum += bsum;
+        wcnt += nsel;
+        if (have_minmax) {
+            wmin = bmin < wmin ? bmin : wmin;
+            wmax = bmax > wmax ? bmax : wmax;
+        }
+        if (flags & KF_FLOAT) wsumf += (double)(int64_t)bsum;
     }
+    flush_partial(partials, cur_group, wsum, wcnt, wmin, wmax, wsumf, lane);
 }
 
 // init kernel: set partials to the fold identity (Map.Reset, function.go)
