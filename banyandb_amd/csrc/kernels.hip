@@ -405,6 +405,195 @@ __device__ uint64_t fold_dod_weighted(const uint8_t *stream, int64_t n_deltas,
     return acc;
 }
 
+// ---------------- tag predicate (dictionary codes) ----------------
+// The reference evaluates tag-equality predicates on decoded tag values
+// per row (vec scan/filter; tags are dictionary-encoded per block,
+// column.go:266-278, dictionary.go:79-115).  Here a resolve prepass maps
+// the predicate value to a per-block CODE mask by comparing it against the
+// block's dictionary values, and the scan walks the block's bit-packed RLE
+// code runs (dictionary.go:158-260) to test rows — codes only, the value
+// bytes are never expanded per row.
+
+struct PredBlock {
+    uint64_t mask[4];      // bit c: dict code c equals the predicate value
+    uint64_t rle_bit_off;  // absolute bit offset of packed RLE entries
+    uint32_t nentries;     // packed entry count (2 per run: value, count)
+    uint8_t width;         // bits per packed entry
+    uint8_t active;        // block has a dictionary tag column
+    uint8_t err;           // unparseable on device (zstd/plain) -> error
+    uint8_t uniform;       // single run covers the whole block
+};
+
+// MSB-first bit read at an arbitrary bit offset (reader.go:39-79 order);
+// n <= 32, so 8 gathered bytes always cover shift+n.
+__device__ __forceinline__ uint64_t rd_bits_be(const uint8_t *p,
+                                               uint64_t bitpos, uint32_t n) {
+    uint64_t byte = bitpos >> 3;
+    uint32_t sh = (uint32_t)(bitpos & 7);
+    uint64_t acc = 0;
+#pragma unroll
+    for (int i = 0; i < 8; i++) acc = (acc << 8) | p[byte + (uint64_t)i];
+    return (acc >> (64 - sh - n)) & ((1ull << n) - 1);
+}
+
+struct PredWalk {
+    const uint8_t *payload;
+    uint64_t bit0;
+    uint32_t nentries, width;
+    uint64_t mask[4];
+    uint32_t entry;            // next entry index (2 per run)
+    int64_t run_lo, run_hi;    // rows [run_lo, run_hi) of the current run
+    bool run_match;
+};
+
+__device__ __forceinline__ void pred_init(PredWalk *pw, const uint8_t *payload,
+                                          const PredBlock *pb) {
+    pw->payload = payload;
+    pw->bit0 = pb->rle_bit_off;
+    pw->nentries = pb->nentries;
+    pw->width = pb->width;
+#pragma unroll
+    for (int i = 0; i < 4; i++) pw->mask[i] = pb->mask[i];
+    pw->entry = 0;
+    pw->run_lo = 0;
+    pw->run_hi = 0;
+    pw->run_match = false;
+}
+
+__device__ __forceinline__ void pred_advance(PredWalk *pw) {
+    if (pw->entry + 1 >= pw->nentries) {   // RLE exhausted: no-match tail
+        pw->run_lo = pw->run_hi;
+        pw->run_hi = INT64_MAX;
+        pw->run_match = false;
+        pw->entry = pw->nentries + 2;
+        return;
+    }
+    uint64_t code = rd_bits_be(pw->payload, pw->bit0 + (uint64_t)pw->entry * pw->width,
+                               pw->width);
+    uint64_t cnt = rd_bits_be(pw->payload,
+                              pw->bit0 + (uint64_t)(pw->entry + 1) * pw->width,
+                              pw->width);
+    pw->entry += 2;
+    pw->run_lo = pw->run_hi;
+    pw->run_hi += (int64_t)cnt;
+    pw->run_match = (pw->mask[(code >> 6) & 3] >> (code & 63)) & 1;
+}
+
+// Assign `match` for each lane's row (rows ascend across calls).  `need`
+// lanes get the run covering their row; the walker only moves forward.
+__device__ __forceinline__ bool pred_match_rows(PredWalk *pw, int64_t row,
+                                                bool need) {
+    bool match = false;
+    while (true) {
+        bool mine = need && row >= pw->run_lo && row < pw->run_hi;
+        if (mine) { match = pw->run_match; need = false; }
+        if (__all(!need)) break;
+        pred_advance(pw);
+    }
+    return match;
+}
+
+// Resolve prepass: one thread per block parses the dictionary header and
+// builds the code mask.  Only plain (<128 B) compress_block sections are
+// parseable on device (bytes.go:291-303); zstd-compressed dictionaries are
+// flagged and surfaced as a device error if a predicate needs them.
+__global__ void k_resolve_pred(const uint8_t *__restrict__ payload,
+                               const bydb_block_desc *__restrict__ blocks,
+                               int64_t n_blocks, const uint8_t *__restrict__ pred,
+                               uint64_t pred_len, PredBlock *__restrict__ out) {
+    int64_t bi = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (bi >= n_blocks) return;
+    const bydb_block_desc *bd = &blocks[bi];
+    PredBlock pb;
+    for (int i = 0; i < 4; i++) pb.mask[i] = 0;
+    pb.rle_bit_off = 0; pb.nentries = 0; pb.width = 0;
+    pb.active = 0; pb.err = 0; pb.uniform = 0;
+    if (bd->tag_len == 0) { out[bi] = pb; return; }
+    const uint8_t *p = payload + bd->tag_off;
+    const uint8_t *end = p + bd->tag_len;
+    if (*p != BYDB_ENC_DICTIONARY) { pb.err = 1; out[bi] = pb; return; }
+    p++;
+    pb.active = 1;
+    // varuint count (int.go:152-199)
+    uint64_t count = 0;
+    unsigned sh = 0;
+    while (p < end) {
+        uint8_t c = *p++;
+        count |= (uint64_t)(c & 0x7f) << sh;
+        if (c < 0x80) break;
+        sh += 7;
+    }
+    // lengths block: compress_block(u64list) — plain only
+    if (p >= end || *p != 0) { pb.err = 1; out[bi] = pb; return; }
+    p++;
+    uint64_t ll = *p++;
+    const uint8_t *lens_blk = p;
+    p += ll;
+    // values payload block: plain only
+    if (p >= end || *p != 0) { pb.err = 1; out[bi] = pb; return; }
+    p++;
+    uint64_t vl = *p++;
+    const uint8_t *vals = p;
+    p += vl;
+    // parse width-typed lengths (bytes.go:205-235)
+    uint8_t wt = lens_blk[0];
+    uint32_t wbytes = wt == 0 ? 1 : wt == 1 ? 2 : wt == 2 ? 4 : 8;
+    const uint8_t *lp = lens_blk + 1;
+    uint64_t voff = 0;
+    for (uint64_t v = 0; v < count && v < 256; v++) {
+        uint64_t alen = 0;
+        for (uint32_t b = 0; b < wbytes; b++) alen = (alen << 8) | lp[v * wbytes + b];
+        if (alen > 0) {
+            uint64_t vlen = alen - 1;
+            if (vlen == pred_len) {
+                bool eq = true;
+                for (uint64_t k = 0; k < vlen; k++)
+                    if (vals[voff + k] != pred[k]) { eq = false; break; }
+                if (eq) pb.mask[(v >> 6) & 3] |= 1ull << (v & 63);
+            }
+            voff += vlen;
+        }
+    }
+    // bit-packed RLE: [32b entry count][8b width][entries...] MSB-first
+    uint64_t bit0 = ((uint64_t)(p - payload)) * 8;
+    uint32_t nentries = (uint32_t)rd_bits_be(payload, bit0, 32);
+    uint32_t width = nentries ? (uint32_t)rd_bits_be(payload, bit0 + 32, 8) : 0;
+    pb.nentries = nentries;
+    pb.width = (uint8_t)width;
+    pb.rle_bit_off = bit0 + 40;
+    if (nentries == 2) {
+        uint64_t cnt = rd_bits_be(payload, pb.rle_bit_off + width, width);
+        if (cnt >= bd->count) pb.uniform = 1;
+    }
+    out[bi] = pb;
+}
+
+// Fold arithmetic-progression blocks (Const dd=0 / DeltaConst) under a
+// per-row predicate: walk rows 64 at a time through the RLE runs.
+__device__ void fold_arith_pred(int64_t first, int64_t dd, int64_t r0,
+                                int64_t r1, int lane, PredWalk *pw,
+                                uint64_t *out_sum, uint64_t *out_cnt,
+                                int64_t *out_mn, int64_t *out_mx) {
+    uint64_t lsum = 0, lcnt = 0;
+    int64_t lmn = INT64_MAX, lmx = INT64_MIN;
+    for (int64_t base = r0; base <= r1; base += 64) {
+        int64_t row = base + lane;
+        bool need = row <= r1;
+        bool match = pred_match_rows(pw, row, need);
+        if (need && match) {
+            int64_t v = (int64_t)((uint64_t)first + (uint64_t)row * (uint64_t)dd);
+            lsum += (uint64_t)v;
+            lcnt++;
+            lmn = v < lmn ? v : lmn;
+            lmx = v > lmx ? v : lmx;
+        }
+    }
+    *out_sum = wave_reduce_add(lsum);
+    *out_cnt = wave_reduce_add(lcnt);
+    *out_mn = wave_reduce_min(lmn);
+    *out_mx = wave_reduce_max(lmx);
+}
+
 // Full value reconstruction over a delta or delta-of-delta stream with a
 // per-value callback encoded as flags (fold min/max/sum/sumf, or count
 // ts-range bounds).  dod=false: values are rows j=1..n-1 (row 0 = first,
@@ -422,7 +611,8 @@ __device__ void scan_stream(const uint8_t *stream, int64_t n_deltas, bool dod,
                             int64_t first_plus /* v at row (dod?1:0) */,
                             int64_t d1_init, int64_t r0, int64_t r1,
                             int64_t lo_bound, int64_t hi_bound, int lane,
-                            ScanFold *f, DevErr *derr, uint64_t bi) {
+                            ScanFold *f, DevErr *derr, uint64_t bi,
+                            PredWalk *pw) {
     uint64_t pos = 0;
     int64_t j = dod ? 2 : 1;
     int64_t jmax = n_deltas;            // always scan the whole stream
@@ -497,16 +687,20 @@ __device__ void scan_stream(const uint8_t *stream, int64_t n_deltas, bool dod,
             tot = (uint64_t)__shfl((long long)s, last_lane);
             v_carry += tot;
         }
-        if (is_term) {
+        bool in_sel = is_term && myj >= r0 && myj <= r1;
+        if (pw) in_sel = pred_match_rows(pw, myj, in_sel) && in_sel;
+        {
             int64_t sv = (int64_t)val;
-            if (myj >= r0 && myj <= r1) {
+            if (in_sel) {
                 l_sum += val;
                 l_nsel++;
                 l_mn = sv < l_mn ? sv : l_mn;
                 l_mx = sv > l_mx ? sv : l_mx;
             }
-            if (sv < lo_bound) l_nlo++;
-            if (sv > hi_bound) l_nhi++;
+            if (is_term) {
+                if (sv < lo_bound) l_nlo++;
+                if (sv > hi_bound) l_nhi++;
+            }
         }
         int nterm = __popcll(emask);
         if (j + nterm > jmax) break;
@@ -545,7 +739,8 @@ __device__ __forceinline__ void flush_partial(bydb_partial *partials,
 __global__ __launch_bounds__(256) void k_scan_agg(
     const uint8_t *__restrict__ payload, const bydb_block_desc *__restrict__ blocks,
     int64_t n_blocks, int64_t min_ts, int64_t max_ts, int flags,
-    bydb_partial *__restrict__ partials, DevErr *derr) {
+    const PredBlock *__restrict__ preds, bydb_partial *__restrict__ partials,
+    DevErr *derr) {
     const int lane = threadIdx.x & 63;
     const int wave_in_block = threadIdx.x >> 6;
     int64_t wave_id = (int64_t)blockIdx.x * (blockDim.x >> 6) + wave_in_block;
@@ -598,7 +793,8 @@ __global__ __launch_bounds__(256) void k_scan_agg(
                 ScanFold tsf;
                 int64_t row1 = (int64_t)((uint64_t)ts_min + (uint64_t)d1);
                 scan_stream(s, n - 1, dod, dod ? row1 : vfirst, d1, 1, 0,
-                            min_ts, max_ts, lane, &tsf, derr, (uint64_t)bi);
+                            min_ts, max_ts, lane, &tsf, derr, (uint64_t)bi,
+                            nullptr);
                 uint64_t nlo = wave_reduce_add(tsf.n_lo);
                 uint64_t nhi = wave_reduce_add(tsf.n_hi);
                 // rows 0 (and 1 for dod) were not in the stream
@@ -615,6 +811,22 @@ __global__ __launch_bounds__(256) void k_scan_agg(
         }
         const uint64_t nsel = (uint64_t)(r1 - r0 + 1);
 
+        // ---- per-row tag predicate (dictionary codes) ----
+        PredWalk pw;
+        bool pred_on = preds != nullptr;
+        if (pred_on) {
+            PredBlock pb = preds[bi];
+            if (pb.err) { dev_set_err(derr, DERR_BAD_ENC, (uint64_t)bi); continue; }
+            if (!pb.active) continue;  // no tag -> nil -> equality never holds
+            pred_init(&pw, payload, &pb);
+            if (pb.uniform) {
+                // one run covers the block: predicate is block-uniform
+                pred_advance(&pw);
+                if (!pw.run_match) continue;
+                pred_on = false;  // fold as if unpredicated
+            }
+        }
+
         // ---- field fold ----
         const uint8_t fenc = bd->field_enc;
         const int64_t first = bd->field_first;
@@ -623,7 +835,58 @@ __global__ __launch_bounds__(256) void k_scan_agg(
         int64_t bmin = INT64_MAX, bmax = INT64_MIN;
         bool have_minmax = false;
 
-        if (fenc == BYDB_ENC_CONST) {
+        uint64_t nsel_eff = nsel;
+        if (pred_on && (fenc == BYDB_ENC_CONST || fenc == BYDB_ENC_DELTA_CONST)) {
+            int64_t dd = 0;
+            if (fenc == BYDB_ENC_DELTA_CONST) {
+                int vl;
+                dd = decode_one_varint(fstream, &vl);
+            }
+            uint64_t psum, pcnt;
+            fold_arith_pred(first, dd, r0, r1, lane, &pw, &psum, &pcnt, &bmin,
+                            &bmax);
+            bsum = lane == 0 ? psum : 0;
+            nsel_eff = pcnt;
+            have_minmax = (flags & KF_NEED_VALUES) && pcnt > 0;
+        } else if (pred_on &&
+                   (fenc == BYDB_ENC_DELTA || fenc == BYDB_ENC_DELTA_OF_DELTA)) {
+            bool dod = fenc == BYDB_ENC_DELTA_OF_DELTA;
+            const uint8_t *s = fstream;
+            int64_t d1 = 0;
+            if (dod) {
+                int vl;
+                d1 = decode_one_varint(s, &vl);
+                s += vl;
+            }
+            int64_t row1 = (int64_t)((uint64_t)first + (uint64_t)d1);
+            // rows outside the stream first (walker rows must ascend)
+            uint64_t lsum = 0, lcnt = 0;
+            int64_t lmn = INT64_MAX, lmx = INT64_MIN;
+            bool need0 = lane == 0 && r0 <= 0 && 0 <= r1;
+            bool m0 = pred_match_rows(&pw, 0, need0);
+            if (need0 && m0) { lsum += (uint64_t)first; lcnt++;
+                lmn = first; lmx = first; }
+            if (dod) {
+                bool need1 = lane == 0 && r0 <= 1 && 1 <= r1;
+                bool m1 = pred_match_rows(&pw, 1, need1);
+                if (need1 && m1) { lsum += (uint64_t)row1; lcnt++;
+                    lmn = row1 < lmn ? row1 : lmn; lmx = row1 > lmx ? row1 : lmx; }
+            }
+            ScanFold ff;
+            scan_stream(s, n - 1, dod, dod ? row1 : first, d1, r0, r1,
+                        INT64_MAX, INT64_MIN, lane, &ff, derr, (uint64_t)bi,
+                        &pw);
+            lsum += ff.sum;
+            lcnt += ff.nsel;
+            lmn = ff.mn < lmn ? ff.mn : lmn;
+            lmx = ff.mx > lmx ? ff.mx : lmx;
+            bsum = wave_reduce_add(lsum);
+            nsel_eff = wave_reduce_add(lcnt);
+            bmin = wave_reduce_min(lmn);
+            bmax = wave_reduce_max(lmx);
+            have_minmax = nsel_eff > 0;
+            bsum = lane == 0 ? bsum : 0;
+        } else if (fenc == BYDB_ENC_CONST) {
             if (lane == 0) bsum = (uint64_t)first * nsel;
             bmin = bmax = first;
             have_minmax = true;
@@ -692,7 +955,8 @@ __global__ __launch_bounds__(256) void k_scan_agg(
                 int64_t row1 = (int64_t)((uint64_t)first + (uint64_t)d1);
                 ScanFold ff;
                 scan_stream(s, n - 1, dod, dod ? row1 : first, d1, r0, r1,
-                            INT64_MAX, INT64_MIN, lane, &ff, derr, (uint64_t)bi);
+                            INT64_MAX, INT64_MIN, lane, &ff, derr, (uint64_t)bi,
+                            nullptr);
                 uint64_t lsum = ff.sum;
                 int64_t lmn = ff.mn, lmx = ff.mx;
                 // rows outside the stream: row 0 (value=first) and, for
@@ -735,7 +999,7 @@ __global__ __launch_bounds__(256) void k_scan_agg(
             wsum = 0; wcnt = 0; wmin = INT64_MAX; wmax = INT64_MIN; wsumf = 0;
         }
         wsum += bsum;
-        wcnt += nsel;
+        wcnt += nsel_eff;
         if (have_minmax) {
             wmin = bmin < wmin ? bmin : wmin;
             wmax = bmax > wmax ? bmax : wmax;
@@ -779,6 +1043,10 @@ struct bydb_session {
     bydb_partial *d_acc = nullptr;        // active accumulation target
     uint64_t partials_cap = 0;
     DevErr *d_err = nullptr;
+    PredBlock *d_preds = nullptr;
+    int64_t preds_cap = 0;
+    uint8_t *d_pred_bytes = nullptr;
+    uint64_t pred_bytes_cap = 0;
     float last_ms = 0.0f;
     bool consumed = false;
     int16_t float_exp = 0;  // shared decimal exponent for float64 restore
@@ -825,6 +1093,8 @@ extern "C" void bydb_session_destroy(bydb_session *s) {
     if (s->d_payload) hipFree(s->d_payload);
     if (s->d_blocks) hipFree(s->d_blocks);
     if (s->d_partials) hipFree(s->d_partials);
+    if (s->d_preds) hipFree(s->d_preds);
+    if (s->d_pred_bytes) hipFree(s->d_pred_bytes);
     if (s->d_err) hipFree(s->d_err);
     if (s->ev_start) hipEventDestroy(s->ev_start);
     if (s->ev_stop) hipEventDestroy(s->ev_stop);
@@ -928,7 +1198,28 @@ extern "C" int bydb_consume(bydb_session *s, int64_t min_ts, int64_t max_ts,
     HIP_TRY(s, hipSetDevice(s->device));
     if (!s->d_acc) { s->err = "configure first"; return BYDB_ERR_STATE; }
     if (s->n_blocks == 0) { s->err = "no part resident"; return BYDB_ERR_STATE; }
-    if (pred_len > 0) { s->err = "predicate not yet supported"; return BYDB_ERR_BAD_ARG; }
+    PredBlock *preds = nullptr;
+    if (pred_len > 0) {
+        if (s->preds_cap < s->n_blocks) {
+            if (s->d_preds) hipFree(s->d_preds);
+            HIP_TRY(s, hipMalloc(&s->d_preds, sizeof(PredBlock) * (size_t)s->n_blocks));
+            s->preds_cap = s->n_blocks;
+        }
+        if (s->pred_bytes_cap < pred_len) {
+            if (s->d_pred_bytes) hipFree(s->d_pred_bytes);
+            HIP_TRY(s, hipMalloc(&s->d_pred_bytes, pred_len));
+            s->pred_bytes_cap = pred_len;
+        }
+        HIP_TRY(s, hipMemcpyAsync(s->d_pred_bytes, pred, pred_len,
+                                  hipMemcpyHostToDevice, s->stream));
+        int rthreads = 256;
+        int rblocks = (int)((s->n_blocks + rthreads - 1) / rthreads);
+        hipLaunchKernelGGL(k_resolve_pred, dim3(rblocks), dim3(rthreads), 0,
+                           s->stream, s->d_payload, s->d_blocks, s->n_blocks,
+                           s->d_pred_bytes, pred_len, s->d_preds);
+        HIP_TRY(s, hipGetLastError());
+        preds = s->d_preds;
+    }
     int flags = 0;
     if (s->func_mask & ((1u << BYDB_AGG_MIN) | (1u << BYDB_AGG_MAX)))
         flags |= KF_NEED_VALUES;
@@ -941,7 +1232,7 @@ extern "C" int bydb_consume(bydb_session *s, int64_t min_ts, int64_t max_ts,
     HIP_TRY(s, hipEventRecord(s->ev_start, s->stream));
     hipLaunchKernelGGL(k_scan_agg, dim3(grid), dim3(threads), 0, s->stream,
                        s->d_payload, s->d_blocks, s->n_blocks, min_ts, max_ts,
-                       flags, s->d_acc, s->d_err);
+                       flags, preds, s->d_acc, s->d_err);
     HIP_TRY(s, hipGetLastError());
     HIP_TRY(s, hipEventRecord(s->ev_stop, s->stream));
     s->consumed = true;

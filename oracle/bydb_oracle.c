@@ -1092,7 +1092,9 @@ static int scan_block(const uint8_t *payload, const bo_block_desc *b,
     int64_t r0, r1;
     if (!bo_find_range(ts_buf, n, min_ts, max_ts, &r0, &r1)) return BO_OK;
     /* predicate rows (tag equality on decoded tag values —
-     * the row path compares raw tag bytes after decode) */
+     * the row path compares raw tag bytes after decode).  A block with no
+     * tag column has nil tags everywhere: equality never holds. */
+    if (pred_len > 0 && b->tag_len == 0) return BO_OK;
     int have_pred = pred_len > 0 && b->tag_len > 0;
     size_t tagdata_len = 0;
     if (have_pred) {
