@@ -112,10 +112,11 @@ def main():
         sess.reset()
         sess.consume(min_ts=t0_r, max_ts=t0_r + n_dp * STRIDE)
         if world > 1:
-            # sync the session stream, then RCCL-merge the partials:
-            # sums+counts reduce with SUM (AggModeReduce Combine semantics)
+            # sync the session stream, then RCCL-merge the partials over
+            # xGMI (AggModeReduce Combine semantics)
             parts = sess.finalize_partials()
-            dist.all_reduce(part_t[0:2], op=dist.ReduceOp.SUM)
+            from banyandb_amd.distributed import allreduce_partials
+            allreduce_partials(dist, part_t, n_groups)
             return parts
         return sess.finalize_partials()
 
@@ -196,6 +197,13 @@ def main():
                 "unit": "GB/s",
                 "frac": achieved_gbs / HBM_PEAK_GBS,
                 "traffic": traffic,
+                # achieved/frac use SURVEY 8d's ALGORITHMIC 16 B/dp; the
+                # kernel reads the ~1.05 B/dp ENCODED stream, so frac > 1
+                # is possible.  traffic_* is what actually moved (PMC).
+                "traffic_gbs": (traffic / (avg_launch_ms / 1e3) / 1e9)
+                               if traffic else None,
+                "traffic_frac": (traffic / (avg_launch_ms / 1e3) / 1e9
+                                 / HBM_PEAK_GBS) if traffic else None,
             },
             "cpu_baseline": cpu_baseline,
         }
