@@ -41,6 +41,10 @@ class BlockDesc(C.Structure):
         ("field_len", C.c_uint64),
         ("tag_off", C.c_uint64),
         ("tag_len", C.c_uint64),
+        ("tag2_off", C.c_uint64),
+        ("tag2_len", C.c_uint64),
+        ("tag3_off", C.c_uint64),
+        ("tag3_len", C.c_uint64),
         ("group_code", C.c_uint32),
         ("_pad2", C.c_uint32),
     ]
@@ -117,6 +121,10 @@ def _load():
     lib.bydb_reset.argtypes = [C.c_void_p]
     lib.bydb_consume.restype = C.c_int
     lib.bydb_consume.argtypes = [C.c_void_p, C.c_int64, C.c_int64, u8p, C.c_uint64]
+    lib.bydb_consume_multi.restype = C.c_int
+    lib.bydb_consume_multi.argtypes = [C.c_void_p, C.c_int64, C.c_int64,
+                                       C.POINTER(u8p), C.POINTER(C.c_uint64),
+                                       C.c_int]
     lib.bydb_finalize.restype = C.c_int
     lib.bydb_finalize.argtypes = [C.c_void_p, C.POINTER(Result), C.c_int64]
     lib.bydb_finalize_partials.restype = C.c_int
@@ -320,7 +328,16 @@ class Session:
     def reset(self):
         self._ck(_lib.bydb_reset(self._h))
 
-    def consume(self, min_ts=INT64_MIN, max_ts=INT64_MAX, pred=b""):
+    def consume(self, min_ts=INT64_MIN, max_ts=INT64_MAX, pred=b"", preds=None):
+        if preds is not None:
+            bufs = [(C.c_uint8 * max(len(p), 1)).from_buffer_copy(p or b"\0")
+                    for p in preds]
+            arr = (C.POINTER(C.c_uint8) * len(preds))(
+                *[C.cast(b, C.POINTER(C.c_uint8)) for b in bufs])
+            lens = (C.c_uint64 * len(preds))(*[len(p) for p in preds])
+            self._ck(_lib.bydb_consume_multi(self._h, min_ts, max_ts, arr,
+                                             lens, len(preds)))
+            return
         buf = (C.c_uint8 * max(len(pred), 1)).from_buffer_copy(pred or b"\0")
         self._ck(_lib.bydb_consume(self._h, min_ts, max_ts, buf, len(pred)))
 

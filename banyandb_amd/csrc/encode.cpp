@@ -502,6 +502,11 @@ extern "C" int bydb_part_builder_set_block_tag(bydb_part_builder *b,
         b->err = "tag row count mismatch";
         return BYDB_ERR_BAD_ARG;
     }
+    int slot = d.tag_len == 0 ? 0 : d.tag2_len == 0 ? 1 : d.tag3_len == 0 ? 2 : 3;
+    if (slot == 3) {
+        b->err = "at most 3 tag columns per block";
+        return BYDB_ERR_BAD_ARG;
+    }
     size_t start = b->payload.size();
     // column.encodeDefault (column.go:266-278): try dictionary, else Plain
     b->payload.push_back(BYDB_ENC_DICTIONARY);
@@ -514,8 +519,11 @@ extern "C" int bydb_part_builder_set_block_tag(bydb_part_builder *b,
             return BYDB_ERR_BAD_DATA;
         }
     }
-    d.tag_off = b->base_off + start;
-    d.tag_len = b->payload.size() - start;
+    uint64_t off = b->base_off + start;
+    uint64_t len = b->payload.size() - start;
+    if (slot == 0) { d.tag_off = off; d.tag_len = len; }
+    else if (slot == 1) { d.tag2_off = off; d.tag2_len = len; }
+    else { d.tag3_off = off; d.tag3_len = len; }
     return BYDB_OK;
 }
 

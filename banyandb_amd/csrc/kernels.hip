@@ -500,17 +500,20 @@ __device__ __forceinline__ bool pred_match_rows(PredWalk *pw, int64_t row,
 __global__ void k_resolve_pred(const uint8_t *__restrict__ payload,
                                const bydb_block_desc *__restrict__ blocks,
                                int64_t n_blocks, const uint8_t *__restrict__ pred,
-                               uint64_t pred_len, PredBlock *__restrict__ out) {
+                               uint64_t pred_len, int slot,
+                               PredBlock *__restrict__ out) {
     int64_t bi = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (bi >= n_blocks) return;
     const bydb_block_desc *bd = &blocks[bi];
+    uint64_t toff = slot == 0 ? bd->tag_off : slot == 1 ? bd->tag2_off : bd->tag3_off;
+    uint64_t tlen = slot == 0 ? bd->tag_len : slot == 1 ? bd->tag2_len : bd->tag3_len;
     PredBlock pb;
     for (int i = 0; i < 4; i++) pb.mask[i] = 0;
     pb.rle_bit_off = 0; pb.nentries = 0; pb.width = 0;
     pb.active = 0; pb.err = 0; pb.uniform = 0;
-    if (bd->tag_len == 0) { out[bi] = pb; return; }
-    const uint8_t *p = payload + bd->tag_off;
-    const uint8_t *end = p + bd->tag_len;
+    if (tlen == 0) { out[bi] = pb; return; }
+    const uint8_t *p = payload + toff;
+    const uint8_t *end = p + tlen;
     if (*p != BYDB_ENC_DICTIONARY) { pb.err = 1; out[bi] = pb; return; }
     p++;
     pb.active = 1;
@@ -571,7 +574,8 @@ __global__ void k_resolve_pred(const uint8_t *__restrict__ payload,
 // Fold arithmetic-progression blocks (Const dd=0 / DeltaConst) under a
 // per-row predicate: walk rows 64 at a time through the RLE runs.
 __device__ void fold_arith_pred(int64_t first, int64_t dd, int64_t r0,
-                                int64_t r1, int lane, PredWalk *pw,
+                                int64_t r1, int lane, PredWalk *pw0,
+                                PredWalk *pw1, PredWalk *pw2,
                                 uint64_t *out_sum, uint64_t *out_cnt,
                                 int64_t *out_mn, int64_t *out_mx) {
     uint64_t lsum = 0, lcnt = 0;
@@ -579,7 +583,10 @@ __device__ void fold_arith_pred(int64_t first, int64_t dd, int64_t r0,
     for (int64_t base = r0; base <= r1; base += 64) {
         int64_t row = base + lane;
         bool need = row <= r1;
-        bool match = pred_match_rows(pw, row, need);
+        bool match = true;
+        if (pw0) match = pred_match_rows(pw0, row, need && match) && match;
+        if (pw1) match = pred_match_rows(pw1, row, need && match) && match;
+        if (pw2) match = pred_match_rows(pw2, row, need && match) && match;
         if (need && match) {
             int64_t v = (int64_t)((uint64_t)first + (uint64_t)row * (uint64_t)dd);
             lsum += (uint64_t)v;
@@ -612,7 +619,7 @@ __device__ void scan_stream(const uint8_t *stream, int64_t n_deltas, bool dod,
                             int64_t d1_init, int64_t r0, int64_t r1,
                             int64_t lo_bound, int64_t hi_bound, int lane,
                             ScanFold *f, DevErr *derr, uint64_t bi,
-                            PredWalk *pw) {
+                            PredWalk *pw0, PredWalk *pw1, PredWalk *pw2) {
     uint64_t pos = 0;
     int64_t j = dod ? 2 : 1;
     int64_t jmax = n_deltas;            // always scan the whole stream
@@ -688,7 +695,9 @@ __device__ void scan_stream(const uint8_t *stream, int64_t n_deltas, bool dod,
             v_carry += tot;
         }
         bool in_sel = is_term && myj >= r0 && myj <= r1;
-        if (pw) in_sel = pred_match_rows(pw, myj, in_sel) && in_sel;
+        if (pw0) in_sel = pred_match_rows(pw0, myj, in_sel) && in_sel;
+        if (pw1) in_sel = pred_match_rows(pw1, myj, in_sel) && in_sel;
+        if (pw2) in_sel = pred_match_rows(pw2, myj, in_sel) && in_sel;
         {
             int64_t sv = (int64_t)val;
             if (in_sel) {
@@ -739,8 +748,8 @@ __device__ __forceinline__ void flush_partial(bydb_partial *partials,
 __global__ __launch_bounds__(256) void k_scan_agg(
     const uint8_t *__restrict__ payload, const bydb_block_desc *__restrict__ blocks,
     int64_t n_blocks, int64_t min_ts, int64_t max_ts, int flags,
-    const PredBlock *__restrict__ preds, bydb_partial *__restrict__ partials,
-    DevErr *derr) {
+    const PredBlock *__restrict__ preds, int n_preds,
+    bydb_partial *__restrict__ partials, DevErr *derr) {
     const int lane = threadIdx.x & 63;
     const int wave_in_block = threadIdx.x >> 6;
     int64_t wave_id = (int64_t)blockIdx.x * (blockDim.x >> 6) + wave_in_block;
@@ -794,7 +803,7 @@ __global__ __launch_bounds__(256) void k_scan_agg(
                 int64_t row1 = (int64_t)((uint64_t)ts_min + (uint64_t)d1);
                 scan_stream(s, n - 1, dod, dod ? row1 : vfirst, d1, 1, 0,
                             min_ts, max_ts, lane, &tsf, derr, (uint64_t)bi,
-                            nullptr);
+                            nullptr, nullptr, nullptr);
                 uint64_t nlo = wave_reduce_add(tsf.n_lo);
                 uint64_t nhi = wave_reduce_add(tsf.n_hi);
                 // rows 0 (and 1 for dod) were not in the stream
@@ -811,21 +820,37 @@ __global__ __launch_bounds__(256) void k_scan_agg(
         }
         const uint64_t nsel = (uint64_t)(r1 - r0 + 1);
 
-        // ---- per-row tag predicate (dictionary codes) ----
-        PredWalk pw;
-        bool pred_on = preds != nullptr;
-        if (pred_on) {
-            PredBlock pb = preds[bi];
-            if (pb.err) { dev_set_err(derr, DERR_BAD_ENC, (uint64_t)bi); continue; }
-            if (!pb.active) continue;  // no tag -> nil -> equality never holds
-            pred_init(&pw, payload, &pb);
-            if (pb.uniform) {
-                // one run covers the block: predicate is block-uniform
-                pred_advance(&pw);
-                if (!pw.run_match) continue;
-                pred_on = false;  // fold as if unpredicated
+        // ---- per-row tag predicates (conjunctive, dictionary codes) ----
+        PredWalk pw0, pw1, pw2;
+        PredWalk *wp0 = nullptr, *wp1 = nullptr, *wp2 = nullptr;
+        bool skip_block = false;
+        if (preds != nullptr) {
+#pragma unroll
+            for (int sl = 0; sl < 3; sl++) {
+                if (sl >= n_preds || skip_block) continue;
+                PredBlock pb = preds[(int64_t)sl * n_blocks + bi];
+                if (pb.err) {
+                    dev_set_err(derr, DERR_BAD_ENC, (uint64_t)bi);
+                    skip_block = true;
+                    continue;
+                }
+                if (!pb.active) { skip_block = true; continue; }  // nil tag
+                PredWalk *w = sl == 0 ? &pw0 : sl == 1 ? &pw1 : &pw2;
+                pred_init(w, payload, &pb);
+                if (pb.uniform) {
+                    // one run covers the block: predicate is block-uniform
+                    pred_advance(w);
+                    if (!w->run_match) skip_block = true;
+                    // uniform + match: drop this walker, fold unpredicated
+                } else {
+                    if (sl == 0) wp0 = w;
+                    else if (sl == 1) wp1 = w;
+                    else wp2 = w;
+                }
             }
+            if (skip_block) continue;
         }
+        bool pred_on = wp0 || wp1 || wp2;
 
         // ---- field fold ----
         const uint8_t fenc = bd->field_enc;
@@ -843,8 +868,8 @@ __global__ __launch_bounds__(256) void k_scan_agg(
                 dd = decode_one_varint(fstream, &vl);
             }
             uint64_t psum, pcnt;
-            fold_arith_pred(first, dd, r0, r1, lane, &pw, &psum, &pcnt, &bmin,
-                            &bmax);
+            fold_arith_pred(first, dd, r0, r1, lane, wp0, wp1, wp2, &psum,
+                            &pcnt, &bmin, &bmax);
             bsum = lane == 0 ? psum : 0;
             nsel_eff = pcnt;
             have_minmax = (flags & KF_NEED_VALUES) && pcnt > 0;
@@ -863,19 +888,25 @@ __global__ __launch_bounds__(256) void k_scan_agg(
             uint64_t lsum = 0, lcnt = 0;
             int64_t lmn = INT64_MAX, lmx = INT64_MIN;
             bool need0 = lane == 0 && r0 <= 0 && 0 <= r1;
-            bool m0 = pred_match_rows(&pw, 0, need0);
+            bool m0 = true;
+            if (wp0) m0 = pred_match_rows(wp0, 0, need0 && m0) && m0;
+            if (wp1) m0 = pred_match_rows(wp1, 0, need0 && m0) && m0;
+            if (wp2) m0 = pred_match_rows(wp2, 0, need0 && m0) && m0;
             if (need0 && m0) { lsum += (uint64_t)first; lcnt++;
                 lmn = first; lmx = first; }
             if (dod) {
                 bool need1 = lane == 0 && r0 <= 1 && 1 <= r1;
-                bool m1 = pred_match_rows(&pw, 1, need1);
+                bool m1 = true;
+                if (wp0) m1 = pred_match_rows(wp0, 1, need1 && m1) && m1;
+                if (wp1) m1 = pred_match_rows(wp1, 1, need1 && m1) && m1;
+                if (wp2) m1 = pred_match_rows(wp2, 1, need1 && m1) && m1;
                 if (need1 && m1) { lsum += (uint64_t)row1; lcnt++;
                     lmn = row1 < lmn ? row1 : lmn; lmx = row1 > lmx ? row1 : lmx; }
             }
             ScanFold ff;
             scan_stream(s, n - 1, dod, dod ? row1 : first, d1, r0, r1,
                         INT64_MAX, INT64_MIN, lane, &ff, derr, (uint64_t)bi,
-                        &pw);
+                        wp0, wp1, wp2);
             lsum += ff.sum;
             lcnt += ff.nsel;
             lmn = ff.mn < lmn ? ff.mn : lmn;
@@ -956,7 +987,7 @@ __global__ __launch_bounds__(256) void k_scan_agg(
                 ScanFold ff;
                 scan_stream(s, n - 1, dod, dod ? row1 : first, d1, r0, r1,
                             INT64_MAX, INT64_MIN, lane, &ff, derr, (uint64_t)bi,
-                            nullptr);
+                            nullptr, nullptr, nullptr);
                 uint64_t lsum = ff.sum;
                 int64_t lmn = ff.mn, lmx = ff.mx;
                 // rows outside the stream: row 0 (value=first) and, for
@@ -1193,31 +1224,50 @@ extern "C" int bydb_reset(bydb_session *s) {
     return BYDB_OK;
 }
 
-extern "C" int bydb_consume(bydb_session *s, int64_t min_ts, int64_t max_ts,
-                            const uint8_t *pred, uint64_t pred_len) {
+extern "C" int bydb_consume_multi(bydb_session *s, int64_t min_ts,
+                                  int64_t max_ts, const uint8_t *const *preds_in,
+                                  const uint64_t *pred_lens, int n_preds) {
     HIP_TRY(s, hipSetDevice(s->device));
     if (!s->d_acc) { s->err = "configure first"; return BYDB_ERR_STATE; }
     if (s->n_blocks == 0) { s->err = "no part resident"; return BYDB_ERR_STATE; }
+    if (n_preds < 0 || n_preds > 3) { s->err = "n_preds must be 0..3"; return BYDB_ERR_BAD_ARG; }
+    // drop trailing disabled slots
+    while (n_preds > 0 && pred_lens[n_preds - 1] == 0) n_preds--;
     PredBlock *preds = nullptr;
-    if (pred_len > 0) {
-        if (s->preds_cap < s->n_blocks) {
+    if (n_preds > 0) {
+        for (int i = 0; i < n_preds; i++) {
+            if (pred_lens[i] == 0) {
+                s->err = "empty predicate in the middle of the slot list";
+                return BYDB_ERR_BAD_ARG;
+            }
+        }
+        if (s->preds_cap < s->n_blocks * 3) {
             if (s->d_preds) hipFree(s->d_preds);
-            HIP_TRY(s, hipMalloc(&s->d_preds, sizeof(PredBlock) * (size_t)s->n_blocks));
-            s->preds_cap = s->n_blocks;
+            HIP_TRY(s, hipMalloc(&s->d_preds,
+                                 sizeof(PredBlock) * (size_t)s->n_blocks * 3));
+            s->preds_cap = s->n_blocks * 3;
         }
-        if (s->pred_bytes_cap < pred_len) {
+        uint64_t total = 0;
+        for (int i = 0; i < n_preds; i++) total += pred_lens[i];
+        if (s->pred_bytes_cap < total) {
             if (s->d_pred_bytes) hipFree(s->d_pred_bytes);
-            HIP_TRY(s, hipMalloc(&s->d_pred_bytes, pred_len));
-            s->pred_bytes_cap = pred_len;
+            HIP_TRY(s, hipMalloc(&s->d_pred_bytes, total));
+            s->pred_bytes_cap = total;
         }
-        HIP_TRY(s, hipMemcpyAsync(s->d_pred_bytes, pred, pred_len,
-                                  hipMemcpyHostToDevice, s->stream));
+        uint64_t off = 0;
         int rthreads = 256;
         int rblocks = (int)((s->n_blocks + rthreads - 1) / rthreads);
-        hipLaunchKernelGGL(k_resolve_pred, dim3(rblocks), dim3(rthreads), 0,
-                           s->stream, s->d_payload, s->d_blocks, s->n_blocks,
-                           s->d_pred_bytes, pred_len, s->d_preds);
-        HIP_TRY(s, hipGetLastError());
+        for (int i = 0; i < n_preds; i++) {
+            HIP_TRY(s, hipMemcpyAsync(s->d_pred_bytes + off, preds_in[i],
+                                      pred_lens[i], hipMemcpyHostToDevice,
+                                      s->stream));
+            hipLaunchKernelGGL(k_resolve_pred, dim3(rblocks), dim3(rthreads), 0,
+                               s->stream, s->d_payload, s->d_blocks, s->n_blocks,
+                               s->d_pred_bytes + off, pred_lens[i], i,
+                               s->d_preds + (int64_t)i * s->n_blocks);
+            HIP_TRY(s, hipGetLastError());
+            off += pred_lens[i];
+        }
         preds = s->d_preds;
     }
     int flags = 0;
@@ -1232,11 +1282,21 @@ extern "C" int bydb_consume(bydb_session *s, int64_t min_ts, int64_t max_ts,
     HIP_TRY(s, hipEventRecord(s->ev_start, s->stream));
     hipLaunchKernelGGL(k_scan_agg, dim3(grid), dim3(threads), 0, s->stream,
                        s->d_payload, s->d_blocks, s->n_blocks, min_ts, max_ts,
-                       flags, preds, s->d_acc, s->d_err);
+                       flags, preds, n_preds, s->d_acc, s->d_err);
     HIP_TRY(s, hipGetLastError());
     HIP_TRY(s, hipEventRecord(s->ev_stop, s->stream));
     s->consumed = true;
     return BYDB_OK;
+}
+
+extern "C" int bydb_consume(bydb_session *s, int64_t min_ts, int64_t max_ts,
+                            const uint8_t *pred, uint64_t pred_len) {
+    if (pred_len == 0) {
+        return bydb_consume_multi(s, min_ts, max_ts, nullptr, nullptr, 0);
+    }
+    const uint8_t *preds[1] = {pred};
+    uint64_t lens[1] = {pred_len};
+    return bydb_consume_multi(s, min_ts, max_ts, preds, lens, 1);
 }
 
 extern "C" double bydb_last_consume_ms(bydb_session *s) { return s->last_ms; }

@@ -99,7 +99,11 @@ typedef struct {
     uint64_t field_off;           /* field varint stream (header-free) */
     uint64_t field_len;
     uint64_t tag_off;             /* tag column payload incl. type byte, or 0 */
-    uint64_t tag_len;
+    uint64_t tag_len;             /* slot 0; slots 1-2 below (conjunctive) */
+    uint64_t tag2_off;
+    uint64_t tag2_len;
+    uint64_t tag3_off;
+    uint64_t tag3_len;
     uint32_t group_code;          /* dense group index (group-by), else 0 */
     uint32_t _pad2;
 } bydb_block_desc;
@@ -164,6 +168,12 @@ int bydb_set_partials_buffer(bydb_session *s, void *dev_ptr, uint64_t len);
  * returns after launch; bydb_finalize syncs. */
 int bydb_consume(bydb_session *s, int64_t min_ts, int64_t max_ts,
                  const uint8_t *pred, uint64_t pred_len);
+
+/* Conjunctive multi-tag predicate (<=3 values; preds[i] applies to the
+ * block's tag slot i).  pred_lens[i] == 0 disables slot i. */
+int bydb_consume_multi(bydb_session *s, int64_t min_ts, int64_t max_ts,
+                       const uint8_t *const *preds, const uint64_t *pred_lens,
+                       int n_preds);
 
 /* ---- finalize + emit (Finalize/NextBatch) ----
  * Syncs the stream, downloads partials and finalises per-group results

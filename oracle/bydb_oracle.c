@@ -1080,9 +1080,11 @@ static void agg_reset(bo_agg_result *r) {
 
 static int scan_block(const uint8_t *payload, const bo_block_desc *b,
                       int field_vtype, int64_t min_ts, int64_t max_ts,
-                      const uint8_t *pred, int64_t pred_len, bo_agg_result *r,
+                      const uint8_t *preds_concat, const int64_t *pred_lens,
+                      bo_agg_result *r,
                       int64_t *ts_buf, int64_t *i64_buf, double *f64_buf,
-                      uint8_t *tagdata_buf, int64_t *taglen_buf) {
+                      uint8_t *tagdata_buf, int64_t *taglen_buf,
+                      uint8_t *rowmatch_buf) {
     int64_t n = (int64_t)b->count;
     int rc = bo_timestamps_decode(ts_buf, NULL, payload + b->ts_off,
                                   (size_t)(b->ts_len + b->ver_len),
@@ -1091,34 +1093,43 @@ static int scan_block(const uint8_t *payload, const bo_block_desc *b,
     if (rc != BO_OK) return rc;
     int64_t r0, r1;
     if (!bo_find_range(ts_buf, n, min_ts, max_ts, &r0, &r1)) return BO_OK;
-    /* predicate rows (tag equality on decoded tag values —
-     * the row path compares raw tag bytes after decode).  A block with no
-     * tag column has nil tags everywhere: equality never holds. */
-    if (pred_len > 0 && b->tag_len == 0) return BO_OK;
-    int have_pred = pred_len > 0 && b->tag_len > 0;
-    size_t tagdata_len = 0;
-    if (have_pred) {
-        uint8_t tag_type = payload[b->tag_off];
+    /* conjunctive tag-equality predicates (up to 3, one per tag slot).
+     * A block lacking a predicated tag column has nil tags: never equal. */
+    int have_pred = 0;
+    for (int64_t i = 0; i < n; i++) rowmatch_buf[i] = 1;
+    int64_t pred_off = 0;
+    for (int sl = 0; sl < 3; sl++) {
+        int64_t plen = pred_lens ? pred_lens[sl] : 0;
+        if (plen == 0) continue;
+        const uint8_t *pred = preds_concat + pred_off;
+        pred_off += plen;
+        have_pred = 1;
+        uint64_t toff = sl == 0 ? b->tag_off : sl == 1 ? b->tag2_off : b->tag3_off;
+        uint64_t tlen = sl == 0 ? b->tag_len : sl == 1 ? b->tag2_len : b->tag3_len;
+        if (tlen == 0) return BO_OK;  /* nil tags: exclude whole block */
+        uint8_t tag_type = payload[toff];
         if (tag_type != BO_ENC_DICTIONARY) return BO_ERR_BAD_TYPE;
+        size_t tagdata_len = 0;
         rc = bo_dictionary_decode(tagdata_buf, (size_t)1 << 24, taglen_buf,
-                                  payload + b->tag_off + 1, b->tag_len - 1, n,
+                                  payload + toff + 1, tlen - 1, n,
                                   &tagdata_len);
-        if (rc != BO_OK) return rc;
-    }
-    if (field_vtype == BO_VT_INT64) {
-        rc = bo_column_i64_decode(i64_buf, payload + b->col_off, b->col_len, n);
         if (rc != BO_OK) return rc;
         size_t tago = 0;
         for (int64_t i = 0; i < n; i++) {
-            int64_t tl = have_pred ? taglen_buf[i] : 0;
-            size_t my_tago = tago;
-            if (have_pred && tl > 0) tago += (size_t)tl;
-            if (i < r0 || i > r1) continue;
-            if (have_pred) {
-                if (tl != pred_len) continue;
-                if (tl > 0 && memcmp(tagdata_buf + my_tago, pred, (size_t)tl) != 0)
-                    continue;
-            }
+            int64_t tl = taglen_buf[i];
+            size_t my = tago;
+            if (tl > 0) tago += (size_t)tl;
+            if (tl != plen ||
+                (tl > 0 && memcmp(tagdata_buf + my, pred, (size_t)tl) != 0))
+                rowmatch_buf[i] = 0;
+        }
+    }
+    (void)have_pred;
+    if (field_vtype == BO_VT_INT64) {
+        rc = bo_column_i64_decode(i64_buf, payload + b->col_off, b->col_len, n);
+        if (rc != BO_OK) return rc;
+        for (int64_t i = r0; i <= r1; i++) {
+            if (!rowmatch_buf[i]) continue;
             int64_t v = i64_buf[i];
             r->sum_i = (int64_t)((uint64_t)r->sum_i + (uint64_t)v);
             r->count++;
@@ -1128,17 +1139,8 @@ static int scan_block(const uint8_t *payload, const bo_block_desc *b,
     } else if (field_vtype == BO_VT_FLOAT64) {
         rc = bo_column_f64_decode(f64_buf, payload + b->col_off, b->col_len, n);
         if (rc != BO_OK) return rc;
-        size_t tago = 0;
-        for (int64_t i = 0; i < n; i++) {
-            int64_t tl = have_pred ? taglen_buf[i] : 0;
-            size_t my_tago = tago;
-            if (have_pred && tl > 0) tago += (size_t)tl;
-            if (i < r0 || i > r1) continue;
-            if (have_pred) {
-                if (tl != pred_len) continue;
-                if (tl > 0 && memcmp(tagdata_buf + my_tago, pred, (size_t)tl) != 0)
-                    continue;
-            }
+        for (int64_t i = r0; i <= r1; i++) {
+            if (!rowmatch_buf[i]) continue;
             double v = f64_buf[i];
             r->sum_f += v;
             r->count++;
@@ -1165,21 +1167,33 @@ int bo_scan_agg_grouped(const uint8_t *payload, const bo_block_desc *blocks,
                         int64_t n_blocks, int field_vtype, int64_t min_ts,
                         int64_t max_ts, const uint8_t *pred, int64_t pred_len,
                         bo_agg_result *out, int64_t n_groups) {
+    int64_t lens[3] = {pred_len, 0, 0};
+    return bo_scan_agg_multi(payload, blocks, n_blocks, field_vtype, min_ts,
+                             max_ts, pred, lens, out, n_groups);
+}
+
+int bo_scan_agg_multi(const uint8_t *payload, const bo_block_desc *blocks,
+                      int64_t n_blocks, int field_vtype, int64_t min_ts,
+                      int64_t max_ts, const uint8_t *preds_concat,
+                      const int64_t pred_lens[3], bo_agg_result *out,
+                      int64_t n_groups) {
     for (int64_t g = 0; g < n_groups; g++) agg_reset(&out[g]);
     int64_t *ts_buf = (int64_t *)malloc(sizeof(int64_t) * MAX_BLOCK_ROWS);
     int64_t *i64_buf = (int64_t *)malloc(sizeof(int64_t) * MAX_BLOCK_ROWS);
     double *f64_buf = (double *)malloc(sizeof(double) * MAX_BLOCK_ROWS);
     uint8_t *tagdata_buf = (uint8_t *)malloc((size_t)1 << 24);
     int64_t *taglen_buf = (int64_t *)malloc(sizeof(int64_t) * MAX_BLOCK_ROWS);
+    uint8_t *rowmatch_buf = (uint8_t *)malloc(MAX_BLOCK_ROWS);
     int rc = BO_OK;
     for (int64_t i = 0; i < n_blocks && rc == BO_OK; i++) {
         uint32_t g = blocks[i].group_code;
         if ((int64_t)g >= n_groups) { rc = BO_ERR_BAD_DATA; break; }
-        rc = scan_block(payload, &blocks[i], field_vtype, min_ts, max_ts, pred,
-                        pred_len, &out[g], ts_buf, i64_buf, f64_buf,
-                        tagdata_buf, taglen_buf);
+        rc = scan_block(payload, &blocks[i], field_vtype, min_ts, max_ts,
+                        preds_concat, pred_lens, &out[g], ts_buf, i64_buf,
+                        f64_buf, tagdata_buf, taglen_buf, rowmatch_buf);
     }
-    free(ts_buf); free(i64_buf); free(f64_buf); free(tagdata_buf); free(taglen_buf);
+    free(ts_buf); free(i64_buf); free(f64_buf); free(tagdata_buf);
+    free(taglen_buf); free(rowmatch_buf);
     return rc;
 }
 

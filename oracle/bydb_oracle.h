@@ -152,7 +152,11 @@ typedef struct {
     uint64_t col_off;  /* offset of full field column payload (incl. header) */
     uint64_t col_len;
     uint64_t tag_off;  /* offset of tag column payload (incl. header), or 0 */
-    uint64_t tag_len;
+    uint64_t tag_len;  /* slot 0; slots 1-2 below */
+    uint64_t tag2_off;
+    uint64_t tag2_len;
+    uint64_t tag3_off;
+    uint64_t tag3_len;
     uint32_t group_code; /* dense group index for group-by parity; else 0 */
     uint32_t _pad2;
 } bo_block_desc;
@@ -181,6 +185,14 @@ int bo_scan_agg_grouped(const uint8_t *payload, const bo_block_desc *blocks,
                         int64_t n_blocks, int field_vtype, int64_t min_ts,
                         int64_t max_ts, const uint8_t *pred, int64_t pred_len,
                         bo_agg_result *out, int64_t n_groups);
+
+/* Conjunctive multi-tag variant: preds = concatenated predicate bytes,
+ * pred_lens[3] (0 disables a slot); preds[i] applies to tag slot i. */
+int bo_scan_agg_multi(const uint8_t *payload, const bo_block_desc *blocks,
+                      int64_t n_blocks, int field_vtype, int64_t min_ts,
+                      int64_t max_ts, const uint8_t *preds_concat,
+                      const int64_t pred_lens[3], bo_agg_result *out,
+                      int64_t n_groups);
 
 /* MEAN finalisation — pkg/query/aggregation/function.go:30-45 (clamps to >=1) */
 int64_t bo_mean_val_i64(int64_t sum, int64_t count);

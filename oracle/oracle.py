@@ -99,6 +99,10 @@ class BlockDesc(C.Structure):
         ("col_len", C.c_uint64),
         ("tag_off", C.c_uint64),
         ("tag_len", C.c_uint64),
+        ("tag2_off", C.c_uint64),
+        ("tag2_len", C.c_uint64),
+        ("tag3_off", C.c_uint64),
+        ("tag3_len", C.c_uint64),
         ("group_code", C.c_uint32),
         ("_pad2", C.c_uint32),
     ]
@@ -123,6 +127,10 @@ _L.bo_scan_agg_grouped.restype = C.c_int
 _L.bo_scan_agg_grouped.argtypes = [u8p, C.POINTER(BlockDesc), C.c_int64, C.c_int,
                                    C.c_int64, C.c_int64, u8p, C.c_int64,
                                    C.POINTER(AggResult), C.c_int64]
+_L.bo_scan_agg_multi.restype = C.c_int
+_L.bo_scan_agg_multi.argtypes = [u8p, C.POINTER(BlockDesc), C.c_int64, C.c_int,
+                                 C.c_int64, C.c_int64, u8p, i64p,
+                                 C.POINTER(AggResult), C.c_int64]
 
 VT_INT64 = 2
 VT_FLOAT64 = 3
@@ -391,20 +399,29 @@ INT64_MAX = 2 ** 63 - 1
 
 
 def scan_agg(payload: bytes, blocks, field_vtype, min_ts=INT64_MIN,
-             max_ts=INT64_MAX, pred: bytes = b"", n_groups=1):
+             max_ts=INT64_MAX, pred: bytes = b"", n_groups=1, preds=None):
     """Run the oracle scan+aggregate over a part payload + block directory.
 
-    blocks: list of dicts with BlockDesc fields.  Returns list of AggResult
-    (length n_groups)."""
+    blocks: list of dicts with BlockDesc fields.  preds: optional list of
+    up to 3 predicate byte strings (conjunctive, one per tag slot);
+    overrides pred.  Returns list of AggResult (length n_groups)."""
     descs = (BlockDesc * len(blocks))()
     for i, b in enumerate(blocks):
         for k, v in b.items():
             setattr(descs[i], k, v)
     out = (AggResult * n_groups)()
     src = (C.c_uint8 * max(len(payload), 1)).from_buffer_copy(payload or b"\0")
-    predbuf = (C.c_uint8 * max(len(pred), 1)).from_buffer_copy(pred or b"\0")
-    rc = _L.bo_scan_agg_grouped(src, descs, len(blocks), field_vtype, min_ts,
-                                max_ts, predbuf, len(pred), out, n_groups)
+    if preds is not None:
+        concat = b"".join(p or b"" for p in preds)
+        lens = [len(p or b"") for p in preds] + [0, 0, 0]
+        predbuf = (C.c_uint8 * max(len(concat), 1)).from_buffer_copy(concat or b"\0")
+        rc = _L.bo_scan_agg_multi(src, descs, len(blocks), field_vtype, min_ts,
+                                  max_ts, predbuf, (C.c_int64 * 3)(*lens[:3]),
+                                  out, n_groups)
+    else:
+        predbuf = (C.c_uint8 * max(len(pred), 1)).from_buffer_copy(pred or b"\0")
+        rc = _L.bo_scan_agg_grouped(src, descs, len(blocks), field_vtype, min_ts,
+                                    max_ts, predbuf, len(pred), out, n_groups)
     if rc != 0:
         raise ValueError(f"scan_agg failed rc={rc}")
     return list(out)

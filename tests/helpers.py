@@ -27,6 +27,8 @@ def oracle_blocks(builder):
             ts_off=d.ts_off, ts_len=d.ts_len, ver_len=0,
             col_off=col_off, col_len=len(col),
             tag_off=d.tag_off, tag_len=d.tag_len,
+            tag2_off=d.tag2_off, tag2_len=d.tag2_len,
+            tag3_off=d.tag3_off, tag3_len=d.tag3_len,
             group_code=d.group_code))
     return bytes(payload), blocks
 

@@ -38,7 +38,7 @@ def test_so_exports_all_header_symbols():
 
 def test_struct_layouts_match_header():
     # ctypes mirrors must track the header structs
-    assert ctypes.sizeof(banyandb_amd.BlockDesc) == 112
+    assert ctypes.sizeof(banyandb_amd.BlockDesc) == 144
     assert ctypes.sizeof(banyandb_amd.Partial) == 48
     assert ctypes.sizeof(banyandb_amd.Result) == 72
 

@@ -132,6 +132,82 @@ def test_config3_float_minmax_avg_with_predicate():
     assert math.isclose(g.mean_f, want_mean, rel_tol=1e-9)
 
 
+def test_config5_three_tag_conjunctive():
+    """SURVEY config 5 shape: 3-tag conjunctive filter (dict cards 4/16/256)
+    + group-by, mixed selectivity."""
+    rng = random.Random(41)
+    b = PartBuilder()
+    n_groups = 32
+    regions = [f"r{i}".encode() for i in range(16)]
+    svcs = [f"svc{i}".encode() for i in range(256)]
+    for sid in range(48):
+        n = 2000
+        ts = [T0 + i * MS for i in range(n)]
+        b.add_block_i64(sid + 1, ts, [1] * n,
+                        [rng.randint(-10**9, 10**9) for _ in range(n)],
+                        group_code=sid % n_groups)
+        b.set_block_tag([ENVS[sid % 4]] * n)                  # env: entity
+        b.set_block_tag([regions[sid % 16]] * n)              # region: entity
+        # svc: row-varying runs; global card 256, but each block draws from
+        # a small subset so the per-block dict values stay in a plain
+        # (<128 B) compress_block — the device-parseable form (zstd'd
+        # dictionaries are a host-side resolve fallback, later round)
+        subset = [svcs[(sid * 7 + k) % 256] for k in range(8)] + [b"svc7"]
+        tags = []
+        while len(tags) < n:
+            run = min(rng.randint(1, 120), n - len(tags))
+            tags.extend([subset[rng.randrange(len(subset))]] * run)
+        b.set_block_tag(tags)
+    preds = [b"prod", b"r2", b"svc7"]
+    orc = oracle_scan(b, VT_INT64, preds=preds, n_groups=n_groups)
+    s = Session(0)
+    s.upload_part(b)
+    s.configure(VT_INT64, [AGG_SUM, AGG_COUNT, AGG_MIN, AGG_MAX],
+                n_groups=n_groups)
+    s.consume(preds=preds)
+    gs = s.finalize()
+    s.close()
+    total = sum(oc.count for oc in orc)
+    assert total > 0, "predicate selected nothing; bad test setup"
+    for g, oc in zip(gs, orc):
+        assert g.count == oc.count
+        assert g.sum_i == oc.sum_i
+        if oc.count:
+            assert g.min_i == oc.min_i and g.max_i == oc.max_i
+
+
+def test_two_tag_conjunctive_rowvarying():
+    rng = random.Random(42)
+    b = PartBuilder()
+    n = 3000
+    ts = [T0 + i * MS for i in range(n)]
+    for sid in range(6):
+        b.add_block_i64(sid + 1, ts, [1] * n,
+                        [rng.randint(0, 10**6) for _ in range(n)])
+        t1, t2 = [], []
+        while len(t1) < n:
+            run = min(rng.randint(1, 97), n - len(t1))
+            t1.extend([ENVS[rng.randrange(4)]] * run)
+        while len(t2) < n:
+            run = min(rng.randint(1, 53), n - len(t2))
+            t2.extend([None if rng.random() < 0.2 else
+                       f"z{rng.randrange(8)}".encode()] * run)
+        b.set_block_tag(t1)
+        b.set_block_tag(t2)
+    preds = [b"dev", b"z3"]
+    orc = oracle_scan(b, VT_INT64, preds=preds)[0]
+    s = Session(0)
+    s.upload_part(b)
+    s.configure(VT_INT64, [AGG_SUM, AGG_COUNT, AGG_MIN, AGG_MAX])
+    s.consume(preds=preds)
+    g = s.finalize()[0]
+    s.close()
+    assert g.count == orc.count
+    assert g.sum_i == orc.sum_i
+    if orc.count:
+        assert g.min_i == orc.min_i and g.max_i == orc.max_i
+
+
 def test_predicate_with_time_clamp():
     rng = random.Random(36)
     b = rowvary_tag_part(rng, n_series=4, n=3000)
