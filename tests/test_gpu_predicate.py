@@ -147,7 +147,7 @@ def test_config5_three_tag_conjunctive():
                         [rng.randint(-10**9, 10**9) for _ in range(n)],
                         group_code=sid % n_groups)
         b.set_block_tag([ENVS[sid % 4]] * n)                  # env: entity
-        b.set_block_tag([regions[sid % 16]] * n)              # region: entity
+        b.set_block_tag([regions[(sid // 4) % 16]] * n)       # region: entity
         # svc: row-varying runs; global card 256, but each block draws from
         # a small subset so the per-block dict values stay in a plain
         # (<128 B) compress_block — the device-parseable form (zstd'd
