@@ -368,3 +368,56 @@ extern "C" int bydb_frame_col_var(bydb_frame_reader *r, uint64_t ci,
     *data_len = o;
     return BYDB_OK;
 }
+
+// ---- top-N over group results ----
+// Restates BatchTop's ordering contract (pkg/query/vectorized/measure/
+// top.go:31-121 + ApplyTopToReduce, reduce.go:290): asc keeps the lowest N
+// values (output smallest first), desc the highest N (largest first);
+// ties surface in insertion order (earlier group wins, top.go:78-81);
+// empty groups (no rows folded) are nulls and sort lowest (top.go:34,95).
+// value_sel: 0 sum_i, 1 count, 2 min_i, 3 max_i, 4 mean_i,
+//            5 sum_f, 6 min_f, 7 max_f, 8 mean_f.
+#include <algorithm>
+
+extern "C" int bydb_top_groups(const bydb_result *results, int64_t n_groups,
+                               int value_sel, int64_t k, int asc,
+                               int64_t *out_idx, int64_t *out_n) {
+    if (k < 0 || value_sel < 0 || value_sel > 8) return BYDB_ERR_BAD_ARG;
+    struct Row { int64_t idx; int64_t iv; double fv; bool is_f; bool null; };
+    std::vector<Row> rows((size_t)n_groups);
+    bool is_f = value_sel >= 5;
+    for (int64_t g = 0; g < n_groups; g++) {
+        const bydb_result *r = &results[g];
+        Row &w = rows[(size_t)g];
+        w.idx = g;
+        w.is_f = is_f;
+        w.null = r->count == 0;
+        switch (value_sel) {
+        case 0: w.iv = r->sum_i; break;
+        case 1: w.iv = r->count; break;
+        case 2: w.iv = r->min_i; break;
+        case 3: w.iv = r->max_i; break;
+        case 4: w.iv = r->mean_i; break;
+        case 5: w.fv = r->sum_f; break;
+        case 6: w.fv = r->min_f; break;
+        case 7: w.fv = r->max_f; break;
+        default: w.fv = r->mean_f; break;
+        }
+    }
+    auto cmp = [&](const Row &a, const Row &b) {
+        // cmpTopVal: nulls lowest (top.go:95-121)
+        int c;
+        if (a.null && b.null) c = 0;
+        else if (a.null) c = -1;
+        else if (b.null) c = 1;
+        else if (is_f) c = a.fv < b.fv ? -1 : a.fv > b.fv ? 1 : 0;
+        else c = a.iv < b.iv ? -1 : a.iv > b.iv ? 1 : 0;
+        if (c != 0) return asc ? c < 0 : c > 0;
+        return a.idx < b.idx;  // ties: insertion order (earlier wins)
+    };
+    std::stable_sort(rows.begin(), rows.end(), cmp);
+    int64_t n = k < n_groups ? k : n_groups;
+    for (int64_t i = 0; i < n; i++) out_idx[i] = rows[(size_t)i].idx;
+    *out_n = n;
+    return BYDB_OK;
+}
