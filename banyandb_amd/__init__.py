@@ -149,6 +149,9 @@ def _load():
                                                     i64p, f64p, C.c_int64, C.c_uint32]
     lib.bydb_part_builder_set_block_tag.restype = C.c_int
     lib.bydb_part_builder_set_block_tag.argtypes = [C.c_void_p, u8p, i64p, C.c_int64]
+    lib.bydb_part_builder_set_tag_table.restype = C.c_int
+    lib.bydb_part_builder_set_tag_table.argtypes = [C.c_void_p, C.c_int, u8p, i64p,
+                                                    C.c_int64]
     lib.bydb_part_builder_payload_len.restype = C.c_uint64
     lib.bydb_part_builder_payload_len.argtypes = [C.c_void_p]
     lib.bydb_part_builder_payload.restype = u8p
@@ -213,6 +216,15 @@ class PartBuilder:
             self._h, series_id, (C.c_int64 * n)(*ts), (C.c_int64 * n)(*versions),
             (C.c_double * n)(*vals), n, group_code))
 
+    def set_tag_table(self, slot, values):
+        """Entity-tag table: generators attach values[series %% len] as a
+        constant tag column (slot) to every generated block."""
+        data = b"".join(v for v in values if v is not None)
+        lens = [(-1 if v is None else len(v)) for v in values]
+        buf = (C.c_uint8 * max(len(data), 1)).from_buffer_copy(data or b"\0")
+        self._ck(_lib.bydb_part_builder_set_tag_table(
+            self._h, slot, buf, (C.c_int64 * len(values))(*lens), len(values)))
+
     def set_block_tag(self, values):
         """values: list of bytes-or-None, one per row of the last block."""
         data = b"".join(v for v in values if v is not None)
@@ -275,9 +287,12 @@ class PartBuilder:
         self._ck(_lib.bydb_part_builder_drain(self._h))
 
     def __del__(self):
-        if getattr(self, "_h", None):
-            _lib.bydb_part_builder_destroy(self._h)
-            self._h = None
+        try:
+            if getattr(self, "_h", None):
+                _lib.bydb_part_builder_destroy(self._h)
+        except Exception:
+            pass  # interpreter shutdown
+        self._h = None
 
 
 class Session:

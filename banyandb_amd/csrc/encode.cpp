@@ -396,6 +396,9 @@ struct bydb_part_builder {
     std::vector<bydb_block_desc> blocks;
     uint64_t base_off = 0;  // absolute offset of payload[0] within the part
     std::string err;
+    // per-slot tag tables: the generators attach tag_table[slot][series %
+    // table_size] as a constant (entity) tag column to every block
+    std::vector<std::vector<uint8_t>> tag_table[3];
     // scratch
     std::vector<int64_t> scratch_i64;
     std::vector<double> scratch_f64;
@@ -546,6 +549,38 @@ extern "C" int bydb_part_builder_drain(bydb_part_builder *b) {
     return BYDB_OK;
 }
 
+extern "C" int bydb_part_builder_set_tag_table(bydb_part_builder *b, int slot,
+                                               const uint8_t *data,
+                                               const int64_t *lens,
+                                               int64_t n_values) {
+    if (slot < 0 || slot > 2) return BYDB_ERR_BAD_ARG;
+    b->tag_table[slot].clear();
+    const uint8_t *p = data;
+    for (int64_t i = 0; i < n_values; i++) {
+        if (lens[i] < 0) { b->tag_table[slot].emplace_back(); continue; }
+        b->tag_table[slot].emplace_back(p, p + lens[i]);
+        p += lens[i];
+    }
+    return BYDB_OK;
+}
+
+// attach each configured tag table's entry for this series to the block
+// just added (entity tags: constant per series)
+static int attach_auto_tags(bydb_part_builder *b, uint64_t series_index,
+                            int64_t n) {
+    for (int slot = 0; slot < 3; slot++) {
+        if (b->tag_table[slot].empty()) continue;
+        const auto &v = b->tag_table[slot][series_index % b->tag_table[slot].size()];
+        std::vector<uint8_t> data;
+        std::vector<int64_t> lens((size_t)n, (int64_t)v.size());
+        data.reserve(v.size() * (size_t)n);
+        for (int64_t i = 0; i < n; i++) data.insert(data.end(), v.begin(), v.end());
+        int rc = bydb_part_builder_set_block_tag(b, data.data(), lens.data(), n);
+        if (rc != BYDB_OK) return rc;
+    }
+    return BYDB_OK;
+}
+
 // ===================== synthetic generator =====================
 // splitmix64 — deterministic synthetic noise stream (seed spec: DESIGN.md).
 static inline uint64_t splitmix64(uint64_t *state) {
@@ -580,6 +615,7 @@ extern "C" int bydb_gen_series_i64(bydb_part_builder *b, uint64_t series_index,
         int rc = bydb_part_builder_add_block_i64(
             b, sid, b->scratch_ts.data(), b->scratch_ver.data(),
             b->scratch_i64.data(), n, group_code);
+        if (rc == BYDB_OK) rc = attach_auto_tags(b, series_index, n);
         if (rc != BYDB_OK) return rc;
         done += n;
     }
@@ -600,17 +636,45 @@ extern "C" int bydb_gen_series_f64(bydb_part_builder *b, uint64_t series_index,
         b->scratch_f64.resize((size_t)n);
         b->scratch_ts.resize((size_t)n);
         b->scratch_ver.resize((size_t)n);
+        b->scratch_i64.resize((size_t)n);
+        bool fast_ok = true;
+        int16_t min_exp = INT16_MAX;
+        std::vector<int16_t> exps((size_t)n);
         for (int64_t i = 0; i < n; i++) {
             int64_t gi = done + i;
             int64_t noise = (int64_t)(splitmix64(&st) % 601) - 300;  // cents
             int64_t cents = base_cents + gi * ramp_cents + noise;
-            b->scratch_f64[(size_t)i] = (double)cents / 100.0;
             b->scratch_ts[(size_t)i] = t0 + gi * stride_ns;
             b->scratch_ver[(size_t)i] = 1;
+            b->scratch_f64[(size_t)i] = (double)cents / 100.0;
+            if (cents > 1000000000000LL || cents < -1000000000000LL)
+                fast_ok = false;
+            // floatToDecimal(cents/100.0) == (cents stripped of trailing
+            // zeros, -2 + strips) for |cents| <= 1e12 (shortest-repr
+            // uniqueness at <= 13 significant digits)
+            int64_t d = cents;
+            int16_t e = -2;
+            if (d == 0) e = 0;
+            else while (d % 10 == 0) { d /= 10; e++; }
+            b->scratch_i64[(size_t)i] = d;
+            exps[(size_t)i] = e;
+            if (e < min_exp) min_exp = e;
         }
-        int rc = bydb_part_builder_add_block_f64(
-            b, sid, b->scratch_ts.data(), b->scratch_ver.data(),
-            b->scratch_f64.data(), n, group_code);
+        int rc;
+        if (fast_ok) {
+            for (int64_t i = 0; i < n; i++) {
+                int diff = exps[(size_t)i] - min_exp;
+                while (diff-- > 0) b->scratch_i64[(size_t)i] *= 10;
+            }
+            rc = add_block_common(b, sid, b->scratch_ts.data(),
+                                  b->scratch_ver.data(), b->scratch_i64.data(),
+                                  min_exp, BYDB_VT_FLOAT64, n, group_code);
+        } else {
+            rc = bydb_part_builder_add_block_f64(
+                b, sid, b->scratch_ts.data(), b->scratch_ver.data(),
+                b->scratch_f64.data(), n, group_code);
+        }
+        if (rc == BYDB_OK) rc = attach_auto_tags(b, series_index, n);
         if (rc != BYDB_OK) return rc;
         done += n;
     }
