@@ -50,6 +50,13 @@ def main():
                     default=int(os.environ.get("BYDB_BENCH_DP", 1_000_000)))
     ap.add_argument("--skip-cpu-baseline", action="store_true",
                     default=os.environ.get("BYDB_SKIP_CPU_BASELINE") == "1")
+    ap.add_argument("--workload", default="i64_sum",
+                    choices=["i64_sum", "f64_pred", "group256",
+                             "group4096_pred3"],
+                    help="i64_sum = BASELINE configs[1] (default, the "
+                         "headline metric); f64_pred = configs[2] shape; "
+                         "group256 = configs[3] shape; group4096_pred3 = "
+                         "configs[4] shape (int64 leg)")
     args = ap.parse_args()
 
     import torch
@@ -70,15 +77,46 @@ def main():
 
     n_series, n_dp = args.series, args.dp
     full_config = n_series == 10000 and n_dp == 1_000_000
-    workload = (f"{n_series}_series_x_{n_dp}_int64_sum_count"
-                if not full_config else "10k_series_x_1M_int64_sum_count")
+    # per-workload parameters
+    ENVS = [b"prod", b"dev", b"staging", b"qa"]
+    REGIONS = [f"r{i}".encode() for i in range(16)]
+    SVCS = [f"s{i}".encode() for i in range(256)]
+    W = {
+        "i64_sum": dict(float=False, n_groups=1, group_mod=0, tags=[],
+                        preds=[], funcs=["sum", "count"], algo_bpd=16,
+                        dtype="int64", agg="sum+count", enc_bpd=1.05),
+        "f64_pred": dict(float=True, n_groups=1, group_mod=0, tags=[ENVS],
+                         preds=[b"prod"], funcs=["sum", "count", "min", "max"],
+                         algo_bpd=17, dtype="f64", agg="min+max+avg",
+                         enc_bpd=2.0),
+        "group256": dict(float=False, n_groups=256, group_mod=256, tags=[],
+                         preds=[], funcs=["sum", "count"], algo_bpd=16,
+                         dtype="int64", agg="sum+count groupby(256)",
+                         enc_bpd=1.05),
+        "group4096_pred3": dict(float=False, n_groups=4096, group_mod=4096,
+                                tags=[ENVS, REGIONS, SVCS],
+                                preds=[b"prod", b"r2", b"s7"],
+                                funcs=["sum", "count"], algo_bpd=19,
+                                dtype="int64",
+                                agg="sum+count groupby(4096) 3-tag filter",
+                                enc_bpd=1.05),
+    }[args.workload]
+    base_name = {"i64_sum": "int64_sum_count", "f64_pred": "f64_minmaxavg_pred",
+                 "group256": "int64_group256",
+                 "group4096_pred3": "int64_group4096_pred3"}[args.workload]
+    workload = (f"10k_series_x_1M_{base_name}" if full_config
+                else f"{n_series}_series_x_{n_dp}_{base_name}")
+    if args.workload == "i64_sum" and full_config:
+        workload = "10k_series_x_1M_int64_sum_count"
 
     # ---- build + upload the rank's shard (untimed) ----
     threads = max(1, (os.cpu_count() or 8) // max(world, 1))
     t0_r = T0 + rank * n_dp * STRIDE
     seed_r = SEED ^ (rank << 32)
     total_dp_rank = n_series * n_dp
-    est_payload = int(total_dp_rank * 1.05) + n_series * 64  # ~1.0 B/dp + block overhead
+    tag_overhead = 64 * len(W["tags"])  # per-block uniform dict columns
+    est_payload = int(total_dp_rank * W["enc_bpd"]) + n_series * (64 + tag_overhead) \
+        + n_series * ((n_dp + 8191) // 8192) * tag_overhead
     est_blocks = n_series * ((n_dp + 8191) // 8192)
 
     sess = ba.Session(local_rank)
@@ -87,19 +125,29 @@ def main():
     gen_t = time.perf_counter()
     chunk = 500
     b = ba.PartBuilder()
+    for slot, table in enumerate(W["tags"]):
+        b.set_tag_table(slot, table)
     uploaded_dp = 0
     for s0 in range(0, n_series, chunk):
         ns = min(chunk, n_series - s0)
-        b.gen_bulk_i64(s0, ns, n_dp, t0_r, STRIDE, 1000, 1, seed_r,
-                       group_mod=0, threads=threads)
+        if W["float"]:
+            b.gen_bulk_f64(s0, ns, n_dp, t0_r, STRIDE, 10.0, 0.01, seed_r,
+                           group_mod=W["group_mod"], threads=threads)
+        else:
+            b.gen_bulk_i64(s0, ns, n_dp, t0_r, STRIDE, 1000, 1, seed_r,
+                           group_mod=W["group_mod"], threads=threads)
         sess.append(b)
         uploaded_dp += ns * n_dp
         b.drain()
     gen_s = time.perf_counter() - gen_t
     log(f"rank {rank}: generated+uploaded {uploaded_dp} dp in {gen_s:.1f}s")
 
-    n_groups = 1
-    sess.configure(ba.VT_INT64, [ba.AGG_SUM, ba.AGG_COUNT], n_groups=n_groups)
+    n_groups = W["n_groups"]
+    fmap = {"sum": ba.AGG_SUM, "count": ba.AGG_COUNT, "min": ba.AGG_MIN,
+            "max": ba.AGG_MAX}
+    sess.configure(ba.VT_FLOAT64 if W["float"] else ba.VT_INT64,
+                   [fmap[f] for f in W["funcs"]], n_groups=n_groups,
+                   float_exp=-2 if W["float"] else 0)
 
     # partials live in a torch CUDA tensor so the merge is RCCL over xGMI
     part_t = None
@@ -108,15 +156,20 @@ def main():
                              device=f"cuda:{local_rank}")
         sess.set_partials_buffer(part_t.data_ptr(), part_t.numel() * 8)
 
+    preds = W["preds"] or None
+    need_minmax = "min" in W["funcs"] or "max" in W["funcs"]
+
     def one_step():
         sess.reset()
-        sess.consume(min_ts=t0_r, max_ts=t0_r + n_dp * STRIDE)
+        sess.consume(min_ts=t0_r, max_ts=t0_r + n_dp * STRIDE, preds=preds)
         if world > 1:
             # sync the session stream, then RCCL-merge the partials over
             # xGMI (AggModeReduce Combine semantics)
             parts = sess.finalize_partials()
             from banyandb_amd.distributed import allreduce_partials
-            allreduce_partials(dist, part_t, n_groups)
+            allreduce_partials(dist, part_t, n_groups,
+                               need_minmax=need_minmax,
+                               need_float=W["float"])
             return parts
         return sess.finalize_partials()
 
@@ -142,17 +195,21 @@ def main():
         elapsed = float(t.item())
 
     # ---- correctness spot-check (outside the timed region) ----
-    parts = sess.finalize_partials() if world == 1 else None
     if world == 1:
-        assert parts[0].count == total_dp_rank, \
-            f"count {parts[0].count} != {total_dp_rank}"
+        parts = sess.finalize_partials()
+        total_count = sum(p.count for p in parts)
+        if not W["preds"]:
+            assert total_count == total_dp_rank, \
+                f"count {total_count} != {total_dp_rank}"
+        else:
+            assert 0 < total_count < total_dp_rank, total_count
 
     ms_per_step = elapsed * 1000.0 / args.steps
     total_dp = total_dp_rank * max(world, 1)
     value = total_dp * args.steps / elapsed
 
     avg_launch_ms = sum(launch_ms) / len(launch_ms)
-    algo_bytes_per_launch = ALGO_BYTES_PER_DP * total_dp_rank
+    algo_bytes_per_launch = W["algo_bpd"] * total_dp_rank
     achieved_gbs = algo_bytes_per_launch / (avg_launch_ms / 1e3) / 1e9
 
     traffic = None
@@ -166,7 +223,8 @@ def main():
             pass
 
     cpu_baseline = None
-    if rank == 0 and world == 1 and not args.skip_cpu_baseline:
+    if (rank == 0 and world == 1 and not args.skip_cpu_baseline
+            and args.workload == "i64_sum"):
         cpu_baseline = run_cpu_baseline(n_dp)
 
     if rank == 0:
@@ -181,13 +239,13 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "int64",
+            "dtype": W["dtype"],
             "data": "synthetic",
             "config": {
                 "workload": workload,
                 "series_per_shard": n_series,
                 "datapoints_per_series": n_dp,
-                "agg": "sum+count",
+                "agg": W["agg"],
                 "parallelism": f"time-bucket shards x{max(world,1)}",
             },
             "roofline": {
