@@ -707,6 +707,8 @@ static int bulk_gen(bydb_part_builder *b, uint64_t first_index,
     if (n_threads > 64) n_threads = 64;
     if ((int64_t)n_threads > n_series) n_threads = (int)n_series;
     std::vector<bydb_part_builder> locals((size_t)n_threads);
+    for (auto &lb : locals)
+        for (int sl = 0; sl < 3; sl++) lb.tag_table[sl] = b->tag_table[sl];
     std::vector<int> rcs((size_t)n_threads, BYDB_OK);
     std::vector<std::thread> threads;
     int64_t per = (n_series + n_threads - 1) / n_threads;

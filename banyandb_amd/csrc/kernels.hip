@@ -226,6 +226,66 @@ __device__ void dense_region(const uint8_t *stream, int64_t lo, int64_t hi,
     *out_sum_jd = acc_jd;
 }
 
+// wave-wide inclusive scan of an int32 (per-window lane sums)
+__device__ __forceinline__ int32_t wave_incl_scan32(int32_t v, int lane) {
+#pragma unroll
+    for (int off = 1; off < 64; off <<= 1) {
+        int32_t t = __shfl_up(v, off);
+        if (lane >= off) v += t;
+    }
+    return v;
+}
+
+// Dense min/max over an all-1-byte delta stream: fold min/max of the
+// values v_j for delta bytes b in [b_lo, b_hi) (v after applying byte b),
+// with base_val = v at delta index b_lo (carry-in).  256-byte windows,
+// SWAR zigzag, per-window lane-sum scan — replaces the latency-chained
+// 64-byte ballot scan for the dominant stream shape.
+__device__ void dense_minmax(const uint8_t *stream, int64_t b_lo, int64_t b_hi,
+                             int64_t base_val, int lane, int64_t *out_mn,
+                             int64_t *out_mx) {
+    int64_t lmn = INT64_MAX, lmx = INT64_MIN;
+    if (b_hi > b_lo) {
+        const uint32_t *wp =
+            (const uint32_t *)((uintptr_t)(stream + b_lo) & ~(uintptr_t)3);
+        int64_t wbase = (int64_t)((uintptr_t)wp - (uintptr_t)stream);
+        uint64_t carry = (uint64_t)base_val;  // wrap-safe (Go int64 wraps)
+        while (wbase < b_hi) {
+            uint32_t w = wp[lane];
+            uint32_t sb = swar_zigzag(w);
+            // mask bytes outside [b_lo, b_hi): zero their deltas and skip fold
+            int64_t g0 = wbase + 4 * lane;
+            uint32_t keep = 0;
+#pragma unroll
+            for (int k = 0; k < 4; k++) {
+                int64_t g = g0 + k;
+                if (g >= b_lo && g < b_hi) keep |= 0xffu << (8 * k);
+            }
+            sb &= keep;
+            int32_t t = dot4_i8(sb, 0x01010101u, 0);
+            int32_t incl = wave_incl_scan32(t, lane);
+            uint64_t lane_base = carry + (uint64_t)(int64_t)(incl - t);
+            int32_t cum = 0;
+#pragma unroll
+            for (int k = 0; k < 4; k++) {
+                int32_t d = (int32_t)(int8_t)(sb >> (8 * k));
+                cum += d;
+                int64_t g = g0 + k;
+                if (g >= b_lo && g < b_hi) {
+                    int64_t v = (int64_t)(lane_base + (uint64_t)(int64_t)cum);
+                    lmn = v < lmn ? v : lmn;
+                    lmx = v > lmx ? v : lmx;
+                }
+            }
+            carry += (uint64_t)(int64_t)__shfl(incl, 63);
+            wp += 64;
+            wbase += 256;
+        }
+    }
+    *out_mn = wave_reduce_min(lmn);
+    *out_mx = wave_reduce_max(lmx);
+}
+
 // Weighted fold over an all-1-byte delta stream:
 //   sum_{j=1..jend} d_j * w(j),  w(j) = nsel (j<=r0) | r1-j+1 (else)
 // via two dense regions (delta j lives at byte j-1):
@@ -975,6 +1035,23 @@ __global__ __launch_bounds__(256) void k_scan_agg(
                         bsum = (uint64_t)first * nsel + (uint64_t)d1 * si + acc;
                     }
                 }
+            } else if (!dod && bd->field_len == (uint64_t)(n - 1)) {
+                // all-1-byte delta stream: dense weighted sum + dense
+                // min/max (no serial window chain)
+                uint64_t acc = dense_delta_weighted(fstream, n - 1, r0, r1,
+                                                    lane);
+                acc = wave_reduce_add(acc);
+                bsum = (uint64_t)first * nsel + acc;
+                bsum = lane == 0 ? bsum : 0;
+                // v at delta index r0 (v_{r0}) = first + sum of bytes [0,r0)
+                int64_t psd, psjd;
+                dense_region(fstream, 0, r0, lane, &psd, &psjd);
+                int64_t vr0 = (int64_t)((uint64_t)first +
+                                        wave_reduce_add((uint64_t)psd));
+                dense_minmax(fstream, r0, r1, vr0, lane, &bmin, &bmax);
+                bmin = vr0 < bmin ? vr0 : bmin;  // row r0 itself
+                bmax = vr0 > bmax ? vr0 : bmax;
+                have_minmax = true;
             } else {
                 const uint8_t *s = fstream;
                 int64_t d1 = 0;
