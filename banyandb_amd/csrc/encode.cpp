@@ -696,6 +696,8 @@ static void splice_builder(bydb_part_builder *dst, bydb_part_builder *src) {
         d.ts_off += shift;
         d.field_off += shift;
         if (d.tag_len) d.tag_off += shift;
+        if (d.tag2_len) d.tag2_off += shift;
+        if (d.tag3_len) d.tag3_off += shift;
         dst->blocks.push_back(d);
     }
 }
