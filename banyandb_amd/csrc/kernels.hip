@@ -784,6 +784,101 @@ __device__ void scan_stream(const uint8_t *stream, int64_t n_deltas, bool dod,
     f->n_hi = l_nhi;
 }
 
+// ---------------- varint segment index ----------------
+// The value-reconstruction path over a mixed-width varint stream is a
+// serial chain of ballot windows (the next window's address depends on the
+// current ballot).  For a RESIDENT part that chain can be broken once:
+// a build pass walks each eligible stream and records, every SEG_ROWS
+// values, the byte offset and running value at the segment start.  The
+// scan then processes (block, segment) pairs independently — 8x more
+// parallel units and no cross-segment dependency.  The index is built
+// once per part upload + agg shape (like the reference's block metadata,
+// it is derived state over immutable part bytes).
+#define SEG_ROWS 1024
+#define MAX_SEGS 8
+#define SEG_INELIGIBLE 0xFFFFFFFFu
+
+struct SegEntry {
+    uint32_t byte_off;   // offset of the segment's first delta in the stream
+    uint32_t _pad;
+    int64_t v_start;     // value at row seg*SEG_ROWS (before the first delta)
+};
+
+__global__ __launch_bounds__(256) void k_build_seg_index(
+    const uint8_t *__restrict__ payload, const bydb_block_desc *__restrict__ blocks,
+    int64_t n_blocks, SegEntry *__restrict__ segs, DevErr *derr) {
+    const int lane = threadIdx.x & 63;
+    int64_t wave_id = (int64_t)blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+    int64_t n_waves = (int64_t)gridDim.x * (blockDim.x >> 6);
+    for (int64_t bi = wave_id; bi < n_blocks; bi += n_waves) {
+        const bydb_block_desc *bd = &blocks[bi];
+        SegEntry *bseg = segs + bi * MAX_SEGS;
+        const int64_t n = (int64_t)bd->count;
+        bool eligible = bd->field_enc == BYDB_ENC_DELTA && n > SEG_ROWS &&
+                        bd->field_len != (uint64_t)(n - 1);
+        if (!eligible) {
+            if (lane == 0) bseg[0].byte_off = SEG_INELIGIBLE;
+            continue;
+        }
+        if (lane == 0) {
+            bseg[0].byte_off = 0;
+            bseg[0].v_start = bd->field_first;
+        }
+        const uint8_t *stream = payload + bd->field_off;
+        const int64_t n_deltas = n - 1;
+        uint64_t pos = 0;
+        int64_t j = 1;
+        uint64_t v_carry = (uint64_t)bd->field_first;
+        while (j <= n_deltas) {
+            uint8_t b = stream[pos + (uint64_t)lane];
+            uint64_t emask = __ballot(b < 0x80);
+            if (emask == 0) { dev_set_err(derr, DERR_BAD_STREAM, (uint64_t)bi); break; }
+            int rank = __popcll(emask & lanemask_lt(lane));
+            int64_t myj = j + rank;
+            bool is_term = (b < 0x80) && (myj <= n_deltas);
+            uint64_t d = 0;
+            if (emask == ~0ull) {
+                d = (uint64_t)zz_dec(b);
+            } else if (is_term) {
+                uint64_t below = emask & lanemask_lt(lane);
+                int start = below ? (64 - __clzll(below)) : 0;
+                uint64_t u = 0;
+                unsigned sh = 0;
+                for (int i = start; i < lane; ++i) {
+                    u |= (uint64_t)(stream[pos + (uint64_t)i] & 0x7f) << sh;
+                    sh += 7;
+                }
+                u |= (uint64_t)b << sh;
+                d = (uint64_t)zz_dec(u);
+            }
+            if (!is_term) d = 0;
+            uint64_t s = wave_incl_scan(d, lane);
+            // a lane whose value index is a SEG_ROWS multiple records the
+            // next segment's start (byte after my terminator; v = v_myj)
+            if (is_term && (myj % SEG_ROWS) == 0 && myj / SEG_ROWS < MAX_SEGS) {
+                SegEntry e;
+                e.byte_off = (uint32_t)(pos + (uint64_t)lane + 1);
+                e._pad = 0;
+                e.v_start = (int64_t)(v_carry + s);
+                bseg[myj / SEG_ROWS] = e;
+            }
+            int nterm_all = __popcll(emask);
+            int64_t nvals = (n_deltas - j + 1) < (int64_t)nterm_all
+                                ? (n_deltas - j + 1) : (int64_t)nterm_all;
+            uint64_t mm = emask;
+            int last_lane = 0;
+            for (int t = 0; t < nvals; t++) {
+                last_lane = __ffsll((unsigned long long)mm) - 1;
+                mm &= mm - 1;
+            }
+            v_carry += (uint64_t)__shfl((long long)s, last_lane);
+            if (j + nterm_all > n_deltas) break;
+            j += nterm_all;
+            pos += (uint64_t)(64 - __clzll(emask));
+        }
+    }
+}
+
 // ---------------- the scan+aggregate kernel ----------------
 
 enum {
@@ -809,7 +904,8 @@ __global__ __launch_bounds__(256) void k_scan_agg(
     const uint8_t *__restrict__ payload, const bydb_block_desc *__restrict__ blocks,
     int64_t n_blocks, int64_t min_ts, int64_t max_ts, int flags,
     const PredBlock *__restrict__ preds, int n_preds,
-    bydb_partial *__restrict__ partials, DevErr *derr) {
+    const SegEntry *__restrict__ segs, bydb_partial *__restrict__ partials,
+    DevErr *derr) {
     const int lane = threadIdx.x & 63;
     const int wave_in_block = threadIdx.x >> 6;
     int64_t wave_id = (int64_t)blockIdx.x * (blockDim.x >> 6) + wave_in_block;
@@ -821,10 +917,19 @@ __global__ __launch_bounds__(256) void k_scan_agg(
     int64_t wmin = INT64_MAX, wmax = INT64_MIN;
     double wsumf = 0.0;
 
-    for (int64_t bi = wave_id; bi < n_blocks; bi += n_waves) {
+    // with a segment index, the work item is a (block, segment) pair —
+    // 8x the parallel units on the value-scan path, no serial chain
+    const int64_t n_items = segs ? n_blocks * MAX_SEGS : n_blocks;
+    for (int64_t wi = wave_id; wi < n_items; wi += n_waves) {
+        const int64_t bi = segs ? wi / MAX_SEGS : wi;
+        const int seg = segs ? (int)(wi % MAX_SEGS) : 0;
+        const bool seg_eligible =
+            segs && segs[bi * MAX_SEGS].byte_off != SEG_INELIGIBLE;
+        if (seg != 0 && !seg_eligible) continue;
         // wave-uniform descriptor loads
         const bydb_block_desc *bd = &blocks[bi];
         const int64_t n = (int64_t)bd->count;
+        if (seg != 0 && (int64_t)seg * SEG_ROWS + 1 > n - 1) continue;
         const int64_t ts_min = bd->ts_min, ts_max = bd->ts_max;
 
         // ---- row clamp (timestamp.FindRange, range.go:143-170) ----
@@ -911,6 +1016,11 @@ __global__ __launch_bounds__(256) void k_scan_agg(
             if (skip_block) continue;
         }
         bool pred_on = wp0 || wp1 || wp2;
+        const bool scan_path = (flags & KF_NEED_VALUES) &&
+                               bd->field_enc == BYDB_ENC_DELTA &&
+                               bd->field_len != (uint64_t)(n - 1);
+        const bool use_seg = seg_eligible && scan_path && !pred_on;
+        if (!use_seg && seg != 0) continue;
 
         // ---- field fold ----
         const uint8_t fenc = bd->field_enc;
@@ -1035,6 +1145,45 @@ __global__ __launch_bounds__(256) void k_scan_agg(
                         bsum = (uint64_t)first * nsel + (uint64_t)d1 * si + acc;
                     }
                 }
+            } else if (!dod && use_seg) {
+                // segment-parallel value scan: this wave folds only rows
+                // [seg*SEG_ROWS+1 .. min((seg+1)*SEG_ROWS, n-1)] (+ row 0
+                // on segment 0), starting from the indexed byte offset and
+                // carry value — no cross-segment dependency
+                const SegEntry e = segs[bi * MAX_SEGS + seg];
+                int64_t sj_lo = (int64_t)seg * SEG_ROWS + 1;
+                int64_t sj_hi = ((int64_t)seg + 1) * SEG_ROWS;
+                if (sj_hi > n - 1) sj_hi = n - 1;
+                int64_t a = sj_lo > r0 ? sj_lo : r0;
+                int64_t b = sj_hi < r1 ? sj_hi : r1;
+                bool handle0 = seg == 0 && r0 <= 0 && 0 <= r1;
+                if (a > b && !handle0) continue;
+                uint64_t lsum = 0, lcnt = 0;
+                int64_t lmn = INT64_MAX, lmx = INT64_MIN;
+                if (a <= b) {
+                    ScanFold ff;
+                    int64_t rel0 = (int64_t)seg * SEG_ROWS;
+                    scan_stream(fstream + e.byte_off, b - rel0, false,
+                                e.v_start, 0, a - rel0, b - rel0, INT64_MAX,
+                                INT64_MIN, lane, &ff, derr, (uint64_t)bi,
+                                nullptr, nullptr, nullptr);
+                    lsum = ff.sum;
+                    lcnt = ff.nsel;
+                    lmn = ff.mn;
+                    lmx = ff.mx;
+                }
+                if (lane == 0 && handle0) {
+                    lsum += (uint64_t)first;
+                    lcnt++;
+                    lmn = first < lmn ? first : lmn;
+                    lmx = first > lmx ? first : lmx;
+                }
+                bsum = wave_reduce_add(lsum);
+                nsel_eff = wave_reduce_add(lcnt);
+                bmin = wave_reduce_min(lmn);
+                bmax = wave_reduce_max(lmx);
+                have_minmax = nsel_eff > 0;
+                bsum = lane == 0 ? bsum : 0;
             } else if (!dod && bd->field_len == (uint64_t)(n - 1)) {
                 // all-1-byte delta stream: dense weighted sum + dense
                 // min/max (no serial window chain)
@@ -1153,6 +1302,9 @@ struct bydb_session {
     DevErr *d_err = nullptr;
     PredBlock *d_preds = nullptr;
     int64_t preds_cap = 0;
+    SegEntry *d_segs = nullptr;
+    int64_t segs_cap = 0;
+    bool segs_built = false;
     uint8_t *d_pred_bytes = nullptr;
     uint64_t pred_bytes_cap = 0;
     float last_ms = 0.0f;
@@ -1203,6 +1355,7 @@ extern "C" void bydb_session_destroy(bydb_session *s) {
     if (s->d_partials) hipFree(s->d_partials);
     if (s->d_preds) hipFree(s->d_preds);
     if (s->d_pred_bytes) hipFree(s->d_pred_bytes);
+    if (s->d_segs) hipFree(s->d_segs);
     if (s->d_err) hipFree(s->d_err);
     if (s->ev_start) hipEventDestroy(s->ev_start);
     if (s->ev_stop) hipEventDestroy(s->ev_stop);
@@ -1228,6 +1381,7 @@ extern "C" int bydb_part_reserve(bydb_session *s, uint64_t payload_bytes,
     s->blocks_cap = n_blocks;
     s->payload_len = 0;
     s->n_blocks = 0;
+    s->segs_built = false;
     return BYDB_OK;
 }
 
@@ -1235,6 +1389,7 @@ extern "C" int bydb_part_append(bydb_session *s, const uint8_t *payload,
                                 uint64_t len, const bydb_block_desc *blocks,
                                 int64_t n_blocks) {
     HIP_TRY(s, hipSetDevice(s->device));
+    s->segs_built = false;
     if (s->payload_len + len > s->payload_cap ||
         s->n_blocks + n_blocks > s->blocks_cap) {
         s->err = "part_append exceeds reservation";
@@ -1352,14 +1507,35 @@ extern "C" int bydb_consume_multi(bydb_session *s, int64_t min_ts,
         flags |= KF_NEED_VALUES;
     if (s->field_vtype == BYDB_VT_FLOAT64) flags |= KF_FLOAT;
     const int threads = 256;                       // 4 waves per workgroup
-    int64_t waves_needed = s->n_blocks;
+    SegEntry *segs = nullptr;
+    if (flags & KF_NEED_VALUES) {
+        // build (once per part) and use the varint segment index
+        if (s->segs_cap < s->n_blocks * MAX_SEGS) {
+            if (s->d_segs) hipFree(s->d_segs);
+            HIP_TRY(s, hipMalloc(&s->d_segs, sizeof(SegEntry) *
+                                                 (size_t)s->n_blocks * MAX_SEGS));
+            s->segs_cap = s->n_blocks * MAX_SEGS;
+            s->segs_built = false;
+        }
+        if (!s->segs_built) {
+            int64_t wgs_b = (s->n_blocks + 3) / 4;
+            int grid_b = (int)(wgs_b < 8192 ? wgs_b : 8192);
+            hipLaunchKernelGGL(k_build_seg_index, dim3(grid_b), dim3(threads),
+                               0, s->stream, s->d_payload, s->d_blocks,
+                               s->n_blocks, s->d_segs, s->d_err);
+            HIP_TRY(s, hipGetLastError());
+            s->segs_built = true;
+        }
+        segs = s->d_segs;
+    }
+    int64_t waves_needed = segs ? s->n_blocks * MAX_SEGS : s->n_blocks;
     int64_t wgs = (waves_needed + 3) / 4;
     int grid = (int)(wgs < 8192 ? wgs : 8192);     // grid-stride beyond
     if (grid < 1) grid = 1;
     HIP_TRY(s, hipEventRecord(s->ev_start, s->stream));
     hipLaunchKernelGGL(k_scan_agg, dim3(grid), dim3(threads), 0, s->stream,
                        s->d_payload, s->d_blocks, s->n_blocks, min_ts, max_ts,
-                       flags, preds, n_preds, s->d_acc, s->d_err);
+                       flags, preds, n_preds, segs, s->d_acc, s->d_err);
     HIP_TRY(s, hipGetLastError());
     HIP_TRY(s, hipEventRecord(s->ev_stop, s->stream));
     s->consumed = true;
