@@ -95,7 +95,7 @@ def main():
                          enc_bpd=1.05),
         "group4096_pred3": dict(float=False, n_groups=4096, group_mod=4096,
                                 tags=[ENVS, REGIONS, SVCS],
-                                preds=[b"prod", b"r2", b"s7"],
+                                preds=[b"prod", b"r4", b"s4"],
                                 funcs=["sum", "count"], algo_bpd=19,
                                 dtype="int64",
                                 agg="sum+count groupby(4096) 3-tag filter",
