@@ -720,22 +720,25 @@ __device__ void scan_stream(const uint8_t *stream, int64_t n_deltas, bool dod,
             // second scan: v_j = v_carry + incl_scan(d1_j over values)
             uint64_t s2 = wave_incl_scan(d1j, lane);
             val = v_carry + s2;
-            // totals: number of values in window and their d1 sum
+            // carry advance = scans at the last CONSUMED terminator lane;
+            // the window is fully consumed except at the stream tail, where
+            // the nvals-th set bit is searched (once per stream)
             int nterm_all = __popcll(emask);
             int64_t nvals = (jmax - j + 1) < (int64_t)nterm_all
                                 ? (jmax - j + 1) : (int64_t)nterm_all;
-            // d1 total over consumed values = s2 at last consumed terminator;
-            // easier: v_carry advances by s2 at the last consumed value lane.
-            // find lane of the nvals-th terminator:
-            uint64_t mm = emask;
-            int last_lane = 0;
-            for (int t = 0; t < nvals; t++) {   // nvals <= 64; wave-uniform loop
-                last_lane = __ffsll((unsigned long long)mm) - 1;
-                mm &= mm - 1;
+            int last_lane;
+            if (nvals == (int64_t)nterm_all) {
+                last_lane = 63 - __clzll(emask);
+            } else {
+                uint64_t mm = emask;
+                last_lane = 0;
+                for (int t = 0; t < nvals; t++) {
+                    last_lane = __ffsll((unsigned long long)mm) - 1;
+                    mm &= mm - 1;
+                }
             }
             tot = (uint64_t)__shfl((long long)s2, last_lane);
             v_carry += tot;
-            // d1_carry advances by scan1 at last consumed value lane
             uint64_t d1tot = (uint64_t)__shfl((long long)s1, last_lane);
             d1_carry += d1tot;
             (void)s1_tot;
@@ -745,11 +748,16 @@ __device__ void scan_stream(const uint8_t *stream, int64_t n_deltas, bool dod,
             int nterm_all = __popcll(emask);
             int64_t nvals = (jmax - j + 1) < (int64_t)nterm_all
                                 ? (jmax - j + 1) : (int64_t)nterm_all;
-            uint64_t mm = emask;
-            int last_lane = 0;
-            for (int t = 0; t < nvals; t++) {
-                last_lane = __ffsll((unsigned long long)mm) - 1;
-                mm &= mm - 1;
+            int last_lane;
+            if (nvals == (int64_t)nterm_all) {
+                last_lane = 63 - __clzll(emask);
+            } else {
+                uint64_t mm = emask;
+                last_lane = 0;
+                for (int t = 0; t < nvals; t++) {
+                    last_lane = __ffsll((unsigned long long)mm) - 1;
+                    mm &= mm - 1;
+                }
             }
             tot = (uint64_t)__shfl((long long)s, last_lane);
             v_carry += tot;
@@ -865,11 +873,16 @@ __global__ __launch_bounds__(256) void k_build_seg_index(
             int nterm_all = __popcll(emask);
             int64_t nvals = (n_deltas - j + 1) < (int64_t)nterm_all
                                 ? (n_deltas - j + 1) : (int64_t)nterm_all;
-            uint64_t mm = emask;
-            int last_lane = 0;
-            for (int t = 0; t < nvals; t++) {
-                last_lane = __ffsll((unsigned long long)mm) - 1;
-                mm &= mm - 1;
+            int last_lane;
+            if (nvals == (int64_t)nterm_all) {
+                last_lane = 63 - __clzll(emask);
+            } else {
+                uint64_t mm = emask;
+                last_lane = 0;
+                for (int t = 0; t < nvals; t++) {
+                    last_lane = __ffsll((unsigned long long)mm) - 1;
+                    mm &= mm - 1;
+                }
             }
             v_carry += (uint64_t)__shfl((long long)s, last_lane);
             if (j + nterm_all > n_deltas) break;
