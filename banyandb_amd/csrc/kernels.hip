@@ -174,24 +174,21 @@ __device__ void dense_region(const uint8_t *stream, int64_t lo, int64_t hi,
         const uint32_t *p = (const uint32_t *)(s0 + (uint64_t)alo);
         int64_t ndw = (ahi - alo) / 4;
         int64_t q = lane;
-        // 4-deep manual batching: issue 4 lane-strided loads, then fold
-        for (; q + 192 < ndw; q += 256) {
-            uint32_t w0 = p[q], w1 = p[q + 64], w2 = p[q + 128], w3 = p[q + 192];
-            uint32_t s_0 = swar_zigzag(w0), s_1 = swar_zigzag(w1),
-                     s_2 = swar_zigzag(w2), s_3 = swar_zigzag(w3);
-            int32_t t0 = dot4_i8(s_0, 0x01010101u, 0);
-            int32_t t1 = dot4_i8(s_1, 0x01010101u, 0);
-            int32_t t2 = dot4_i8(s_2, 0x01010101u, 0);
-            int32_t t3 = dot4_i8(s_3, 0x01010101u, 0);
-            int32_t k0 = dot4_i8(s_0, 0x03020100u, 0);
-            int32_t k1 = dot4_i8(s_1, 0x03020100u, 0);
-            int32_t k2 = dot4_i8(s_2, 0x03020100u, 0);
-            int32_t k3 = dot4_i8(s_3, 0x03020100u, 0);
+        // 8-deep manual batching: issue 8 lane-strided loads (2 KiB per
+        // wave-batch), then fold — keeps 8 loads in flight per lane
+        for (; q + 448 < ndw; q += 512) {
+            uint32_t w[8];
+#pragma unroll
+            for (int t = 0; t < 8; t++) w[t] = p[q + 64 * t];
             int32_t b0 = (int32_t)(alo + 4 * q);
-            acc_d += t0 + t1 + t2 + t3;
-            acc_jd32 += k0 + k1 + k2 + k3;
-            acc_jd32 += b0 * t0 + (b0 + 256) * t1 + (b0 + 512) * t2 +
-                        (b0 + 768) * t3;
+#pragma unroll
+            for (int t = 0; t < 8; t++) {
+                uint32_t sv = swar_zigzag(w[t]);
+                int32_t tt = dot4_i8(sv, 0x01010101u, 0);
+                int32_t kk = dot4_i8(sv, 0x03020100u, 0);
+                acc_d += tt;
+                acc_jd32 += kk + (b0 + 256 * t) * tt;
+            }
         }
         for (; q < ndw; q += 64) {
             uint32_t w0 = p[q];
@@ -931,9 +928,14 @@ __global__ __launch_bounds__(256) void k_scan_agg(
     double wsumf = 0.0;
 
     // with a segment index, the work item is a (block, segment) pair —
-    // 8x the parallel units on the value-scan path, no serial chain
+    // 8x the parallel units on the value-scan path, no serial chain.
+    // Each wave takes a CONTIGUOUS item range: successive streams are
+    // adjacent in the payload, so a wave reads DRAM sequentially.
     const int64_t n_items = segs ? n_blocks * MAX_SEGS : n_blocks;
-    for (int64_t wi = wave_id; wi < n_items; wi += n_waves) {
+    const int64_t per_wave = (n_items + n_waves - 1) / n_waves;
+    const int64_t wi_lo = wave_id * per_wave;
+    const int64_t wi_hi = wi_lo + per_wave < n_items ? wi_lo + per_wave : n_items;
+    for (int64_t wi = wi_lo; wi < wi_hi; wi += 1) {
         const int64_t bi = segs ? wi / MAX_SEGS : wi;
         const int seg = segs ? (int)(wi % MAX_SEGS) : 0;
         const bool seg_eligible =
