@@ -929,13 +929,10 @@ __global__ __launch_bounds__(256) void k_scan_agg(
 
     // with a segment index, the work item is a (block, segment) pair —
     // 8x the parallel units on the value-scan path, no serial chain.
-    // Each wave takes a CONTIGUOUS item range: successive streams are
-    // adjacent in the payload, so a wave reads DRAM sequentially.
+    // Strided assignment (not contiguous ranges): predicates skip blocks
+    // in long runs, and contiguous ranges turn those runs into idle waves.
     const int64_t n_items = segs ? n_blocks * MAX_SEGS : n_blocks;
-    const int64_t per_wave = (n_items + n_waves - 1) / n_waves;
-    const int64_t wi_lo = wave_id * per_wave;
-    const int64_t wi_hi = wi_lo + per_wave < n_items ? wi_lo + per_wave : n_items;
-    for (int64_t wi = wi_lo; wi < wi_hi; wi += 1) {
+    for (int64_t wi = wave_id; wi < n_items; wi += n_waves) {
         const int64_t bi = segs ? wi / MAX_SEGS : wi;
         const int seg = segs ? (int)(wi % MAX_SEGS) : 0;
         const bool seg_eligible =
