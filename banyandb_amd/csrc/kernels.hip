@@ -174,21 +174,23 @@ __device__ void dense_region(const uint8_t *stream, int64_t lo, int64_t hi,
         const uint32_t *p = (const uint32_t *)(s0 + (uint64_t)alo);
         int64_t ndw = (ahi - alo) / 4;
         int64_t q = lane;
-        // 8-deep manual batching: issue 8 lane-strided loads (2 KiB per
-        // wave-batch), then fold — keeps 8 loads in flight per lane
-        for (; q + 448 < ndw; q += 512) {
-            uint32_t w[8];
-#pragma unroll
-            for (int t = 0; t < 8; t++) w[t] = p[q + 64 * t];
+        // 4-deep manual batching: issue 4 lane-strided loads, then fold
+        for (; q + 192 < ndw; q += 256) {
+            uint32_t w0 = p[q], w1 = p[q + 64], w2 = p[q + 128], w3 = p[q + 192];
             int32_t b0 = (int32_t)(alo + 4 * q);
-#pragma unroll
-            for (int t = 0; t < 8; t++) {
-                uint32_t sv = swar_zigzag(w[t]);
-                int32_t tt = dot4_i8(sv, 0x01010101u, 0);
-                int32_t kk = dot4_i8(sv, 0x03020100u, 0);
-                acc_d += tt;
-                acc_jd32 += kk + (b0 + 256 * t) * tt;
-            }
+            uint32_t s_0 = swar_zigzag(w0), s_1 = swar_zigzag(w1),
+                     s_2 = swar_zigzag(w2), s_3 = swar_zigzag(w3);
+            int32_t t0 = dot4_i8(s_0, 0x01010101u, 0);
+            int32_t t1 = dot4_i8(s_1, 0x01010101u, 0);
+            int32_t t2 = dot4_i8(s_2, 0x01010101u, 0);
+            int32_t t3 = dot4_i8(s_3, 0x01010101u, 0);
+            acc_d += t0 + t1 + t2 + t3;
+            acc_jd32 += dot4_i8(s_0, 0x03020100u, 0) +
+                        dot4_i8(s_1, 0x03020100u, 0) +
+                        dot4_i8(s_2, 0x03020100u, 0) +
+                        dot4_i8(s_3, 0x03020100u, 0);
+            acc_jd32 += b0 * t0 + (b0 + 256) * t1 + (b0 + 512) * t2 +
+                        (b0 + 768) * t3;
         }
         for (; q < ndw; q += 64) {
             uint32_t w0 = p[q];
@@ -910,12 +912,19 @@ __device__ __forceinline__ void flush_partial(bydb_partial *partials,
     if (wsumf != 0.0) atomicAdd(&p->sum_f, wsumf);
 }
 
-__global__ __launch_bounds__(256) void k_scan_agg(
+// Compile-time specialization: the FAST instantiation (no value
+// reconstruction, no predicates) sheds the scan/walker machinery and its
+// register pressure; the full one keeps everything.  (EN_* are constant
+// guards — dead branches are eliminated per instantiation.)
+template <bool EN_VALUES, bool EN_PREDS>
+__global__ __launch_bounds__(256) void k_scan_agg_t(
     const uint8_t *__restrict__ payload, const bydb_block_desc *__restrict__ blocks,
     int64_t n_blocks, int64_t min_ts, int64_t max_ts, int flags,
-    const PredBlock *__restrict__ preds, int n_preds,
-    const SegEntry *__restrict__ segs, bydb_partial *__restrict__ partials,
+    const PredBlock *__restrict__ preds_in, int n_preds,
+    const SegEntry *__restrict__ segs_in, bydb_partial *__restrict__ partials,
     DevErr *derr) {
+    const PredBlock *preds = EN_PREDS ? preds_in : nullptr;
+    const SegEntry *segs = EN_VALUES ? segs_in : nullptr;
     const int lane = threadIdx.x & 63;
     const int wave_in_block = threadIdx.x >> 6;
     int64_t wave_id = (int64_t)blockIdx.x * (blockDim.x >> 6) + wave_in_block;
@@ -1028,7 +1037,7 @@ __global__ __launch_bounds__(256) void k_scan_agg(
             if (skip_block) continue;
         }
         bool pred_on = wp0 || wp1 || wp2;
-        const bool scan_path = (flags & KF_NEED_VALUES) &&
+        const bool scan_path = EN_VALUES && (flags & KF_NEED_VALUES) &&
                                bd->field_enc == BYDB_ENC_DELTA &&
                                bd->field_len != (uint64_t)(n - 1);
         const bool use_seg = seg_eligible && scan_path && !pred_on;
@@ -1054,7 +1063,7 @@ __global__ __launch_bounds__(256) void k_scan_agg(
                             &pcnt, &bmin, &bmax);
             bsum = lane == 0 ? psum : 0;
             nsel_eff = pcnt;
-            have_minmax = (flags & KF_NEED_VALUES) && pcnt > 0;
+            have_minmax = (EN_VALUES && (flags & KF_NEED_VALUES)) && pcnt > 0;
         } else if (pred_on &&
                    (fenc == BYDB_ENC_DELTA || fenc == BYDB_ENC_DELTA_OF_DELTA)) {
             bool dod = fenc == BYDB_ENC_DELTA_OF_DELTA;
@@ -1111,7 +1120,7 @@ __global__ __launch_bounds__(256) void k_scan_agg(
                 uint64_t si = (uint64_t)(r0 + r1) * nsel / 2;
                 bsum = (uint64_t)first * nsel + (uint64_t)dd * si;
             }
-            if (flags & KF_NEED_VALUES) {
+            if (EN_VALUES && (flags & KF_NEED_VALUES)) {
                 // reconstruct min/max exactly (wrap-safe): walk rows by lanes
                 int64_t lmn = INT64_MAX, lmx = INT64_MIN;
                 for (int64_t base = r0; base <= r1; base += WAVE) {
@@ -1128,7 +1137,7 @@ __global__ __launch_bounds__(256) void k_scan_agg(
             }
         } else if (fenc == BYDB_ENC_DELTA || fenc == BYDB_ENC_DELTA_OF_DELTA) {
             bool dod = fenc == BYDB_ENC_DELTA_OF_DELTA;
-            if (!(flags & KF_NEED_VALUES)) {
+            if (!(EN_VALUES && (flags & KF_NEED_VALUES))) {
                 uint64_t acc;
                 if (!dod) {
                     if (bd->field_len == (uint64_t)(n - 1)) {
@@ -1545,7 +1554,13 @@ extern "C" int bydb_consume_multi(bydb_session *s, int64_t min_ts,
     int grid = (int)(wgs < 8192 ? wgs : 8192);     // grid-stride beyond
     if (grid < 1) grid = 1;
     HIP_TRY(s, hipEventRecord(s->ev_start, s->stream));
-    hipLaunchKernelGGL(k_scan_agg, dim3(grid), dim3(threads), 0, s->stream,
+    const bool en_values = (flags & KF_NEED_VALUES) != 0;
+    const bool en_preds = n_preds > 0;
+    auto kfn = en_values ? (en_preds ? k_scan_agg_t<true, true>
+                                     : k_scan_agg_t<true, false>)
+                         : (en_preds ? k_scan_agg_t<false, true>
+                                     : k_scan_agg_t<false, false>);
+    hipLaunchKernelGGL(kfn, dim3(grid), dim3(threads), 0, s->stream,
                        s->d_payload, s->d_blocks, s->n_blocks, min_ts, max_ts,
                        flags, preds, n_preds, segs, s->d_acc, s->d_err);
     HIP_TRY(s, hipGetLastError());
