@@ -693,6 +693,38 @@ __device__ void scan_stream(const uint8_t *stream, int64_t n_deltas, bool dod,
         int rank = __popcll(emask & lanemask_lt(lane));
         int64_t myj = j + rank;
         bool is_term = (b < 0x80) && (myj <= jmax);
+        // fast window: every varint is 1 or 2 bytes (no two consecutive
+        // continuation bits) and the whole window is consumed — decode via
+        // the left neighbour's byte (shfl) and scan in 32-bit (window
+        // delta sums fit: |d| <= 2^13 per value)
+        uint64_t cont = ~emask;
+        if (!dod && pw0 == nullptr && pw1 == nullptr && pw2 == nullptr &&
+            (cont & (cont << 1)) == 0 && (int64_t)__popcll(emask) <= jmax - j + 1) {
+            uint32_t prev = (uint32_t)__shfl_up((int)b, 1);
+            bool is2 = lane > 0 && (prev & 0x80u);
+            uint32_t u = is2 ? ((prev & 0x7fu) | ((uint32_t)b << 7)) : b;
+            int32_t d32 = (int32_t)(u >> 1) ^ -(int32_t)(u & 1);
+            if (!(b < 0x80)) d32 = 0;          // continuation lanes carry 0
+            int32_t s32 = wave_incl_scan32(d32, lane);
+            if (b < 0x80) {
+                int64_t sv = (int64_t)(v_carry + (uint64_t)(int64_t)s32);
+                if (myj >= r0 && myj <= r1) {
+                    l_sum += (uint64_t)sv;
+                    l_nsel++;
+                    l_mn = sv < l_mn ? sv : l_mn;
+                    l_mx = sv > l_mx ? sv : l_mx;
+                }
+                if (sv < lo_bound) l_nlo++;
+                if (sv > hi_bound) l_nhi++;
+            }
+            int last_lane = 63 - __clzll(emask);
+            v_carry += (uint64_t)(int64_t)__shfl(s32, last_lane);
+            int nterm = __popcll(emask);
+            if (j + nterm > jmax) break;
+            j += nterm;
+            pos += (uint64_t)(last_lane + 1);
+            continue;
+        }
         uint64_t d = 0;
         if (emask == ~0ull) {
             d = (uint64_t)zz_dec(b);
@@ -917,7 +949,7 @@ __device__ __forceinline__ void flush_partial(bydb_partial *partials,
 // register pressure; the full one keeps everything.  (EN_* are constant
 // guards — dead branches are eliminated per instantiation.)
 template <bool EN_VALUES, bool EN_PREDS>
-__global__ __launch_bounds__(256) void k_scan_agg_t(
+__global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS) ? 4 : 6) void k_scan_agg_t(
     const uint8_t *__restrict__ payload, const bydb_block_desc *__restrict__ blocks,
     int64_t n_blocks, int64_t min_ts, int64_t max_ts, int flags,
     const PredBlock *__restrict__ preds_in, int n_preds,
