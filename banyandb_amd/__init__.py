@@ -175,6 +175,11 @@ def _load():
                                              C.c_int64, C.c_int64, C.c_int64,
                                              C.c_int64, C.c_int64, C.c_uint64,
                                              C.c_uint32, C.c_int]
+    lib.bydb_part_write_dir.restype = C.c_int
+    lib.bydb_part_write_dir.argtypes = [C.c_void_p, C.c_char_p, C.c_char_p,
+                                        C.c_char_p, C.POINTER(C.c_char_p), C.c_int]
+    lib.bydb_part_read_dir.restype = C.c_int
+    lib.bydb_part_read_dir.argtypes = [C.c_void_p, C.c_char_p]
     lib.bydb_gen_series_bulk_f64.restype = C.c_int
     lib.bydb_gen_series_bulk_f64.argtypes = [C.c_void_p, C.c_uint64, C.c_int64,
                                              C.c_int64, C.c_int64, C.c_int64,
@@ -282,6 +287,20 @@ class PartBuilder:
     def blocks(self):
         ptr = _lib.bydb_part_builder_blocks(self._h)
         return [ptr[i] for i in range(self.n_blocks)]
+
+    def write_dir(self, path, field_name="value", tag_family="default",
+                  tag_names=()):
+        """Write the part as a reference-layout on-disk part directory."""
+        arr = (C.c_char_p * max(len(tag_names), 1))(
+            *[t.encode() for t in tag_names] or [b""])
+        self._ck(_lib.bydb_part_write_dir(self._h, path.encode(),
+                                          field_name.encode(),
+                                          tag_family.encode(), arr,
+                                          len(tag_names)))
+
+    def read_dir(self, path):
+        """Load a part directory written by write_dir (or the reference)."""
+        self._ck(_lib.bydb_part_read_dir(self._h, path.encode()))
 
     def drain(self):
         self._ck(_lib.bydb_part_builder_drain(self._h))

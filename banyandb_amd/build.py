@@ -16,7 +16,8 @@ OUT = os.path.join(PKG, "libbydb_gpu.so")
 def build(verbose=True):
     hipcc = os.environ.get("HIPCC", "/opt/rocm/bin/hipcc")
     srcs = [os.path.join(CSRC, "kernels.hip"), os.path.join(CSRC, "encode.cpp"),
-            os.path.join(CSRC, "frame.cpp")]
+            os.path.join(CSRC, "frame.cpp"),
+            os.path.join(CSRC, "part_io.cpp")]
     cmd = [
         hipcc, "--offload-arch=gfx950", "-O3", "-std=c++17", "-fPIC",
         "-shared", "-o", OUT, *srcs, "-ldl",

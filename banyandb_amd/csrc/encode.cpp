@@ -542,6 +542,20 @@ extern "C" int64_t bydb_part_builder_n_blocks(bydb_part_builder *b) {
 extern "C" const bydb_block_desc *bydb_part_builder_blocks(bydb_part_builder *b) {
     return b->blocks.data();
 }
+// raw helpers for the part reader (part_io.cpp): append payload bytes /
+// a pre-built descriptor with offsets relative to the whole part
+extern "C" int bydb_part_builder_append_raw(bydb_part_builder *b,
+                                            const uint8_t *data, uint64_t len) {
+    b->payload.insert(b->payload.end(), data, data + len);
+    return BYDB_OK;
+}
+
+extern "C" int bydb_part_builder_append_desc(bydb_part_builder *b,
+                                             const bydb_block_desc *d) {
+    b->blocks.push_back(*d);
+    return BYDB_OK;
+}
+
 extern "C" int bydb_part_builder_drain(bydb_part_builder *b) {
     b->base_off += b->payload.size();
     b->payload.clear();
