@@ -383,6 +383,30 @@ __device__ uint64_t fold_delta_weighted_fast(const uint8_t *stream,
             int rank = __popcll(emask & lanemask_lt(lane));
             int64_t myj = j + rank;
             bool is_term = (b < 0x80) && (myj <= jend);
+            // <=2-byte window fast case: decode via the left neighbour,
+            // no backward byte loop, no 64-bit bookkeeping
+            uint64_t cont2 = ~emask;
+            if ((cont2 & (cont2 << 1)) == 0 &&
+                (int64_t)__popcll(emask) <= jend - j + 1) {
+                uint32_t prevb = (uint32_t)__shfl_up((int)b, 1);
+                bool is2b = lane > 0 && (prevb & 0x80u);
+                uint32_t u2 = is2b ? ((prevb & 0x7fu) | ((uint32_t)b << 7))
+                                   : (uint32_t)b;
+                int32_t d32 = (int32_t)(u2 >> 1) ^ -(int32_t)(u2 & 1);
+                if (b < 0x80) {
+                    uint64_t wt = myj <= r0 ? nsel : (uint64_t)(r1 - myj + 1);
+                    gen_acc += (uint64_t)(int64_t)d32 * wt;
+                }
+                int ll2 = 63 - __clzll(emask);
+                int nterm2 = __popcll(emask);
+                if (j + nterm2 > jend) { j = jend + 1; break; }
+                j += nterm2;
+                pos += (uint64_t)(ll2 + 1);
+                wp = (const uint32_t *)((uintptr_t)(stream + pos) & ~(uintptr_t)3);
+                off = (unsigned)((uintptr_t)(stream + pos) - (uintptr_t)wp);
+                w = wp[lane];
+                continue;
+            }
             int64_t d;
             if (emask == ~0ull) {
                 d = zz_dec(b);
