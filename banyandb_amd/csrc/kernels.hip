@@ -84,12 +84,6 @@ __device__ __forceinline__ int64_t wave_reduce_max(int64_t v) {
     return v;
 }
 
-__device__ __forceinline__ double wave_reduce_addf(double v) {
-#pragma unroll
-    for (int off = 32; off > 0; off >>= 1) v += __shfl_xor(v, off);
-    return v;
-}
-
 // Decode one varint serially starting at p (used for wave-uniform headers:
 // the DeltaConst stride and the delta-of-delta d1 — every lane runs the
 // same loop on the same bytes, so the result is wave-uniform).
@@ -1420,23 +1414,23 @@ extern "C" bydb_session *bydb_session_create(int device) {
         delete s;
         return nullptr;
     }
-    hipMemset(s->d_err, 0, sizeof(DevErr));
+    (void)hipMemset(s->d_err, 0, sizeof(DevErr));
     return s;
 }
 
 extern "C" void bydb_session_destroy(bydb_session *s) {
     if (!s) return;
-    hipSetDevice(s->device);
-    if (s->d_payload) hipFree(s->d_payload);
-    if (s->d_blocks) hipFree(s->d_blocks);
-    if (s->d_partials) hipFree(s->d_partials);
-    if (s->d_preds) hipFree(s->d_preds);
-    if (s->d_pred_bytes) hipFree(s->d_pred_bytes);
-    if (s->d_segs) hipFree(s->d_segs);
-    if (s->d_err) hipFree(s->d_err);
-    if (s->ev_start) hipEventDestroy(s->ev_start);
-    if (s->ev_stop) hipEventDestroy(s->ev_stop);
-    if (s->stream) hipStreamDestroy(s->stream);
+    (void)hipSetDevice(s->device);
+    if (s->d_payload) (void)hipFree(s->d_payload);
+    if (s->d_blocks) (void)hipFree(s->d_blocks);
+    if (s->d_partials) (void)hipFree(s->d_partials);
+    if (s->d_preds) (void)hipFree(s->d_preds);
+    if (s->d_pred_bytes) (void)hipFree(s->d_pred_bytes);
+    if (s->d_segs) (void)hipFree(s->d_segs);
+    if (s->d_err) (void)hipFree(s->d_err);
+    if (s->ev_start) (void)hipEventDestroy(s->ev_start);
+    if (s->ev_stop) (void)hipEventDestroy(s->ev_stop);
+    if (s->stream) (void)hipStreamDestroy(s->stream);
     delete s;
 }
 
@@ -1447,8 +1441,8 @@ extern "C" const char *bydb_last_error(bydb_session *s) {
 extern "C" int bydb_part_reserve(bydb_session *s, uint64_t payload_bytes,
                                  int64_t n_blocks) {
     HIP_TRY(s, hipSetDevice(s->device));
-    if (s->d_payload) { hipFree(s->d_payload); s->d_payload = nullptr; }
-    if (s->d_blocks) { hipFree(s->d_blocks); s->d_blocks = nullptr; }
+    if (s->d_payload) { (void)hipFree(s->d_payload); s->d_payload = nullptr; }
+    if (s->d_blocks) { (void)hipFree(s->d_blocks); s->d_blocks = nullptr; }
     // +1KiB slack: the 256-byte window loop prefetches one window ahead
     // and may read past the last stream's end
     HIP_TRY(s, hipMalloc(&s->d_payload, payload_bytes + 1024));
@@ -1498,7 +1492,7 @@ extern "C" int bydb_agg_configure(bydb_session *s, int field_vtype,
     s->n_groups = n_groups;
     s->mode = mode;
     if (s->partials_cap < n_groups) {
-        if (s->d_partials) hipFree(s->d_partials);
+        if (s->d_partials) (void)hipFree(s->d_partials);
         HIP_TRY(s, hipMalloc(&s->d_partials, sizeof(bydb_partial) * n_groups));
         s->partials_cap = n_groups;
     }
@@ -1551,7 +1545,7 @@ extern "C" int bydb_consume_multi(bydb_session *s, int64_t min_ts,
             }
         }
         if (s->preds_cap < s->n_blocks * 3) {
-            if (s->d_preds) hipFree(s->d_preds);
+            if (s->d_preds) (void)hipFree(s->d_preds);
             HIP_TRY(s, hipMalloc(&s->d_preds,
                                  sizeof(PredBlock) * (size_t)s->n_blocks * 3));
             s->preds_cap = s->n_blocks * 3;
@@ -1559,7 +1553,7 @@ extern "C" int bydb_consume_multi(bydb_session *s, int64_t min_ts,
         uint64_t total = 0;
         for (int i = 0; i < n_preds; i++) total += pred_lens[i];
         if (s->pred_bytes_cap < total) {
-            if (s->d_pred_bytes) hipFree(s->d_pred_bytes);
+            if (s->d_pred_bytes) (void)hipFree(s->d_pred_bytes);
             HIP_TRY(s, hipMalloc(&s->d_pred_bytes, total));
             s->pred_bytes_cap = total;
         }
@@ -1588,7 +1582,7 @@ extern "C" int bydb_consume_multi(bydb_session *s, int64_t min_ts,
     if (flags & KF_NEED_VALUES) {
         // build (once per part) and use the varint segment index
         if (s->segs_cap < s->n_blocks * MAX_SEGS) {
-            if (s->d_segs) hipFree(s->d_segs);
+            if (s->d_segs) (void)hipFree(s->d_segs);
             HIP_TRY(s, hipMalloc(&s->d_segs, sizeof(SegEntry) *
                                                  (size_t)s->n_blocks * MAX_SEGS));
             s->segs_cap = s->n_blocks * MAX_SEGS;
