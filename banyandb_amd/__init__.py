@@ -115,6 +115,11 @@ def _load():
     lib.bydb_part_append.argtypes = [C.c_void_p, u8p, C.c_uint64, bp, C.c_int64]
     lib.bydb_agg_configure.restype = C.c_int
     lib.bydb_agg_configure.argtypes = [C.c_void_p, C.c_int, C.c_uint32, C.c_uint32, C.c_int]
+    lib.bydb_agg_configure_by_tag.restype = C.c_int
+    lib.bydb_agg_configure_by_tag.argtypes = [C.c_void_p, C.c_int, C.c_uint32,
+                                              C.c_int, u8p,
+                                              C.POINTER(C.c_uint64), C.c_uint32,
+                                              C.c_int]
     lib.bydb_set_partials_buffer.restype = C.c_int
     lib.bydb_set_partials_buffer.argtypes = [C.c_void_p, C.c_void_p, C.c_uint64]
     lib.bydb_reset.restype = C.c_int
@@ -354,6 +359,25 @@ class Session:
         self.n_groups = n_groups
         self.field_vtype = field_vtype
         self._ck(_lib.bydb_agg_configure(self._h, field_vtype, mask, n_groups, mode))
+        self._ck(_lib.bydb_set_float_exp(self._h, float_exp))
+
+    def configure_by_tag(self, field_vtype, funcs, tag_slot, domain,
+                         mode=MODE_ALL, float_exp=0):
+        """Per-row group-by on the dictionary tag in tag_slot; group id =
+        index into domain (list of bytes values)."""
+        mask = 0
+        for f in funcs:
+            mask |= 1 << f
+        blob = b"".join(domain)
+        offs = [0]
+        for v in domain:
+            offs.append(offs[-1] + len(v))
+        self.n_groups = len(domain)
+        self.field_vtype = field_vtype
+        buf = (C.c_uint8 * max(len(blob), 1)).from_buffer_copy(blob or b"\0")
+        self._ck(_lib.bydb_agg_configure_by_tag(
+            self._h, field_vtype, mask, tag_slot, buf,
+            (C.c_uint64 * len(offs))(*offs), len(domain), mode))
         self._ck(_lib.bydb_set_float_exp(self._h, float_exp))
 
     def set_partials_buffer(self, dev_ptr, nbytes):

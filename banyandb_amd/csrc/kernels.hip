@@ -570,6 +570,145 @@ __device__ __forceinline__ bool pred_match_rows(PredWalk *pw, int64_t row,
     return match;
 }
 
+// ---- per-row group-by on a dictionary tag ----
+// The reference groups rows by the encoded tag-value key, materialising
+// groups in first-seen order (computeKey/appendKeyComponent,
+// vectorized/measure/aggregation.go:523, groupby.go:287-364).  Here the
+// host supplies the group DOMAIN (value -> dense group id); a resolve
+// prepass maps each block's dictionary codes to group ids — per-block
+// uniform tags (entity tags, the common case) collapse to a single id and
+// keep the fast fold paths; row-varying tags fold per RLE run.
+struct GroupDomain {
+    const uint8_t *blob;     // concatenated domain value bytes
+    const uint64_t *offs;    // n+1 offsets into blob
+    const uint64_t *hashes;  // open-addressing table: slot -> hash
+    const uint32_t *gids;    // slot -> gid (0xFFFFFFFF empty)
+    uint32_t table_size;     // power of two
+    uint32_t n;
+};
+
+struct GroupBlock {
+    uint32_t uniform_gid;   // gid when the whole block is one run;
+                            // 0xFFFFFFFE = row-varying, 0xFFFFFFFF = no tag
+    uint32_t map_off;       // offset into the code->gid arena (row-varying)
+    uint64_t rle_bit_off;
+    uint32_t nentries;
+    uint8_t width;
+    uint8_t err;
+    uint8_t _p[2];
+};
+
+#define GID_NONE 0xFFFFFFFFu
+#define GID_VARYING 0xFFFFFFFEu
+
+// FNV-1a 64 over value bytes (hash choice is internal; equality is by
+// bytes, mirroring the reference's "equality is on key bytes" note)
+__device__ __host__ inline uint64_t fnv1a(const uint8_t *p, uint64_t n) {
+    uint64_t h = 1469598103934665603ULL;
+    for (uint64_t i = 0; i < n; i++) {
+        h ^= p[i];
+        h *= 1099511628211ULL;
+    }
+    return h;
+}
+
+__device__ inline uint32_t domain_lookup(const GroupDomain *d, const uint8_t *v,
+                                         uint64_t len) {
+    uint64_t h = fnv1a(v, len);
+    uint32_t mask = d->table_size - 1;
+    uint32_t slot = (uint32_t)h & mask;
+    for (uint32_t probe = 0; probe <= mask; probe++) {
+        uint32_t g = d->gids[slot];
+        if (g == GID_NONE) return GID_NONE;
+        if (d->hashes[slot] == h) {
+            uint64_t lo = d->offs[g], hi = d->offs[g + 1];
+            if (hi - lo == len) {
+                bool eq = true;
+                for (uint64_t k = 0; k < len; k++)
+                    if (d->blob[lo + k] != v[k]) { eq = false; break; }
+                if (eq) return g;
+            }
+        }
+        slot = (slot + 1) & mask;
+    }
+    return GID_NONE;
+}
+
+// one thread per block: parse the tag dictionary, map codes to gids
+__global__ void k_resolve_groups(const uint8_t *__restrict__ payload,
+                                 const bydb_block_desc *__restrict__ blocks,
+                                 int64_t n_blocks, int slot, GroupDomain dom,
+                                 GroupBlock *__restrict__ out,
+                                 uint16_t *__restrict__ map_arena) {
+    int64_t bi = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (bi >= n_blocks) return;
+    const bydb_block_desc *bd = &blocks[bi];
+    uint64_t toff = slot == 0 ? bd->tag_off : slot == 1 ? bd->tag2_off : bd->tag3_off;
+    uint64_t tlen = slot == 0 ? bd->tag_len : slot == 1 ? bd->tag2_len : bd->tag3_len;
+    GroupBlock gb;
+    gb.uniform_gid = GID_NONE;
+    gb.map_off = (uint32_t)(bi * 256);
+    gb.rle_bit_off = 0; gb.nentries = 0; gb.width = 0; gb.err = 0;
+    gb._p[0] = gb._p[1] = 0;
+    if (tlen == 0) { out[bi] = gb; return; }
+    const uint8_t *p = payload + toff;
+    const uint8_t *end = p + tlen;
+    if (*p != BYDB_ENC_DICTIONARY) { gb.err = 1; out[bi] = gb; return; }
+    p++;
+    uint64_t count = 0;
+    unsigned sh = 0;
+    while (p < end) {
+        uint8_t c = *p++;
+        count |= (uint64_t)(c & 0x7f) << sh;
+        if (c < 0x80) break;
+        sh += 7;
+    }
+    if (p >= end || *p != 0) { gb.err = 1; out[bi] = gb; return; }
+    p++;
+    uint64_t ll = *p++;
+    const uint8_t *lens_blk = p;
+    p += ll;
+    if (p >= end || *p != 0) { gb.err = 1; out[bi] = gb; return; }
+    p++;
+    uint64_t vl = *p++;
+    const uint8_t *vals = p;
+    p += vl;
+    uint8_t wt = lens_blk[0];
+    uint32_t wbytes = wt == 0 ? 1 : wt == 1 ? 2 : wt == 2 ? 4 : 8;
+    const uint8_t *lp = lens_blk + 1;
+    uint64_t voff = 0;
+    uint16_t *map = map_arena + gb.map_off;
+    for (uint64_t v = 0; v < count && v < 256; v++) {
+        uint64_t alen = 0;
+        for (uint32_t b = 0; b < wbytes; b++) alen = (alen << 8) | lp[v * wbytes + b];
+        uint32_t g = GID_NONE;
+        if (alen > 0) {
+            uint64_t vlen = alen - 1;
+            g = domain_lookup(&dom, vals + voff, vlen);
+            voff += vlen;
+        }
+        map[v] = g == GID_NONE ? 0xFFFFu : (uint16_t)g;
+    }
+    uint64_t bit0 = ((uint64_t)(p - payload)) * 8;
+    uint32_t nentries = (uint32_t)rd_bits_be(payload, bit0, 32);
+    uint32_t width = nentries ? (uint32_t)rd_bits_be(payload, bit0 + 32, 8) : 0;
+    gb.nentries = nentries;
+    gb.width = (uint8_t)width;
+    gb.rle_bit_off = bit0 + 40;
+    if (nentries == 2) {
+        uint64_t code = rd_bits_be(payload, gb.rle_bit_off, width);
+        uint64_t cnt = rd_bits_be(payload, gb.rle_bit_off + width, width);
+        if (cnt >= bd->count) {
+            uint16_t m = map[code < 256 ? code : 0];
+            gb.uniform_gid = m == 0xFFFFu ? GID_NONE : m;
+            out[bi] = gb;
+            return;
+        }
+    }
+    gb.uniform_gid = GID_VARYING;
+    out[bi] = gb;
+}
+
 // Resolve prepass: one thread per block parses the dictionary header and
 // builds the code mask.  Only plain (<128 B) compress_block sections are
 // parseable on device (bytes.go:291-303); zstd-compressed dictionaries are
@@ -962,19 +1101,131 @@ __device__ __forceinline__ void flush_partial(bydb_partial *partials,
     if (wsumf != 0.0) atomicAdd(&p->sum_f, wsumf);
 }
 
+// Fold rows [a, b] of a block's field column (no predicates): returns the
+// wrapping sum, selected count and optional min/max.  Shared by the
+// per-run grouped fold; mirrors the main kernel's dispatch.
+__device__ void fold_range(const uint8_t *fstream, uint8_t fenc, int64_t first,
+                           uint64_t field_len, int64_t n, int64_t a, int64_t b,
+                           bool need_values, int lane, uint64_t *out_sum,
+                           int64_t *out_mn, int64_t *out_mx, bool *out_have_mm,
+                           DevErr *derr, uint64_t bi) {
+    uint64_t nsel = (uint64_t)(b - a + 1);
+    uint64_t bsum = 0;
+    int64_t bmn = INT64_MAX, bmx = INT64_MIN;
+    bool have = false;
+    if (fenc == BYDB_ENC_CONST) {
+        if (lane == 0) bsum = (uint64_t)first * nsel;
+        bmn = bmx = first;
+        have = true;
+    } else if (fenc == BYDB_ENC_DELTA_CONST) {
+        int vl;
+        int64_t dd = decode_one_varint(fstream, &vl);
+        if (lane == 0) {
+            uint64_t si = (uint64_t)(a + b) * nsel / 2;
+            bsum = (uint64_t)first * nsel + (uint64_t)dd * si;
+        }
+        if (need_values) {
+            int64_t lmn = INT64_MAX, lmx = INT64_MIN;
+            for (int64_t base = a; base <= b; base += 64) {
+                int64_t i = base + lane;
+                if (i <= b) {
+                    int64_t v = (int64_t)((uint64_t)first + (uint64_t)i * (uint64_t)dd);
+                    lmn = v < lmn ? v : lmn;
+                    lmx = v > lmx ? v : lmx;
+                }
+            }
+            bmn = wave_reduce_min(lmn);
+            bmx = wave_reduce_max(lmx);
+            have = true;
+        }
+    } else if (fenc == BYDB_ENC_DELTA) {
+        uint64_t acc;
+        bool dense = field_len == (uint64_t)(n - 1);
+        if (dense) acc = dense_delta_weighted(fstream, n - 1, a, b, lane);
+        else acc = fold_delta_weighted_fast(fstream, n - 1, a, b, lane, derr, bi);
+        acc = wave_reduce_add(acc);
+        if (lane == 0) bsum = (uint64_t)first * nsel + acc;
+        if (need_values) {
+            if (dense) {
+                int64_t psd, psjd;
+                dense_region(fstream, 0, a, lane, &psd, &psjd);
+                int64_t vr0 = (int64_t)((uint64_t)first +
+                                        wave_reduce_add((uint64_t)psd));
+                dense_minmax(fstream, a, b, vr0, lane, &bmn, &bmx);
+                bmn = vr0 < bmn ? vr0 : bmn;
+                bmx = vr0 > bmx ? vr0 : bmx;
+            } else {
+                ScanFold ff;
+                scan_stream(fstream, n - 1, false, first, 0, a, b, INT64_MAX,
+                            INT64_MIN, lane, &ff, derr, bi, nullptr, nullptr,
+                            nullptr);
+                uint64_t lsum2 = ff.sum;
+                int64_t lmn = ff.mn, lmx = ff.mx;
+                if (lane == 0 && a <= 0 && 0 <= b) {
+                    lmn = first < lmn ? first : lmn;
+                    lmx = first > lmx ? first : lmx;
+                }
+                (void)lsum2;
+                bmn = wave_reduce_min(lmn);
+                bmx = wave_reduce_max(lmx);
+            }
+            have = true;
+        }
+    } else if (fenc == BYDB_ENC_DELTA_OF_DELTA) {
+        int vl;
+        int64_t d1 = decode_one_varint(fstream, &vl);
+        uint64_t acc = fold_dod_weighted(fstream + vl, n - 1, a, b, lane, derr, bi);
+        acc = wave_reduce_add(acc);
+        if (lane == 0) {
+            uint64_t si = (uint64_t)(a + b) * nsel / 2;
+            bsum = (uint64_t)first * nsel + (uint64_t)d1 * si + acc;
+        }
+        if (need_values) {
+            int64_t row1 = (int64_t)((uint64_t)first + (uint64_t)d1);
+            ScanFold ff;
+            scan_stream(fstream + vl, n - 1, true, row1, d1, a, b, INT64_MAX,
+                        INT64_MIN, lane, &ff, derr, bi, nullptr, nullptr,
+                        nullptr);
+            int64_t lmn = ff.mn, lmx = ff.mx;
+            if (lane == 0) {
+                if (a <= 0 && 0 <= b) {
+                    lmn = first < lmn ? first : lmn;
+                    lmx = first > lmx ? first : lmx;
+                }
+                if (a <= 1 && 1 <= b) {
+                    lmn = row1 < lmn ? row1 : lmn;
+                    lmx = row1 > lmx ? row1 : lmx;
+                }
+            }
+            bmn = wave_reduce_min(lmn);
+            bmx = wave_reduce_max(lmx);
+            have = true;
+        }
+    } else {
+        dev_set_err(derr, DERR_BAD_ENC, bi);
+    }
+    *out_sum = (uint64_t)__shfl((long long)bsum, 0);
+    *out_mn = bmn;
+    *out_mx = bmx;
+    *out_have_mm = have;
+}
+
 // Compile-time specialization: the FAST instantiation (no value
 // reconstruction, no predicates) sheds the scan/walker machinery and its
 // register pressure; the full one keeps everything.  (EN_* are constant
 // guards — dead branches are eliminated per instantiation.)
-template <bool EN_VALUES, bool EN_PREDS>
-__global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS) ? 4 : 6) void k_scan_agg_t(
+template <bool EN_VALUES, bool EN_PREDS, bool EN_GROUPS>
+__global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) void k_scan_agg_t(
     const uint8_t *__restrict__ payload, const bydb_block_desc *__restrict__ blocks,
     int64_t n_blocks, int64_t min_ts, int64_t max_ts, int flags,
     const PredBlock *__restrict__ preds_in, int n_preds,
-    const SegEntry *__restrict__ segs_in, bydb_partial *__restrict__ partials,
-    DevErr *derr) {
+    const SegEntry *__restrict__ segs_in,
+    const GroupBlock *__restrict__ groups_in,
+    const uint16_t *__restrict__ gmap_in,
+    bydb_partial *__restrict__ partials, DevErr *derr) {
     const PredBlock *preds = EN_PREDS ? preds_in : nullptr;
     const SegEntry *segs = EN_VALUES ? segs_in : nullptr;
+    const GroupBlock *groups = EN_GROUPS ? groups_in : nullptr;
     const int lane = threadIdx.x & 63;
     const int wave_in_block = threadIdx.x >> 6;
     int64_t wave_id = (int64_t)blockIdx.x * (blockDim.x >> 6) + wave_in_block;
@@ -1087,6 +1338,62 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS) ? 4 : 6) void k_scan_a
             if (skip_block) continue;
         }
         bool pred_on = wp0 || wp1 || wp2;
+
+        // ---- per-row group-by on a dictionary tag (EN_GROUPS) ----
+        int64_t block_group = (int64_t)bd->group_code;
+        if (EN_GROUPS && groups) {
+            GroupBlock gb = groups[bi];
+            if (gb.err) { dev_set_err(derr, DERR_BAD_ENC, (uint64_t)bi); continue; }
+            if (gb.uniform_gid == GID_NONE) continue;   // nil/unmapped tag
+            if (gb.uniform_gid != GID_VARYING) {
+                block_group = (int64_t)gb.uniform_gid;  // entity tag: fast paths
+            } else {
+                if (pred_on) {  // v1: row-varying groups + row-varying preds
+                    dev_set_err(derr, DERR_BAD_ENC, (uint64_t)bi);
+                    continue;
+                }
+                if (seg != 0) continue;  // whole block folded at seg 0
+                const uint16_t *map16 = gmap_in + gb.map_off;
+                const uint8_t *gstream = payload + bd->field_off;
+                uint64_t bit = gb.rle_bit_off;
+                int64_t lo = 0;
+                for (uint32_t e = 0; e + 1 < gb.nentries && lo < n; e += 2) {
+                    uint64_t code = rd_bits_be(payload, bit, gb.width);
+                    bit += gb.width;
+                    uint64_t cnt = rd_bits_be(payload, bit, gb.width);
+                    bit += gb.width;
+                    int64_t hi = lo + (int64_t)cnt - 1;
+                    int64_t aa = lo > r0 ? lo : r0;
+                    int64_t bb2 = hi < r1 ? hi : r1;
+                    lo = hi + 1;
+                    if (aa > bb2) continue;
+                    uint16_t m = map16[code < 256 ? code : 0];
+                    if (m == 0xFFFFu) continue;
+                    uint64_t rsum;
+                    int64_t rmn, rmx;
+                    bool rhave;
+                    fold_range(gstream, bd->field_enc, bd->field_first,
+                               bd->field_len, n, aa, bb2,
+                               EN_VALUES && (flags & KF_NEED_VALUES), lane,
+                               &rsum, &rmn, &rmx, &rhave, derr, (uint64_t)bi);
+                    if ((int64_t)m != cur_group) {
+                        flush_partial(partials, cur_group, wsum, wcnt, wmin,
+                                      wmax, wsumf, lane);
+                        cur_group = (int64_t)m;
+                        wsum = 0; wcnt = 0; wmin = INT64_MAX; wmax = INT64_MIN;
+                        wsumf = 0;
+                    }
+                    wsum += rsum;
+                    wcnt += (uint64_t)(bb2 - aa + 1);
+                    if (rhave) {
+                        wmin = rmn < wmin ? rmn : wmin;
+                        wmax = rmx > wmax ? rmx : wmax;
+                    }
+                    if (flags & KF_FLOAT) wsumf += (double)(int64_t)rsum;
+                }
+                continue;
+            }
+        }
         const bool scan_path = EN_VALUES && (flags & KF_NEED_VALUES) &&
                                bd->field_enc == BYDB_ENC_DELTA &&
                                bd->field_len != (uint64_t)(n - 1);
@@ -1320,10 +1627,10 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS) ? 4 : 6) void k_scan_a
         // scan; instead fold block totals into per-wave registers and
         // flush only when the group changes or the wave is done.
         bsum = (uint64_t)__shfl((long long)bsum, 0);  // lane0 holds closed forms
-        if ((int64_t)bd->group_code != cur_group) {
+        if (block_group != cur_group) {
             flush_partial(partials, cur_group, wsum, wcnt, wmin, wmax, wsumf,
                           lane);
-            cur_group = (int64_t)bd->group_code;
+            cur_group = block_group;
             wsum = 0; wcnt = 0; wmin = INT64_MAX; wmax = INT64_MIN; wsumf = 0;
         }
         wsum += bsum;
@@ -1376,6 +1683,18 @@ struct bydb_session {
     SegEntry *d_segs = nullptr;
     int64_t segs_cap = 0;
     bool segs_built = false;
+    // per-row group-by on a dictionary tag
+    int group_slot = -1;               // -1 = group by block group_code
+    GroupBlock *d_groups = nullptr;
+    uint16_t *d_gmap = nullptr;
+    int64_t groups_cap = 0;
+    bool groups_built = false;
+    uint8_t *d_dom_blob = nullptr;
+    uint64_t *d_dom_offs = nullptr;
+    uint64_t *d_dom_hashes = nullptr;
+    uint32_t *d_dom_gids = nullptr;
+    uint32_t dom_table_size = 0;
+    uint32_t dom_n = 0;
     uint8_t *d_pred_bytes = nullptr;
     uint64_t pred_bytes_cap = 0;
     float last_ms = 0.0f;
@@ -1427,6 +1746,12 @@ extern "C" void bydb_session_destroy(bydb_session *s) {
     if (s->d_preds) (void)hipFree(s->d_preds);
     if (s->d_pred_bytes) (void)hipFree(s->d_pred_bytes);
     if (s->d_segs) (void)hipFree(s->d_segs);
+    if (s->d_groups) (void)hipFree(s->d_groups);
+    if (s->d_gmap) (void)hipFree(s->d_gmap);
+    if (s->d_dom_blob) (void)hipFree(s->d_dom_blob);
+    if (s->d_dom_offs) (void)hipFree(s->d_dom_offs);
+    if (s->d_dom_hashes) (void)hipFree(s->d_dom_hashes);
+    if (s->d_dom_gids) (void)hipFree(s->d_dom_gids);
     if (s->d_err) (void)hipFree(s->d_err);
     if (s->ev_start) (void)hipEventDestroy(s->ev_start);
     if (s->ev_stop) (void)hipEventDestroy(s->ev_stop);
@@ -1453,6 +1778,7 @@ extern "C" int bydb_part_reserve(bydb_session *s, uint64_t payload_bytes,
     s->payload_len = 0;
     s->n_blocks = 0;
     s->segs_built = false;
+    s->groups_built = false;
     return BYDB_OK;
 }
 
@@ -1461,6 +1787,7 @@ extern "C" int bydb_part_append(bydb_session *s, const uint8_t *payload,
                                 int64_t n_blocks) {
     HIP_TRY(s, hipSetDevice(s->device));
     s->segs_built = false;
+    s->groups_built = false;
     if (s->payload_len + len > s->payload_cap ||
         s->n_blocks + n_blocks > s->blocks_cap) {
         s->err = "part_append exceeds reservation";
@@ -1497,7 +1824,61 @@ extern "C" int bydb_agg_configure(bydb_session *s, int field_vtype,
         s->partials_cap = n_groups;
     }
     s->d_acc = s->d_partials;
+    s->group_slot = -1;
     return bydb_reset(s);
+}
+
+// Group rows by the dictionary tag in `slot`: the host supplies the group
+// DOMAIN (dense gid = index into the value list, mirroring the reference's
+// first-seen group materialisation order, which the host layer controls —
+// computeKey semantics, aggregation.go:523).  Values: concatenated blob +
+// n+1 offsets.  n_groups = n domain values.
+extern "C" int bydb_agg_configure_by_tag(bydb_session *s, int field_vtype,
+                                         uint32_t func_mask, int tag_slot,
+                                         const uint8_t *dom_blob,
+                                         const uint64_t *dom_offs,
+                                         uint32_t n_values, int mode) {
+    if (tag_slot < 0 || tag_slot > 2 || n_values < 1 || n_values > 65000) {
+        s->err = "bad tag slot or domain size";
+        return BYDB_ERR_BAD_ARG;
+    }
+    int rc = bydb_agg_configure(s, field_vtype, func_mask, n_values, mode);
+    if (rc != BYDB_OK) return rc;
+    s->group_slot = tag_slot;
+    s->groups_built = false;
+    // host-side open-addressing table: slot -> (fnv1a hash, gid)
+    uint32_t tsize = 4;
+    while (tsize < n_values * 4) tsize <<= 1;
+    std::vector<uint64_t> hashes(tsize, 0);
+    std::vector<uint32_t> gids(tsize, GID_NONE);
+    for (uint32_t g = 0; g < n_values; g++) {
+        const uint8_t *v = dom_blob + dom_offs[g];
+        uint64_t len = dom_offs[g + 1] - dom_offs[g];
+        uint64_t h = fnv1a(v, len);
+        uint32_t sl = (uint32_t)h & (tsize - 1);
+        while (gids[sl] != GID_NONE) sl = (sl + 1) & (tsize - 1);
+        hashes[sl] = h;
+        gids[sl] = g;
+    }
+    uint64_t blob_len = dom_offs[n_values];
+    if (s->d_dom_blob) { (void)hipFree(s->d_dom_blob); s->d_dom_blob = nullptr; }
+    if (s->d_dom_offs) { (void)hipFree(s->d_dom_offs); s->d_dom_offs = nullptr; }
+    if (s->d_dom_hashes) { (void)hipFree(s->d_dom_hashes); s->d_dom_hashes = nullptr; }
+    if (s->d_dom_gids) { (void)hipFree(s->d_dom_gids); s->d_dom_gids = nullptr; }
+    HIP_TRY(s, hipMalloc(&s->d_dom_blob, blob_len ? blob_len : 1));
+    HIP_TRY(s, hipMalloc(&s->d_dom_offs, sizeof(uint64_t) * (n_values + 1)));
+    HIP_TRY(s, hipMalloc(&s->d_dom_hashes, sizeof(uint64_t) * tsize));
+    HIP_TRY(s, hipMalloc(&s->d_dom_gids, sizeof(uint32_t) * tsize));
+    HIP_TRY(s, hipMemcpy(s->d_dom_blob, dom_blob, blob_len, hipMemcpyHostToDevice));
+    HIP_TRY(s, hipMemcpy(s->d_dom_offs, dom_offs, sizeof(uint64_t) * (n_values + 1),
+                         hipMemcpyHostToDevice));
+    HIP_TRY(s, hipMemcpy(s->d_dom_hashes, hashes.data(), sizeof(uint64_t) * tsize,
+                         hipMemcpyHostToDevice));
+    HIP_TRY(s, hipMemcpy(s->d_dom_gids, gids.data(), sizeof(uint32_t) * tsize,
+                         hipMemcpyHostToDevice));
+    s->dom_table_size = tsize;
+    s->dom_n = n_values;
+    return BYDB_OK;
 }
 
 extern "C" int bydb_set_partials_buffer(bydb_session *s, void *dev_ptr,
@@ -1599,6 +1980,37 @@ extern "C" int bydb_consume_multi(bydb_session *s, int64_t min_ts,
         }
         segs = s->d_segs;
     }
+    GroupBlock *groups = nullptr;
+    if (s->group_slot >= 0) {
+        if (s->groups_cap < s->n_blocks) {
+            if (s->d_groups) (void)hipFree(s->d_groups);
+            if (s->d_gmap) (void)hipFree(s->d_gmap);
+            HIP_TRY(s, hipMalloc(&s->d_groups,
+                                 sizeof(GroupBlock) * (size_t)s->n_blocks));
+            HIP_TRY(s, hipMalloc(&s->d_gmap,
+                                 sizeof(uint16_t) * 256 * (size_t)s->n_blocks));
+            s->groups_cap = s->n_blocks;
+            s->groups_built = false;
+        }
+        if (!s->groups_built) {
+            GroupDomain dom;
+            dom.blob = s->d_dom_blob;
+            dom.offs = s->d_dom_offs;
+            dom.hashes = s->d_dom_hashes;
+            dom.gids = s->d_dom_gids;
+            dom.table_size = s->dom_table_size;
+            dom.n = s->dom_n;
+            int rthreads = 256;
+            int rblocks = (int)((s->n_blocks + rthreads - 1) / rthreads);
+            hipLaunchKernelGGL(k_resolve_groups, dim3(rblocks), dim3(rthreads),
+                               0, s->stream, s->d_payload, s->d_blocks,
+                               s->n_blocks, s->group_slot, dom, s->d_groups,
+                               s->d_gmap);
+            HIP_TRY(s, hipGetLastError());
+            s->groups_built = true;
+        }
+        groups = s->d_groups;
+    }
     int64_t waves_needed = segs ? s->n_blocks * MAX_SEGS : s->n_blocks;
     int64_t wgs = (waves_needed + 3) / 4;
     int grid = (int)(wgs < 8192 ? wgs : 8192);     // grid-stride beyond
@@ -1606,13 +2018,25 @@ extern "C" int bydb_consume_multi(bydb_session *s, int64_t min_ts,
     HIP_TRY(s, hipEventRecord(s->ev_start, s->stream));
     const bool en_values = (flags & KF_NEED_VALUES) != 0;
     const bool en_preds = n_preds > 0;
-    auto kfn = en_values ? (en_preds ? k_scan_agg_t<true, true>
-                                     : k_scan_agg_t<true, false>)
-                         : (en_preds ? k_scan_agg_t<false, true>
-                                     : k_scan_agg_t<false, false>);
+    const bool en_groups = groups != nullptr;
+    void (*kfn)(const uint8_t *, const bydb_block_desc *, int64_t, int64_t,
+                int64_t, int, const PredBlock *, int, const SegEntry *,
+                const GroupBlock *, const uint16_t *, bydb_partial *, DevErr *);
+    if (en_values) {
+        if (en_preds) kfn = en_groups ? k_scan_agg_t<true, true, true>
+                                      : k_scan_agg_t<true, true, false>;
+        else kfn = en_groups ? k_scan_agg_t<true, false, true>
+                             : k_scan_agg_t<true, false, false>;
+    } else {
+        if (en_preds) kfn = en_groups ? k_scan_agg_t<false, true, true>
+                                      : k_scan_agg_t<false, true, false>;
+        else kfn = en_groups ? k_scan_agg_t<false, false, true>
+                             : k_scan_agg_t<false, false, false>;
+    }
     hipLaunchKernelGGL(kfn, dim3(grid), dim3(threads), 0, s->stream,
                        s->d_payload, s->d_blocks, s->n_blocks, min_ts, max_ts,
-                       flags, preds, n_preds, segs, s->d_acc, s->d_err);
+                       flags, preds, n_preds, segs, groups, s->d_gmap,
+                       s->d_acc, s->d_err);
     HIP_TRY(s, hipGetLastError());
     HIP_TRY(s, hipEventRecord(s->ev_stop, s->stream));
     s->consumed = true;

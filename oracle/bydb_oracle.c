@@ -1197,6 +1197,88 @@ int bo_scan_agg_multi(const uint8_t *payload, const bo_block_desc *blocks,
     return rc;
 }
 
+int bo_scan_agg_bytag(const uint8_t *payload, const bo_block_desc *blocks,
+                      int64_t n_blocks, int field_vtype, int64_t min_ts,
+                      int64_t max_ts, int slot, const uint8_t *dom_blob,
+                      const int64_t *dom_lens, int64_t n_dom,
+                      bo_agg_result *out) {
+    for (int64_t g = 0; g < n_dom; g++) agg_reset(&out[g]);
+    int64_t *ts_buf = (int64_t *)malloc(sizeof(int64_t) * MAX_BLOCK_ROWS);
+    int64_t *i64_buf = (int64_t *)malloc(sizeof(int64_t) * MAX_BLOCK_ROWS);
+    double *f64_buf = (double *)malloc(sizeof(double) * MAX_BLOCK_ROWS);
+    uint8_t *tagdata = (uint8_t *)malloc((size_t)1 << 24);
+    int64_t *taglen = (int64_t *)malloc(sizeof(int64_t) * MAX_BLOCK_ROWS);
+    const uint8_t **dvals = (const uint8_t **)malloc(sizeof(void *) * (size_t)n_dom);
+    {
+        const uint8_t *p = dom_blob;
+        for (int64_t g = 0; g < n_dom; g++) {
+            dvals[g] = p;
+            p += dom_lens[g] > 0 ? dom_lens[g] : 0;
+        }
+    }
+    int rc = BO_OK;
+    for (int64_t i = 0; i < n_blocks && rc == BO_OK; i++) {
+        const bo_block_desc *b = &blocks[i];
+        int64_t n = (int64_t)b->count;
+        rc = bo_timestamps_decode(ts_buf, NULL, payload + b->ts_off,
+                                  (size_t)(b->ts_len + b->ver_len),
+                                  b->ts_enc_with_version, b->ts_min, b->ts_len,
+                                  b->version_enc, b->version_first, n);
+        if (rc != BO_OK) break;
+        int64_t r0, r1;
+        if (!bo_find_range(ts_buf, n, min_ts, max_ts, &r0, &r1)) continue;
+        uint64_t toff = slot == 0 ? b->tag_off : slot == 1 ? b->tag2_off : b->tag3_off;
+        uint64_t tlen = slot == 0 ? b->tag_len : slot == 1 ? b->tag2_len : b->tag3_len;
+        if (tlen == 0) continue;  /* nil tag rows: no group */
+        if (payload[toff] != BO_ENC_DICTIONARY) { rc = BO_ERR_BAD_TYPE; break; }
+        size_t tdl = 0;
+        rc = bo_dictionary_decode(tagdata, (size_t)1 << 24, taglen,
+                                  payload + toff + 1, tlen - 1, n, &tdl);
+        if (rc != BO_OK) break;
+        if (field_vtype == BO_VT_INT64) {
+            rc = bo_column_i64_decode(i64_buf, payload + b->col_off, b->col_len, n);
+        } else if (field_vtype == BO_VT_FLOAT64) {
+            rc = bo_column_f64_decode(f64_buf, payload + b->col_off, b->col_len, n);
+        } else {
+            rc = BO_ERR_BAD_TYPE;
+        }
+        if (rc != BO_OK) break;
+        size_t tago = 0;
+        for (int64_t r = 0; r < n; r++) {
+            int64_t tl = taglen[r];
+            size_t my = tago;
+            if (tl > 0) tago += (size_t)tl;
+            if (r < r0 || r > r1 || tl < 0) continue;
+            int64_t gid = -1;
+            for (int64_t g = 0; g < n_dom; g++) {
+                if (dom_lens[g] == tl &&
+                    (tl == 0 || memcmp(dvals[g], tagdata + my, (size_t)tl) == 0)) {
+                    gid = g;
+                    break;
+                }
+            }
+            if (gid < 0) continue;
+            bo_agg_result *o = &out[gid];
+            if (field_vtype == BO_VT_INT64) {
+                int64_t v = i64_buf[r];
+                o->sum_i = (int64_t)((uint64_t)o->sum_i + (uint64_t)v);
+                o->count++;
+                if (v < o->min_i) o->min_i = v;
+                if (v > o->max_i) o->max_i = v;
+            } else {
+                double v = f64_buf[r];
+                o->sum_f += v;
+                o->count++;
+                if (v < o->min_f) o->min_f = v;
+                if (v > o->max_f) o->max_f = v;
+            }
+        }
+    }
+    free(ts_buf); free(i64_buf); free(f64_buf); free(tagdata); free(taglen);
+    free((void *)dvals);
+    return rc;
+}
+
 /* ===================== xxhash64 =====================
  * Canonical XXH64 (seed 0) — pins cespare/xxhash v2.3.0 used for
  * Entity -> SeriesID (pkg/convert/hash.go:23). */

@@ -194,6 +194,16 @@ int bo_scan_agg_multi(const uint8_t *payload, const bo_block_desc *blocks,
                       const int64_t pred_lens[3], bo_agg_result *out,
                       int64_t n_groups);
 
+/* Per-row group-by on a dictionary tag: group id = index of the row's
+ * tag value in the domain list (host-controlled order = the reference's
+ * first-seen materialisation, computeKey aggregation.go:523).  Rows with
+ * nil or out-of-domain tags are dropped.  slot: 0..2. */
+int bo_scan_agg_bytag(const uint8_t *payload, const bo_block_desc *blocks,
+                      int64_t n_blocks, int field_vtype, int64_t min_ts,
+                      int64_t max_ts, int slot, const uint8_t *dom_blob,
+                      const int64_t *dom_lens, int64_t n_dom,
+                      bo_agg_result *out);
+
 /* MEAN finalisation — pkg/query/aggregation/function.go:30-45 (clamps to >=1) */
 int64_t bo_mean_val_i64(int64_t sum, int64_t count);
 double bo_mean_val_f64(double sum, double count);

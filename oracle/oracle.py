@@ -7,6 +7,9 @@ import ctypes as C
 import os
 import subprocess
 
+INT64_MIN = -(2 ** 63)
+INT64_MAX = 2 ** 63 - 1
+
 _DIR = os.path.dirname(os.path.abspath(__file__))
 _SO = os.path.join(_DIR, "libbydb_oracle.so")
 
@@ -127,6 +130,10 @@ _L.bo_scan_agg_grouped.restype = C.c_int
 _L.bo_scan_agg_grouped.argtypes = [u8p, C.POINTER(BlockDesc), C.c_int64, C.c_int,
                                    C.c_int64, C.c_int64, u8p, C.c_int64,
                                    C.POINTER(AggResult), C.c_int64]
+_L.bo_scan_agg_bytag.restype = C.c_int
+_L.bo_scan_agg_bytag.argtypes = [u8p, C.POINTER(BlockDesc), C.c_int64, C.c_int,
+                                 C.c_int64, C.c_int64, C.c_int, u8p, i64p,
+                                 C.c_int64, C.POINTER(AggResult)]
 _L.bo_scan_agg_multi.restype = C.c_int
 _L.bo_scan_agg_multi.argtypes = [u8p, C.POINTER(BlockDesc), C.c_int64, C.c_int,
                                  C.c_int64, C.c_int64, u8p, i64p,
@@ -379,6 +386,26 @@ def bytes_block_decode(payload, n):
             out.append(raw[pos: pos + ln])
             pos += ln
     return out
+
+
+def scan_agg_bytag(payload: bytes, blocks, field_vtype, slot, domain,
+                   min_ts=INT64_MIN, max_ts=INT64_MAX):
+    """Per-row group-by on a dictionary tag; gid = index in domain."""
+    descs = (BlockDesc * len(blocks))()
+    for i, b in enumerate(blocks):
+        for k, v in b.items():
+            setattr(descs[i], k, v)
+    blob = b"".join(v for v in domain if v is not None)
+    lens = [(-1 if v is None else len(v)) for v in domain]
+    out = (AggResult * len(domain))()
+    src = (C.c_uint8 * max(len(payload), 1)).from_buffer_copy(payload or b"\0")
+    bb = (C.c_uint8 * max(len(blob), 1)).from_buffer_copy(blob or b"\0")
+    rc = _L.bo_scan_agg_bytag(src, descs, len(blocks), field_vtype, min_ts,
+                              max_ts, slot, bb, (C.c_int64 * len(domain))(*lens),
+                              len(domain), out)
+    if rc != 0:
+        raise ValueError(f"scan_agg_bytag rc={rc}")
+    return list(out)
 
 
 def mean_val_i64(s, c):
