@@ -115,6 +115,7 @@ enum {
     DERR_BAD_STREAM = 1,    // no terminator in a 64-byte window
     DERR_BAD_ENC = 2,       // unsupported encode type in kernel
     DERR_DESC_TS = 3,       // descending timestamps (reference never writes them)
+    DERR_GROUP_RANGE = 4,   // block group_code >= configured n_groups
 };
 
 __device__ __forceinline__ void dev_set_err(DevErr *e, unsigned code, uint64_t bi) {
@@ -1221,7 +1222,7 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
     const PredBlock *__restrict__ preds_in, int n_preds,
     const SegEntry *__restrict__ segs_in,
     const GroupBlock *__restrict__ groups_in,
-    const uint16_t *__restrict__ gmap_in,
+    const uint16_t *__restrict__ gmap_in, int64_t n_groups,
     bydb_partial *__restrict__ partials, DevErr *derr) {
     const PredBlock *preds = EN_PREDS ? preds_in : nullptr;
     const SegEntry *segs = EN_VALUES ? segs_in : nullptr;
@@ -1341,6 +1342,10 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
 
         // ---- per-row group-by on a dictionary tag (EN_GROUPS) ----
         int64_t block_group = (int64_t)bd->group_code;
+        if (!EN_GROUPS && block_group >= n_groups) {
+            dev_set_err(derr, DERR_GROUP_RANGE, (uint64_t)bi);
+            continue;
+        }
         if (EN_GROUPS && groups) {
             GroupBlock gb = groups[bi];
             if (gb.err) { dev_set_err(derr, DERR_BAD_ENC, (uint64_t)bi); continue; }
@@ -2021,7 +2026,8 @@ extern "C" int bydb_consume_multi(bydb_session *s, int64_t min_ts,
     const bool en_groups = groups != nullptr;
     void (*kfn)(const uint8_t *, const bydb_block_desc *, int64_t, int64_t,
                 int64_t, int, const PredBlock *, int, const SegEntry *,
-                const GroupBlock *, const uint16_t *, bydb_partial *, DevErr *);
+                const GroupBlock *, const uint16_t *, int64_t,
+                bydb_partial *, DevErr *);
     if (en_values) {
         if (en_preds) kfn = en_groups ? k_scan_agg_t<true, true, true>
                                       : k_scan_agg_t<true, true, false>;
@@ -2036,7 +2042,7 @@ extern "C" int bydb_consume_multi(bydb_session *s, int64_t min_ts,
     hipLaunchKernelGGL(kfn, dim3(grid), dim3(threads), 0, s->stream,
                        s->d_payload, s->d_blocks, s->n_blocks, min_ts, max_ts,
                        flags, preds, n_preds, segs, groups, s->d_gmap,
-                       s->d_acc, s->d_err);
+                       (int64_t)s->n_groups, s->d_acc, s->d_err);
     HIP_TRY(s, hipGetLastError());
     HIP_TRY(s, hipEventRecord(s->ev_stop, s->stream));
     s->consumed = true;

@@ -61,7 +61,7 @@ def random_tags(rng, n, kind):
     return tags
 
 
-def build_scenario(rng):
+def build_scenario(rng, n_groups=1):
     is_float = rng.random() < 0.3
     n_blocks = rng.randint(2, 24)
     tag_kind = rng.randrange(3)
@@ -74,11 +74,11 @@ def build_scenario(rng):
             cents = random_values(rng, n, rng.choice([0, 1, 2]))
             cents = [c % 10**12 - 5 * 10**11 for c in cents]
             b.add_block_f64(sid + 1, ts, [1] * n, [c / 100.0 for c in cents],
-                            group_code=sid % 3)
+                            group_code=sid % n_groups)
         else:
             b.add_block_i64(sid + 1, ts, [1] * n,
                             random_values(rng, n, rng.randrange(6)),
-                            group_code=sid % 3)
+                            group_code=sid % n_groups)
         t = random_tags(rng, n, tag_kind)
         if t is not None:
             b.set_block_tag(t)
@@ -149,3 +149,47 @@ def test_fuzz_scenario(seed):
     check_scalar(rng, b, is_float, tag_kind)
     if tag_kind:
         check_bytag(rng, b, is_float)
+
+
+@pytest.mark.parametrize("seed", range(6))
+def test_fuzz_grouped_by_code(seed):
+    rng = random.Random(0x1F0 + seed)
+    b, is_float, _ = build_scenario(rng, n_groups=3)
+    vtype = VT_FLOAT64 if is_float else VT_INT64
+    exp = b.blocks()[0].exp if is_float else 0
+    orc = oracle_scan(b, vtype, n_groups=3)
+    s = Session(0)
+    s.upload_part(b)
+    s.configure(vtype, [AGG_SUM, AGG_COUNT, AGG_MIN, AGG_MAX], n_groups=3,
+                float_exp=exp)
+    s.consume()
+    gs = s.finalize()
+    s.close()
+    for g, oc in zip(gs, orc):
+        assert g.count == oc.count
+        if is_float:
+            if oc.count:
+                assert g.min_f == oc.min_f and g.max_f == oc.max_f
+                assert math.isclose(g.sum_f, oc.sum_f, rel_tol=1e-9,
+                                    abs_tol=1e-6)
+        else:
+            assert g.sum_i == oc.sum_i
+            if oc.count:
+                assert g.min_i == oc.min_i and g.max_i == oc.max_i
+
+
+def test_group_code_out_of_range_fails_loud():
+    b = PartBuilder()
+    ts = [T0 + i * MS for i in range(100)]
+    b.add_block_i64(1, ts, [1] * 100, list(range(100)), group_code=5)
+    s = Session(0)
+    s.upload_part(b)
+    s.configure(VT_INT64, [AGG_SUM, AGG_COUNT], n_groups=1)
+    s.consume()
+    try:
+        s.finalize()
+        raise AssertionError("out-of-range group_code must be a loud error")
+    except RuntimeError as e:
+        assert "decode error 4" in str(e)
+    finally:
+        s.close()
