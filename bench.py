@@ -65,6 +65,11 @@ def main():
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    # test hooks: BYDB_FORCE_DEVICE maps every rank onto one GPU and
+    # BYDB_DIST_BACKEND=gloo swaps the merge transport, so the multi-rank
+    # code path can be smoke-tested on a single-GPU box
+    device = int(os.environ.get("BYDB_FORCE_DEVICE", local_rank))
+    backend = os.environ.get("BYDB_DIST_BACKEND", "nccl")
     if args.gpus > 1 and world == 1:
         log("WARNING: --gpus > 1 without torchrun; measuring 1 rank")
     n_gpus = world if world > 1 else 1
@@ -72,8 +77,9 @@ def main():
     if world > 1:
         import torch.distributed as dist_mod
         dist = dist_mod
-        torch.cuda.set_device(local_rank)
-        dist.init_process_group("nccl")
+        if backend == "nccl":
+            torch.cuda.set_device(device)
+        dist.init_process_group(backend)
 
     n_series, n_dp = args.series, args.dp
     full_config = n_series == 10000 and n_dp == 1_000_000
@@ -119,7 +125,7 @@ def main():
         + n_series * ((n_dp + 8191) // 8192) * tag_overhead
     est_blocks = n_series * ((n_dp + 8191) // 8192)
 
-    sess = ba.Session(local_rank)
+    sess = ba.Session(device)
     sess.reserve(est_payload, est_blocks)
 
     gen_t = time.perf_counter()
@@ -151,9 +157,9 @@ def main():
 
     # partials live in a torch CUDA tensor so the merge is RCCL over xGMI
     part_t = None
-    if world > 1:
+    if world > 1 and backend == "nccl":
         part_t = torch.zeros(n_groups * 6, dtype=torch.int64,
-                             device=f"cuda:{local_rank}")
+                             device=f"cuda:{device}")
         sess.set_partials_buffer(part_t.data_ptr(), part_t.numel() * 8)
 
     preds = W["preds"] or None
@@ -163,13 +169,21 @@ def main():
         sess.reset()
         sess.consume(min_ts=t0_r, max_ts=t0_r + n_dp * STRIDE, preds=preds)
         if world > 1:
-            # sync the session stream, then RCCL-merge the partials over
-            # xGMI (AggModeReduce Combine semantics)
+            # sync the session stream, then merge the partials
+            # (AggModeReduce Combine semantics): RCCL over xGMI on the
+            # device tensor, or gloo over host copies in the smoke config
             parts = sess.finalize_partials()
-            from banyandb_amd.distributed import allreduce_partials
-            allreduce_partials(dist, part_t, n_groups,
-                               need_minmax=need_minmax,
-                               need_float=W["float"])
+            from banyandb_amd.distributed import (allreduce_partials,
+                                                  partials_from_structs)
+            if backend == "nccl":
+                allreduce_partials(dist, part_t, n_groups,
+                                   need_minmax=need_minmax,
+                                   need_float=W["float"])
+            else:
+                t = partials_from_structs(parts)
+                allreduce_partials(dist, t, n_groups,
+                                   need_minmax=need_minmax,
+                                   need_float=W["float"])
             return parts
         return sess.finalize_partials()
 
@@ -180,17 +194,18 @@ def main():
     launch_ms = []
     if world > 1:
         dist.barrier()
-    torch.cuda.synchronize(local_rank) if torch.cuda.is_available() else None
+    torch.cuda.synchronize(device) if torch.cuda.is_available() else None
     t_start = time.perf_counter()
     for _ in range(args.steps):
         one_step()
         launch_ms.append(sess.last_consume_ms())
-    torch.cuda.synchronize(local_rank) if torch.cuda.is_available() else None
+    torch.cuda.synchronize(device) if torch.cuda.is_available() else None
     if world > 1:
         dist.barrier()
     elapsed = time.perf_counter() - t_start
     if world > 1:
-        t = torch.tensor([elapsed], device=f"cuda:{local_rank}")
+        dev = f"cuda:{device}" if backend == "nccl" else "cpu"
+        t = torch.tensor([elapsed], device=dev)
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
 
