@@ -115,6 +115,12 @@ def _load():
     lib.bydb_part_append.argtypes = [C.c_void_p, u8p, C.c_uint64, bp, C.c_int64]
     lib.bydb_agg_configure.restype = C.c_int
     lib.bydb_agg_configure.argtypes = [C.c_void_p, C.c_int, C.c_uint32, C.c_uint32, C.c_int]
+    lib.bydb_agg_configure_by_tags.restype = C.c_int
+    lib.bydb_agg_configure_by_tags.argtypes = [C.c_void_p, C.c_int, C.c_uint32,
+                                               C.POINTER(C.c_int), C.c_int,
+                                               C.POINTER(u8p),
+                                               C.POINTER(C.POINTER(C.c_uint64)),
+                                               C.POINTER(C.c_uint32), C.c_int]
     lib.bydb_agg_configure_by_tag.restype = C.c_int
     lib.bydb_agg_configure_by_tag.argtypes = [C.c_void_p, C.c_int, C.c_uint32,
                                               C.c_int, u8p,
@@ -378,6 +384,38 @@ class Session:
         self._ck(_lib.bydb_agg_configure_by_tag(
             self._h, field_vtype, mask, tag_slot, buf,
             (C.c_uint64 * len(offs))(*offs), len(domain), mode))
+        self._ck(_lib.bydb_set_float_exp(self._h, float_exp))
+
+    def configure_by_tags(self, field_vtype, funcs, slots, domains,
+                          mode=MODE_ALL, float_exp=0):
+        """Composite group-by over multiple dictionary tags; group id =
+        g0 + n0*g1 + n0*n1*g2 (domains in slot-list order)."""
+        mask = 0
+        for f in funcs:
+            mask |= 1 << f
+        u8p = C.POINTER(C.c_uint8)
+        bufs = []
+        blob_ptrs = (u8p * len(domains))()
+        off_ptrs = (C.POINTER(C.c_uint64) * len(domains))()
+        nvals = (C.c_uint32 * len(domains))()
+        total = 1
+        for i, dom in enumerate(domains):
+            blob = b"".join(dom)
+            offs = [0]
+            for v in dom:
+                offs.append(offs[-1] + len(v))
+            buf = (C.c_uint8 * max(len(blob), 1)).from_buffer_copy(blob or b"\0")
+            oa = (C.c_uint64 * len(offs))(*offs)
+            bufs.append((buf, oa))
+            blob_ptrs[i] = C.cast(buf, u8p)
+            off_ptrs[i] = oa
+            nvals[i] = len(dom)
+            total *= len(dom)
+        self.n_groups = total
+        self.field_vtype = field_vtype
+        self._ck(_lib.bydb_agg_configure_by_tags(
+            self._h, field_vtype, mask, (C.c_int * len(slots))(*slots),
+            len(slots), blob_ptrs, off_ptrs, nvals, mode))
         self._ck(_lib.bydb_set_float_exp(self._h, float_exp))
 
     def set_partials_buffer(self, dev_ptr, nbytes):

@@ -640,7 +640,8 @@ __global__ void k_resolve_groups(const uint8_t *__restrict__ payload,
                                  const bydb_block_desc *__restrict__ blocks,
                                  int64_t n_blocks, int slot, GroupDomain dom,
                                  GroupBlock *__restrict__ out,
-                                 uint16_t *__restrict__ map_arena) {
+                                 uint16_t *__restrict__ map_arena,
+                                 uint32_t map_base) {
     int64_t bi = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (bi >= n_blocks) return;
     const bydb_block_desc *bd = &blocks[bi];
@@ -648,7 +649,7 @@ __global__ void k_resolve_groups(const uint8_t *__restrict__ payload,
     uint64_t tlen = slot == 0 ? bd->tag_len : slot == 1 ? bd->tag2_len : bd->tag3_len;
     GroupBlock gb;
     gb.uniform_gid = GID_NONE;
-    gb.map_off = (uint32_t)(bi * 256);
+    gb.map_off = map_base + (uint32_t)(bi * 256);
     gb.rle_bit_off = 0; gb.nentries = 0; gb.width = 0; gb.err = 0;
     gb._p[0] = gb._p[1] = 0;
     if (tlen == 0) { out[bi] = gb; return; }
@@ -1222,8 +1223,10 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
     const PredBlock *__restrict__ preds_in, int n_preds,
     const SegEntry *__restrict__ segs_in,
     const GroupBlock *__restrict__ groups_in,
-    const uint16_t *__restrict__ gmap_in, int64_t n_groups,
+    const uint16_t *__restrict__ gmap_in, int n_gslots, int64_t gm0,
+    int64_t gm1, int64_t gm2, int64_t n_groups,
     bydb_partial *__restrict__ partials, DevErr *derr) {
+    const int64_t gmul[3] = {gm0, gm1, gm2};
     const PredBlock *preds = EN_PREDS ? preds_in : nullptr;
     const SegEntry *segs = EN_VALUES ? segs_in : nullptr;
     const GroupBlock *groups = EN_GROUPS ? groups_in : nullptr;
@@ -1340,61 +1343,115 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
         }
         bool pred_on = wp0 || wp1 || wp2;
 
-        // ---- per-row group-by on a dictionary tag (EN_GROUPS) ----
+        // ---- per-row group-by on dictionary tags (EN_GROUPS) ----
+        // Composite keys over up to 3 tag slots: gid = g0 + n0*g1 +
+        // n0*n1*g2 (computeKey's multi-component key, aggregation.go:523).
+        // All-uniform blocks (entity tags) take the fast fold paths; any
+        // row-varying slot folds per intersection of the RLE runs.
         int64_t block_group = (int64_t)bd->group_code;
         if (!EN_GROUPS && block_group >= n_groups) {
             dev_set_err(derr, DERR_GROUP_RANGE, (uint64_t)bi);
             continue;
         }
         if (EN_GROUPS && groups) {
-            GroupBlock gb = groups[bi];
-            if (gb.err) { dev_set_err(derr, DERR_BAD_ENC, (uint64_t)bi); continue; }
-            if (gb.uniform_gid == GID_NONE) continue;   // nil/unmapped tag
-            if (gb.uniform_gid != GID_VARYING) {
-                block_group = (int64_t)gb.uniform_gid;  // entity tag: fast paths
+            GroupBlock gb[3];
+            bool varying = false, skip_blk = false, errd = false;
+            for (int sl = 0; sl < n_gslots; sl++) {
+                gb[sl] = groups[(int64_t)sl * n_blocks + bi];
+                if (gb[sl].err) errd = true;
+                else if (gb[sl].uniform_gid == GID_NONE) skip_blk = true;
+                else if (gb[sl].uniform_gid == GID_VARYING) varying = true;
+            }
+            if (errd) { dev_set_err(derr, DERR_BAD_ENC, (uint64_t)bi); continue; }
+            if (skip_blk) continue;   // nil/unmapped on a grouped slot
+            if (!varying) {
+                block_group = 0;
+                for (int sl = 0; sl < n_gslots; sl++)
+                    block_group += (int64_t)gb[sl].uniform_gid * gmul[sl];
             } else {
                 if (pred_on) {  // v1: row-varying groups + row-varying preds
                     dev_set_err(derr, DERR_BAD_ENC, (uint64_t)bi);
                     continue;
                 }
                 if (seg != 0) continue;  // whole block folded at seg 0
-                const uint16_t *map16 = gmap_in + gb.map_off;
                 const uint8_t *gstream = payload + bd->field_off;
-                uint64_t bit = gb.rle_bit_off;
+                // per-slot run cursors (uniform slots are one infinite run)
+                uint64_t bit[3];
+                uint32_t ent[3];
+                int64_t run_hi_s[3];
+                int64_t gid_s[3];
+                for (int sl = 0; sl < n_gslots; sl++) {
+                    if (gb[sl].uniform_gid != GID_VARYING) {
+                        gid_s[sl] = (int64_t)gb[sl].uniform_gid;
+                        run_hi_s[sl] = n;
+                        ent[sl] = gb[sl].nentries;  // exhausted
+                        bit[sl] = 0;
+                    } else {
+                        bit[sl] = gb[sl].rle_bit_off;
+                        ent[sl] = 0;
+                        run_hi_s[sl] = 0;
+                        gid_s[sl] = -1;
+                    }
+                }
+                auto advance = [&](int sl) {
+                    if (ent[sl] + 1 >= gb[sl].nentries) {
+                        run_hi_s[sl] = n;
+                        gid_s[sl] = -1;  // RLE exhausted: no group
+                        return;
+                    }
+                    uint64_t code = rd_bits_be(payload, bit[sl], gb[sl].width);
+                    bit[sl] += gb[sl].width;
+                    uint64_t cnt = rd_bits_be(payload, bit[sl], gb[sl].width);
+                    bit[sl] += gb[sl].width;
+                    ent[sl] += 2;
+                    uint16_t m = (gmap_in + gb[sl].map_off)[code < 256 ? code : 0];
+                    gid_s[sl] = m == 0xFFFFu ? -1 : (int64_t)m;
+                    run_hi_s[sl] += (int64_t)cnt;
+                };
+                for (int sl = 0; sl < n_gslots; sl++)
+                    if (gb[sl].uniform_gid == GID_VARYING) advance(sl);
                 int64_t lo = 0;
-                for (uint32_t e = 0; e + 1 < gb.nentries && lo < n; e += 2) {
-                    uint64_t code = rd_bits_be(payload, bit, gb.width);
-                    bit += gb.width;
-                    uint64_t cnt = rd_bits_be(payload, bit, gb.width);
-                    bit += gb.width;
-                    int64_t hi = lo + (int64_t)cnt - 1;
+                while (lo < n) {
+                    int64_t hi = n;
+                    for (int sl = 0; sl < n_gslots; sl++)
+                        hi = run_hi_s[sl] < hi ? run_hi_s[sl] : hi;
                     int64_t aa = lo > r0 ? lo : r0;
-                    int64_t bb2 = hi < r1 ? hi : r1;
-                    lo = hi + 1;
-                    if (aa > bb2) continue;
-                    uint16_t m = map16[code < 256 ? code : 0];
-                    if (m == 0xFFFFu) continue;
-                    uint64_t rsum;
-                    int64_t rmn, rmx;
-                    bool rhave;
-                    fold_range(gstream, bd->field_enc, bd->field_first,
-                               bd->field_len, n, aa, bb2,
-                               EN_VALUES && (flags & KF_NEED_VALUES), lane,
-                               &rsum, &rmn, &rmx, &rhave, derr, (uint64_t)bi);
-                    if ((int64_t)m != cur_group) {
-                        flush_partial(partials, cur_group, wsum, wcnt, wmin,
-                                      wmax, wsumf, lane);
-                        cur_group = (int64_t)m;
-                        wsum = 0; wcnt = 0; wmin = INT64_MAX; wmax = INT64_MIN;
-                        wsumf = 0;
+                    int64_t bb2 = (hi - 1) < r1 ? (hi - 1) : r1;
+                    int64_t comp = 0;
+                    bool ok = true;
+                    for (int sl = 0; sl < n_gslots; sl++) {
+                        if (gid_s[sl] < 0) ok = false;
+                        else comp += gid_s[sl] * gmul[sl];
                     }
-                    wsum += rsum;
-                    wcnt += (uint64_t)(bb2 - aa + 1);
-                    if (rhave) {
-                        wmin = rmn < wmin ? rmn : wmin;
-                        wmax = rmx > wmax ? rmx : wmax;
+                    if (ok && aa <= bb2) {
+                        uint64_t rsum;
+                        int64_t rmn, rmx;
+                        bool rhave;
+                        fold_range(gstream, bd->field_enc, bd->field_first,
+                                   bd->field_len, n, aa, bb2,
+                                   EN_VALUES && (flags & KF_NEED_VALUES), lane,
+                                   &rsum, &rmn, &rmx, &rhave, derr,
+                                   (uint64_t)bi);
+                        if (comp != cur_group) {
+                            flush_partial(partials, cur_group, wsum, wcnt,
+                                          wmin, wmax, wsumf, lane);
+                            cur_group = comp;
+                            wsum = 0; wcnt = 0; wmin = INT64_MAX;
+                            wmax = INT64_MIN; wsumf = 0;
+                        }
+                        wsum += rsum;
+                        wcnt += (uint64_t)(bb2 - aa + 1);
+                        if (rhave) {
+                            wmin = rmn < wmin ? rmn : wmin;
+                            wmax = rmx > wmax ? rmx : wmax;
+                        }
+                        if (flags & KF_FLOAT) wsumf += (double)(int64_t)rsum;
                     }
-                    if (flags & KF_FLOAT) wsumf += (double)(int64_t)rsum;
+                    for (int sl = 0; sl < n_gslots; sl++)
+                        if (run_hi_s[sl] == hi && gb[sl].uniform_gid == GID_VARYING)
+                            advance(sl);
+                    if (hi <= lo) break;  // safety against stuck cursors
+                    lo = hi;
                 }
                 continue;
             }
@@ -1688,18 +1745,20 @@ struct bydb_session {
     SegEntry *d_segs = nullptr;
     int64_t segs_cap = 0;
     bool segs_built = false;
-    // per-row group-by on a dictionary tag
-    int group_slot = -1;               // -1 = group by block group_code
-    GroupBlock *d_groups = nullptr;
-    uint16_t *d_gmap = nullptr;
+    // per-row group-by on dictionary tags (composite keys over <=3 slots)
+    int n_gslots = 0;                  // 0 = group by block group_code
+    int gslots[3] = {-1, -1, -1};
+    int64_t gmul[3] = {1, 1, 1};
+    GroupBlock *d_groups = nullptr;    // [slot][block]
+    uint16_t *d_gmap = nullptr;        // [slot][block][256]
     int64_t groups_cap = 0;
     bool groups_built = false;
-    uint8_t *d_dom_blob = nullptr;
-    uint64_t *d_dom_offs = nullptr;
-    uint64_t *d_dom_hashes = nullptr;
-    uint32_t *d_dom_gids = nullptr;
-    uint32_t dom_table_size = 0;
-    uint32_t dom_n = 0;
+    uint8_t *d_dom_blob[3] = {};
+    uint64_t *d_dom_offs[3] = {};
+    uint64_t *d_dom_hashes[3] = {};
+    uint32_t *d_dom_gids[3] = {};
+    uint32_t dom_table_size[3] = {};
+    uint32_t dom_n[3] = {};
     uint8_t *d_pred_bytes = nullptr;
     uint64_t pred_bytes_cap = 0;
     float last_ms = 0.0f;
@@ -1753,10 +1812,12 @@ extern "C" void bydb_session_destroy(bydb_session *s) {
     if (s->d_segs) (void)hipFree(s->d_segs);
     if (s->d_groups) (void)hipFree(s->d_groups);
     if (s->d_gmap) (void)hipFree(s->d_gmap);
-    if (s->d_dom_blob) (void)hipFree(s->d_dom_blob);
-    if (s->d_dom_offs) (void)hipFree(s->d_dom_offs);
-    if (s->d_dom_hashes) (void)hipFree(s->d_dom_hashes);
-    if (s->d_dom_gids) (void)hipFree(s->d_dom_gids);
+    for (int sl = 0; sl < 3; sl++) {
+        if (s->d_dom_blob[sl]) (void)hipFree(s->d_dom_blob[sl]);
+        if (s->d_dom_offs[sl]) (void)hipFree(s->d_dom_offs[sl]);
+        if (s->d_dom_hashes[sl]) (void)hipFree(s->d_dom_hashes[sl]);
+        if (s->d_dom_gids[sl]) (void)hipFree(s->d_dom_gids[sl]);
+    }
     if (s->d_err) (void)hipFree(s->d_err);
     if (s->ev_start) (void)hipEventDestroy(s->ev_start);
     if (s->ev_stop) (void)hipEventDestroy(s->ev_stop);
@@ -1829,29 +1890,14 @@ extern "C" int bydb_agg_configure(bydb_session *s, int field_vtype,
         s->partials_cap = n_groups;
     }
     s->d_acc = s->d_partials;
-    s->group_slot = -1;
+    s->n_gslots = 0;
     return bydb_reset(s);
 }
 
-// Group rows by the dictionary tag in `slot`: the host supplies the group
-// DOMAIN (dense gid = index into the value list, mirroring the reference's
-// first-seen group materialisation order, which the host layer controls —
-// computeKey semantics, aggregation.go:523).  Values: concatenated blob +
-// n+1 offsets.  n_groups = n domain values.
-extern "C" int bydb_agg_configure_by_tag(bydb_session *s, int field_vtype,
-                                         uint32_t func_mask, int tag_slot,
-                                         const uint8_t *dom_blob,
-                                         const uint64_t *dom_offs,
-                                         uint32_t n_values, int mode) {
-    if (tag_slot < 0 || tag_slot > 2 || n_values < 1 || n_values > 65000) {
-        s->err = "bad tag slot or domain size";
-        return BYDB_ERR_BAD_ARG;
-    }
-    int rc = bydb_agg_configure(s, field_vtype, func_mask, n_values, mode);
-    if (rc != BYDB_OK) return rc;
-    s->group_slot = tag_slot;
-    s->groups_built = false;
-    // host-side open-addressing table: slot -> (fnv1a hash, gid)
+// Load one slot's group domain onto the device (open-addressing table,
+// equality by bytes).
+static int load_domain(bydb_session *s, int pos, const uint8_t *dom_blob,
+                       const uint64_t *dom_offs, uint32_t n_values) {
     uint32_t tsize = 4;
     while (tsize < n_values * 4) tsize <<= 1;
     std::vector<uint64_t> hashes(tsize, 0);
@@ -1866,24 +1912,71 @@ extern "C" int bydb_agg_configure_by_tag(bydb_session *s, int field_vtype,
         gids[sl] = g;
     }
     uint64_t blob_len = dom_offs[n_values];
-    if (s->d_dom_blob) { (void)hipFree(s->d_dom_blob); s->d_dom_blob = nullptr; }
-    if (s->d_dom_offs) { (void)hipFree(s->d_dom_offs); s->d_dom_offs = nullptr; }
-    if (s->d_dom_hashes) { (void)hipFree(s->d_dom_hashes); s->d_dom_hashes = nullptr; }
-    if (s->d_dom_gids) { (void)hipFree(s->d_dom_gids); s->d_dom_gids = nullptr; }
-    HIP_TRY(s, hipMalloc(&s->d_dom_blob, blob_len ? blob_len : 1));
-    HIP_TRY(s, hipMalloc(&s->d_dom_offs, sizeof(uint64_t) * (n_values + 1)));
-    HIP_TRY(s, hipMalloc(&s->d_dom_hashes, sizeof(uint64_t) * tsize));
-    HIP_TRY(s, hipMalloc(&s->d_dom_gids, sizeof(uint32_t) * tsize));
-    HIP_TRY(s, hipMemcpy(s->d_dom_blob, dom_blob, blob_len, hipMemcpyHostToDevice));
-    HIP_TRY(s, hipMemcpy(s->d_dom_offs, dom_offs, sizeof(uint64_t) * (n_values + 1),
-                         hipMemcpyHostToDevice));
-    HIP_TRY(s, hipMemcpy(s->d_dom_hashes, hashes.data(), sizeof(uint64_t) * tsize,
-                         hipMemcpyHostToDevice));
-    HIP_TRY(s, hipMemcpy(s->d_dom_gids, gids.data(), sizeof(uint32_t) * tsize,
-                         hipMemcpyHostToDevice));
-    s->dom_table_size = tsize;
-    s->dom_n = n_values;
+    if (s->d_dom_blob[pos]) { (void)hipFree(s->d_dom_blob[pos]); s->d_dom_blob[pos] = nullptr; }
+    if (s->d_dom_offs[pos]) { (void)hipFree(s->d_dom_offs[pos]); s->d_dom_offs[pos] = nullptr; }
+    if (s->d_dom_hashes[pos]) { (void)hipFree(s->d_dom_hashes[pos]); s->d_dom_hashes[pos] = nullptr; }
+    if (s->d_dom_gids[pos]) { (void)hipFree(s->d_dom_gids[pos]); s->d_dom_gids[pos] = nullptr; }
+    HIP_TRY(s, hipMalloc(&s->d_dom_blob[pos], blob_len ? blob_len : 1));
+    HIP_TRY(s, hipMalloc(&s->d_dom_offs[pos], sizeof(uint64_t) * (n_values + 1)));
+    HIP_TRY(s, hipMalloc(&s->d_dom_hashes[pos], sizeof(uint64_t) * tsize));
+    HIP_TRY(s, hipMalloc(&s->d_dom_gids[pos], sizeof(uint32_t) * tsize));
+    HIP_TRY(s, hipMemcpy(s->d_dom_blob[pos], dom_blob, blob_len, hipMemcpyHostToDevice));
+    HIP_TRY(s, hipMemcpy(s->d_dom_offs[pos], dom_offs,
+                         sizeof(uint64_t) * (n_values + 1), hipMemcpyHostToDevice));
+    HIP_TRY(s, hipMemcpy(s->d_dom_hashes[pos], hashes.data(),
+                         sizeof(uint64_t) * tsize, hipMemcpyHostToDevice));
+    HIP_TRY(s, hipMemcpy(s->d_dom_gids[pos], gids.data(),
+                         sizeof(uint32_t) * tsize, hipMemcpyHostToDevice));
+    s->dom_table_size[pos] = tsize;
+    s->dom_n[pos] = n_values;
     return BYDB_OK;
+}
+
+// Group rows by up to 3 dictionary tags (composite key, computeKey
+// semantics): gid = g0 + n0*g1 + n0*n1*g2; per-slot domains concatenated
+// in blobs/offs (offs holds, per slot, n_i+1 cumulative offsets).
+extern "C" int bydb_agg_configure_by_tags(bydb_session *s, int field_vtype,
+                                          uint32_t func_mask, const int *slots,
+                                          int n_slots,
+                                          const uint8_t *const *dom_blobs,
+                                          const uint64_t *const *dom_offs,
+                                          const uint32_t *n_values, int mode) {
+    if (n_slots < 1 || n_slots > 3) { s->err = "n_slots 1..3"; return BYDB_ERR_BAD_ARG; }
+    uint64_t total = 1;
+    for (int i = 0; i < n_slots; i++) {
+        if (slots[i] < 0 || slots[i] > 2 || n_values[i] < 1) {
+            s->err = "bad slot/domain";
+            return BYDB_ERR_BAD_ARG;
+        }
+        total *= n_values[i];
+    }
+    if (total > 65000) { s->err = "composite group space too large"; return BYDB_ERR_BAD_ARG; }
+    int rc = bydb_agg_configure(s, field_vtype, func_mask, (uint32_t)total, mode);
+    if (rc != BYDB_OK) return rc;
+    s->n_gslots = n_slots;
+    s->groups_built = false;
+    int64_t mul = 1;
+    for (int i = 0; i < n_slots; i++) {
+        s->gslots[i] = slots[i];
+        s->gmul[i] = mul;
+        mul *= n_values[i];
+        rc = load_domain(s, i, dom_blobs[i], dom_offs[i], n_values[i]);
+        if (rc != BYDB_OK) return rc;
+    }
+    return BYDB_OK;
+}
+
+extern "C" int bydb_agg_configure_by_tag(bydb_session *s, int field_vtype,
+                                         uint32_t func_mask, int tag_slot,
+                                         const uint8_t *dom_blob,
+                                         const uint64_t *dom_offs,
+                                         uint32_t n_values, int mode) {
+    const int slots[1] = {tag_slot};
+    const uint8_t *blobs[1] = {dom_blob};
+    const uint64_t *offs[1] = {dom_offs};
+    const uint32_t nv[1] = {n_values};
+    return bydb_agg_configure_by_tags(s, field_vtype, func_mask, slots, 1,
+                                      blobs, offs, nv, mode);
 }
 
 extern "C" int bydb_set_partials_buffer(bydb_session *s, void *dev_ptr,
@@ -1986,32 +2079,35 @@ extern "C" int bydb_consume_multi(bydb_session *s, int64_t min_ts,
         segs = s->d_segs;
     }
     GroupBlock *groups = nullptr;
-    if (s->group_slot >= 0) {
+    if (s->n_gslots > 0) {
         if (s->groups_cap < s->n_blocks) {
             if (s->d_groups) (void)hipFree(s->d_groups);
             if (s->d_gmap) (void)hipFree(s->d_gmap);
             HIP_TRY(s, hipMalloc(&s->d_groups,
-                                 sizeof(GroupBlock) * (size_t)s->n_blocks));
+                                 sizeof(GroupBlock) * 3 * (size_t)s->n_blocks));
             HIP_TRY(s, hipMalloc(&s->d_gmap,
-                                 sizeof(uint16_t) * 256 * (size_t)s->n_blocks));
+                                 sizeof(uint16_t) * 256 * 3 * (size_t)s->n_blocks));
             s->groups_cap = s->n_blocks;
             s->groups_built = false;
         }
         if (!s->groups_built) {
-            GroupDomain dom;
-            dom.blob = s->d_dom_blob;
-            dom.offs = s->d_dom_offs;
-            dom.hashes = s->d_dom_hashes;
-            dom.gids = s->d_dom_gids;
-            dom.table_size = s->dom_table_size;
-            dom.n = s->dom_n;
             int rthreads = 256;
             int rblocks = (int)((s->n_blocks + rthreads - 1) / rthreads);
-            hipLaunchKernelGGL(k_resolve_groups, dim3(rblocks), dim3(rthreads),
-                               0, s->stream, s->d_payload, s->d_blocks,
-                               s->n_blocks, s->group_slot, dom, s->d_groups,
-                               s->d_gmap);
-            HIP_TRY(s, hipGetLastError());
+            for (int i = 0; i < s->n_gslots; i++) {
+                GroupDomain dom;
+                dom.blob = s->d_dom_blob[i];
+                dom.offs = s->d_dom_offs[i];
+                dom.hashes = s->d_dom_hashes[i];
+                dom.gids = s->d_dom_gids[i];
+                dom.table_size = s->dom_table_size[i];
+                dom.n = s->dom_n[i];
+                hipLaunchKernelGGL(
+                    k_resolve_groups, dim3(rblocks), dim3(rthreads), 0,
+                    s->stream, s->d_payload, s->d_blocks, s->n_blocks,
+                    s->gslots[i], dom, s->d_groups + (int64_t)i * s->n_blocks,
+                    s->d_gmap, (uint32_t)((int64_t)i * 256 * s->n_blocks));
+                HIP_TRY(s, hipGetLastError());
+            }
             s->groups_built = true;
         }
         groups = s->d_groups;
@@ -2026,8 +2122,8 @@ extern "C" int bydb_consume_multi(bydb_session *s, int64_t min_ts,
     const bool en_groups = groups != nullptr;
     void (*kfn)(const uint8_t *, const bydb_block_desc *, int64_t, int64_t,
                 int64_t, int, const PredBlock *, int, const SegEntry *,
-                const GroupBlock *, const uint16_t *, int64_t,
-                bydb_partial *, DevErr *);
+                const GroupBlock *, const uint16_t *, int, int64_t, int64_t,
+                int64_t, int64_t, bydb_partial *, DevErr *);
     if (en_values) {
         if (en_preds) kfn = en_groups ? k_scan_agg_t<true, true, true>
                                       : k_scan_agg_t<true, true, false>;
@@ -2042,6 +2138,7 @@ extern "C" int bydb_consume_multi(bydb_session *s, int64_t min_ts,
     hipLaunchKernelGGL(kfn, dim3(grid), dim3(threads), 0, s->stream,
                        s->d_payload, s->d_blocks, s->n_blocks, min_ts, max_ts,
                        flags, preds, n_preds, segs, groups, s->d_gmap,
+                       s->n_gslots, s->gmul[0], s->gmul[1], s->gmul[2],
                        (int64_t)s->n_groups, s->d_acc, s->d_err);
     HIP_TRY(s, hipGetLastError());
     HIP_TRY(s, hipEventRecord(s->ev_stop, s->stream));

@@ -1197,25 +1197,21 @@ int bo_scan_agg_multi(const uint8_t *payload, const bo_block_desc *blocks,
     return rc;
 }
 
-int bo_scan_agg_bytag(const uint8_t *payload, const bo_block_desc *blocks,
-                      int64_t n_blocks, int field_vtype, int64_t min_ts,
-                      int64_t max_ts, int slot, const uint8_t *dom_blob,
-                      const int64_t *dom_lens, int64_t n_dom,
-                      bo_agg_result *out) {
-    for (int64_t g = 0; g < n_dom; g++) agg_reset(&out[g]);
+int bo_scan_agg_bytags(const uint8_t *payload, const bo_block_desc *blocks,
+                       int64_t n_blocks, int field_vtype, int64_t min_ts,
+                       int64_t max_ts, const int *slots, int n_slots,
+                       const uint8_t *const *dom_blobs,
+                       const int64_t *const *dom_lens, const int64_t *n_doms,
+                       bo_agg_result *out) {
+    int64_t total = 1;
+    for (int i = 0; i < n_slots; i++) total *= n_doms[i];
+    for (int64_t g = 0; g < total; g++) agg_reset(&out[g]);
     int64_t *ts_buf = (int64_t *)malloc(sizeof(int64_t) * MAX_BLOCK_ROWS);
     int64_t *i64_buf = (int64_t *)malloc(sizeof(int64_t) * MAX_BLOCK_ROWS);
     double *f64_buf = (double *)malloc(sizeof(double) * MAX_BLOCK_ROWS);
     uint8_t *tagdata = (uint8_t *)malloc((size_t)1 << 24);
     int64_t *taglen = (int64_t *)malloc(sizeof(int64_t) * MAX_BLOCK_ROWS);
-    const uint8_t **dvals = (const uint8_t **)malloc(sizeof(void *) * (size_t)n_dom);
-    {
-        const uint8_t *p = dom_blob;
-        for (int64_t g = 0; g < n_dom; g++) {
-            dvals[g] = p;
-            p += dom_lens[g] > 0 ? dom_lens[g] : 0;
-        }
-    }
+    int64_t *rowgid = (int64_t *)malloc(sizeof(int64_t) * MAX_BLOCK_ROWS);
     int rc = BO_OK;
     for (int64_t i = 0; i < n_blocks && rc == BO_OK; i++) {
         const bo_block_desc *b = &blocks[i];
@@ -1227,38 +1223,57 @@ int bo_scan_agg_bytag(const uint8_t *payload, const bo_block_desc *blocks,
         if (rc != BO_OK) break;
         int64_t r0, r1;
         if (!bo_find_range(ts_buf, n, min_ts, max_ts, &r0, &r1)) continue;
-        uint64_t toff = slot == 0 ? b->tag_off : slot == 1 ? b->tag2_off : b->tag3_off;
-        uint64_t tlen = slot == 0 ? b->tag_len : slot == 1 ? b->tag2_len : b->tag3_len;
-        if (tlen == 0) continue;  /* nil tag rows: no group */
-        if (payload[toff] != BO_ENC_DICTIONARY) { rc = BO_ERR_BAD_TYPE; break; }
-        size_t tdl = 0;
-        rc = bo_dictionary_decode(tagdata, (size_t)1 << 24, taglen,
-                                  payload + toff + 1, tlen - 1, n, &tdl);
-        if (rc != BO_OK) break;
-        if (field_vtype == BO_VT_INT64) {
-            rc = bo_column_i64_decode(i64_buf, payload + b->col_off, b->col_len, n);
-        } else if (field_vtype == BO_VT_FLOAT64) {
-            rc = bo_column_f64_decode(f64_buf, payload + b->col_off, b->col_len, n);
-        } else {
-            rc = BO_ERR_BAD_TYPE;
+        for (int64_t r = 0; r < n; r++) rowgid[r] = 0;
+        int64_t mul = 1;
+        int dropped = 0;
+        for (int si = 0; si < n_slots && rc == BO_OK && !dropped; si++) {
+            int slot = slots[si];
+            uint64_t toff = slot == 0 ? b->tag_off : slot == 1 ? b->tag2_off : b->tag3_off;
+            uint64_t tlen = slot == 0 ? b->tag_len : slot == 1 ? b->tag2_len : b->tag3_len;
+            if (tlen == 0) { dropped = 1; break; }
+            if (payload[toff] != BO_ENC_DICTIONARY) { rc = BO_ERR_BAD_TYPE; break; }
+            size_t tdl = 0;
+            rc = bo_dictionary_decode(tagdata, (size_t)1 << 24, taglen,
+                                      payload + toff + 1, tlen - 1, n, &tdl);
+            if (rc != BO_OK) break;
+            const uint8_t *dv = dom_blobs[si];
+            const int64_t *dl = dom_lens[si];
+            int64_t nd = n_doms[si];
+            size_t tago = 0;
+            for (int64_t r = 0; r < n; r++) {
+                int64_t tl = taglen[r];
+                size_t my = tago;
+                if (tl > 0) tago += (size_t)tl;
+                if (rowgid[r] < 0) continue;
+                int64_t gid = -1;
+                if (tl >= 0) {
+                    const uint8_t *p = dv;
+                    for (int64_t g = 0; g < nd; g++) {
+                        int64_t gl = dl[g] < 0 ? -1 : dl[g];
+                        if (gl == tl &&
+                            (tl == 0 || memcmp(p, tagdata + my, (size_t)tl) == 0)) {
+                            gid = g;
+                            break;
+                        }
+                        if (gl > 0) p += gl;
+                    }
+                }
+                if (gid < 0) rowgid[r] = -1;
+                else rowgid[r] += gid * mul;
+            }
+            mul *= nd;
         }
         if (rc != BO_OK) break;
-        size_t tago = 0;
-        for (int64_t r = 0; r < n; r++) {
-            int64_t tl = taglen[r];
-            size_t my = tago;
-            if (tl > 0) tago += (size_t)tl;
-            if (r < r0 || r > r1 || tl < 0) continue;
-            int64_t gid = -1;
-            for (int64_t g = 0; g < n_dom; g++) {
-                if (dom_lens[g] == tl &&
-                    (tl == 0 || memcmp(dvals[g], tagdata + my, (size_t)tl) == 0)) {
-                    gid = g;
-                    break;
-                }
-            }
-            if (gid < 0) continue;
-            bo_agg_result *o = &out[gid];
+        if (dropped) continue;
+        if (field_vtype == BO_VT_INT64)
+            rc = bo_column_i64_decode(i64_buf, payload + b->col_off, b->col_len, n);
+        else if (field_vtype == BO_VT_FLOAT64)
+            rc = bo_column_f64_decode(f64_buf, payload + b->col_off, b->col_len, n);
+        else rc = BO_ERR_BAD_TYPE;
+        if (rc != BO_OK) break;
+        for (int64_t r = r0; r <= r1; r++) {
+            if (rowgid[r] < 0) continue;
+            bo_agg_result *o = &out[rowgid[r]];
             if (field_vtype == BO_VT_INT64) {
                 int64_t v = i64_buf[r];
                 o->sum_i = (int64_t)((uint64_t)o->sum_i + (uint64_t)v);
@@ -1275,8 +1290,21 @@ int bo_scan_agg_bytag(const uint8_t *payload, const bo_block_desc *blocks,
         }
     }
     free(ts_buf); free(i64_buf); free(f64_buf); free(tagdata); free(taglen);
-    free((void *)dvals);
+    free(rowgid);
     return rc;
+}
+
+int bo_scan_agg_bytag(const uint8_t *payload, const bo_block_desc *blocks,
+                      int64_t n_blocks, int field_vtype, int64_t min_ts,
+                      int64_t max_ts, int slot, const uint8_t *dom_blob,
+                      const int64_t *dom_lens, int64_t n_dom,
+                      bo_agg_result *out) {
+    const int slots[1] = {slot};
+    const uint8_t *blobs[1] = {dom_blob};
+    const int64_t *lens[1] = {dom_lens};
+    const int64_t nd[1] = {n_dom};
+    return bo_scan_agg_bytags(payload, blocks, n_blocks, field_vtype, min_ts,
+                              max_ts, slots, 1, blobs, lens, nd, out);
 }
 
 /* ===================== xxhash64 =====================

@@ -198,6 +198,12 @@ int bo_scan_agg_multi(const uint8_t *payload, const bo_block_desc *blocks,
  * tag value in the domain list (host-controlled order = the reference's
  * first-seen materialisation, computeKey aggregation.go:523).  Rows with
  * nil or out-of-domain tags are dropped.  slot: 0..2. */
+int bo_scan_agg_bytags(const uint8_t *payload, const bo_block_desc *blocks,
+                       int64_t n_blocks, int field_vtype, int64_t min_ts,
+                       int64_t max_ts, const int *slots, int n_slots,
+                       const uint8_t *const *dom_blobs,
+                       const int64_t *const *dom_lens, const int64_t *n_doms,
+                       bo_agg_result *out);
 int bo_scan_agg_bytag(const uint8_t *payload, const bo_block_desc *blocks,
                       int64_t n_blocks, int field_vtype, int64_t min_ts,
                       int64_t max_ts, int slot, const uint8_t *dom_blob,

@@ -130,6 +130,11 @@ _L.bo_scan_agg_grouped.restype = C.c_int
 _L.bo_scan_agg_grouped.argtypes = [u8p, C.POINTER(BlockDesc), C.c_int64, C.c_int,
                                    C.c_int64, C.c_int64, u8p, C.c_int64,
                                    C.POINTER(AggResult), C.c_int64]
+_L.bo_scan_agg_bytags.restype = C.c_int
+_L.bo_scan_agg_bytags.argtypes = [u8p, C.POINTER(BlockDesc), C.c_int64, C.c_int,
+                                  C.c_int64, C.c_int64, C.POINTER(C.c_int),
+                                  C.c_int, C.POINTER(u8p), C.POINTER(i64p),
+                                  i64p, C.POINTER(AggResult)]
 _L.bo_scan_agg_bytag.restype = C.c_int
 _L.bo_scan_agg_bytag.argtypes = [u8p, C.POINTER(BlockDesc), C.c_int64, C.c_int,
                                  C.c_int64, C.c_int64, C.c_int, u8p, i64p,
@@ -405,6 +410,38 @@ def scan_agg_bytag(payload: bytes, blocks, field_vtype, slot, domain,
                               len(domain), out)
     if rc != 0:
         raise ValueError(f"scan_agg_bytag rc={rc}")
+    return list(out)
+
+
+def scan_agg_bytags(payload: bytes, blocks, field_vtype, slots, domains,
+                    min_ts=INT64_MIN, max_ts=INT64_MAX):
+    """Composite group-by over multiple tag slots; gid = g0 + n0*g1 + ..."""
+    descs = (BlockDesc * len(blocks))()
+    for i, b in enumerate(blocks):
+        for k, v in b.items():
+            setattr(descs[i], k, v)
+    total = 1
+    blobs, lens_arrs, nds = [], [], []
+    bufs = []
+    for dom in domains:
+        blob = b"".join(v for v in dom if v is not None)
+        lens = [(-1 if v is None else len(v)) for v in dom]
+        buf = (C.c_uint8 * max(len(blob), 1)).from_buffer_copy(blob or b"\0")
+        la = (C.c_int64 * len(dom))(*lens)
+        bufs.append((buf, la))
+        blobs.append(C.cast(buf, u8p))
+        lens_arrs.append(C.cast(la, i64p))
+        nds.append(len(dom))
+        total *= len(dom)
+    out = (AggResult * total)()
+    src = (C.c_uint8 * max(len(payload), 1)).from_buffer_copy(payload or b"\0")
+    rc = _L.bo_scan_agg_bytags(
+        src, descs, len(blocks), field_vtype, min_ts, max_ts,
+        (C.c_int * len(slots))(*slots), len(slots),
+        (u8p * len(domains))(*blobs), (i64p * len(domains))(*lens_arrs),
+        (C.c_int64 * len(domains))(*nds), out)
+    if rc != 0:
+        raise ValueError(f"scan_agg_bytags rc={rc}")
     return list(out)
 
 

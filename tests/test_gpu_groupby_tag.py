@@ -148,3 +148,71 @@ def test_float_grouping():
         if oc.count:
             assert g.min_f == oc.min_f and g.max_f == oc.max_f
             assert math.isclose(g.sum_f, oc.sum_f, rel_tol=1e-9)
+
+
+def test_composite_two_tag_grouping():
+    """Composite (env, region) key: gid = env + 4*region."""
+    rng = random.Random(61)
+    regions = [b"r0", b"r1", b"r2"]
+    b = PartBuilder()
+    for sid in range(10):
+        n = 3000
+        ts = [T0 + i * MS for i in range(n)]
+        b.add_block_i64(sid + 1, ts, [1] * n,
+                        [rng.randint(-10**6, 10**6) for _ in range(n)])
+        # slot 0: row-varying env; slot 1: entity region
+        tags = []
+        while len(tags) < n:
+            run = min(rng.randint(1, 120), n - len(tags))
+            tags.extend([ENVS[rng.randrange(4)]] * run)
+        b.set_block_tag(tags)
+        b.set_block_tag([regions[sid % 3]] * n)
+    payload, blocks = oracle_blocks(b)
+    orc = o.scan_agg_bytags(payload, blocks, VT_INT64, [0, 1],
+                            [ENVS, regions])
+    s = Session(0)
+    s.upload_part(b)
+    s.configure_by_tags(VT_INT64, [AGG_SUM, AGG_COUNT, AGG_MIN, AGG_MAX],
+                        [0, 1], [ENVS, regions])
+    s.consume()
+    gs = s.finalize()
+    s.close()
+    assert len(gs) == 12
+    assert sum(oc.count for oc in orc) == 10 * 3000
+    for g, oc in zip(gs, orc):
+        assert g.count == oc.count
+        assert g.sum_i == oc.sum_i
+        if oc.count:
+            assert g.min_i == oc.min_i and g.max_i == oc.max_i
+
+
+def test_composite_both_rowvarying():
+    rng = random.Random(62)
+    regions = [b"x", b"yy"]
+    b = PartBuilder()
+    for sid in range(6):
+        n = 2500
+        ts = [T0 + i * MS for i in range(n)]
+        b.add_block_i64(sid + 1, ts, [1] * n,
+                        [rng.randint(0, 9999) for _ in range(n)])
+        for table, maxrun in ((ENVS, 90), (regions, 130)):
+            tags = []
+            while len(tags) < n:
+                run = min(rng.randint(1, maxrun), n - len(tags))
+                v = None if rng.random() < 0.08 else table[rng.randrange(len(table))]
+                tags.extend([v] * run)
+            b.set_block_tag(tags)
+    payload, blocks = oracle_blocks(b)
+    orc = o.scan_agg_bytags(payload, blocks, VT_INT64, [0, 1],
+                            [ENVS, regions])
+    s = Session(0)
+    s.upload_part(b)
+    s.configure_by_tags(VT_INT64, [AGG_SUM, AGG_COUNT], [0, 1],
+                        [ENVS, regions])
+    s.consume()
+    gs = s.finalize()
+    s.close()
+    assert sum(oc.count for oc in orc) > 0
+    for g, oc in zip(gs, orc):
+        assert g.count == oc.count
+        assert g.sum_i == oc.sum_i
