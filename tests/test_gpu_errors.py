@@ -1,0 +1,100 @@
+"""Loud error paths: corrupt or unsupported inputs must surface as device
+errors at finalize, never silently wrong results."""
+import ctypes as C
+
+import pytest
+
+import banyandb_amd as ba
+from banyandb_amd import PartBuilder, Session, VT_INT64, AGG_SUM, AGG_COUNT
+
+pytestmark = pytest.mark.gpu
+
+T0 = 1_700_000_000_000_000_000
+MS = 10 ** 6
+
+
+def _desc_copy(d):
+    new = ba.BlockDesc()
+    C.memmove(C.byref(new), C.byref(d), C.sizeof(ba.BlockDesc))
+    return new
+
+
+def _upload_raw(payload: bytes, descs):
+    s = Session(0)
+    lib = ba.lib()
+    s._ck(lib.bydb_part_reserve(s._h, len(payload), len(descs)))
+    buf = (C.c_uint8 * len(payload)).from_buffer_copy(payload)
+    arr = (ba.BlockDesc * len(descs))(*descs)
+    s._ck(lib.bydb_part_append(s._h, buf, len(payload), arr, len(descs)))
+    return s
+
+
+def test_unsupported_encode_type_fails_loud():
+    b = PartBuilder()
+    ts = [T0 + i * MS for i in range(64)]
+    b.add_block_i64(1, ts, [1] * 64, list(range(0, 6400, 100)))
+    d = _desc_copy(b.blocks()[0])
+    d.field_enc = 9  # Plain — numeric fold does not support it
+    s = _upload_raw(b.payload, [d])
+    s.configure(VT_INT64, [AGG_SUM, AGG_COUNT])
+    s.consume()
+    with pytest.raises(RuntimeError, match="decode error"):
+        s.finalize()
+    s.close()
+
+
+def test_descending_timestamps_fail_loud():
+    b = PartBuilder()
+    ts = [T0 + i * MS for i in range(64)]
+    b.add_block_i64(1, ts, [1] * 64, list(range(64)))
+    d = _desc_copy(b.blocks()[0])
+    d.ts_min, d.ts_max = d.ts_max, d.ts_min  # descending metadata
+    s = _upload_raw(b.payload, [d])
+    s.configure(VT_INT64, [AGG_SUM, AGG_COUNT])
+    s.consume()
+    with pytest.raises(RuntimeError, match="decode error"):
+        s.finalize()
+    s.close()
+
+
+def test_corrupt_varint_stream_fails_loud():
+    b = PartBuilder()
+    import random
+    rng = random.Random(3)
+    ts = [T0 + i * MS for i in range(300)]
+    b.add_block_i64(1, ts, [1] * 300,
+                    [rng.randint(-2**61, 2**61) for _ in range(300)])
+    d = b.blocks()[0]
+    payload = bytearray(b.payload)
+    # make a 64-byte run of continuation bytes inside the field stream
+    for k in range(70):
+        payload[d.field_off + 16 + k] |= 0x80
+    s = _upload_raw(bytes(payload), [_desc_copy(d)])
+    s.configure(VT_INT64, [AGG_SUM, AGG_COUNT])
+    s.consume()
+    with pytest.raises(RuntimeError, match="decode error"):
+        s.finalize()
+    s.close()
+
+
+def test_sticky_error_then_reset_recovers():
+    b = PartBuilder()
+    ts = [T0 + i * MS for i in range(64)]
+    b.add_block_i64(1, ts, [1] * 64, [7] * 64)
+    d_bad = _desc_copy(b.blocks()[0])
+    d_bad.field_enc = 9
+    s = _upload_raw(b.payload, [d_bad])
+    s.configure(VT_INT64, [AGG_SUM, AGG_COUNT])
+    s.consume()
+    with pytest.raises(RuntimeError):
+        s.finalize()
+    # a reset + good part recovers the session
+    s._ck(ba.lib().bydb_part_reserve(s._h, b.payload_len, 1))
+    buf = (C.c_uint8 * b.payload_len).from_buffer_copy(b.payload)
+    arr = (ba.BlockDesc * 1)(_desc_copy(b.blocks()[0]))
+    s._ck(ba.lib().bydb_part_append(s._h, buf, b.payload_len, arr, 1))
+    s.reset()
+    s.consume()
+    g = s.finalize()[0]
+    assert g.count == 64 and g.sum_i == 7 * 64
+    s.close()
