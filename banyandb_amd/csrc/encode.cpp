@@ -228,10 +228,12 @@ inline void cell_append(std::vector<uint8_t> &dst, int64_t v) {
 // compressBlock plain path only (<128 B); zstd framing via libzstd at need.
 extern "C" {
 typedef size_t (*zstd_compress_fn)(void *, size_t, const void *, size_t, int);
+typedef size_t (*zstd_decompress_fn)(void *, size_t, const void *, size_t);
 typedef unsigned (*zstd_iserr_fn)(size_t);
 }
 #include <dlfcn.h>
 zstd_compress_fn p_zstd_compress = nullptr;
+zstd_decompress_fn p_zstd_decompress = nullptr;
 zstd_iserr_fn p_zstd_iserr = nullptr;
 bool zstd_load() {
     static int loaded = -1;
@@ -240,10 +242,42 @@ bool zstd_load() {
     if (!h) h = dlopen("libzstd.so", RTLD_NOW | RTLD_GLOBAL);
     if (h) {
         p_zstd_compress = (zstd_compress_fn)dlsym(h, "ZSTD_compress");
+        p_zstd_decompress = (zstd_decompress_fn)dlsym(h, "ZSTD_decompress");
         p_zstd_iserr = (zstd_iserr_fn)dlsym(h, "ZSTD_isError");
     }
-    loaded = (p_zstd_compress && p_zstd_iserr) ? 1 : 0;
+    loaded = (p_zstd_compress && p_zstd_decompress && p_zstd_iserr) ? 1 : 0;
     return loaded;
+}
+
+// decompressBlock (bytes.go:306-345): [0][u8 len][bytes] plain (<128 B) or
+// [1][varuint clen][zstd frame].  Returns bytes consumed in *consumed.
+bool decompress_block_host(std::vector<uint8_t> &out, const uint8_t *src,
+                           uint64_t src_len, uint64_t dst_hint,
+                           uint64_t *consumed) {
+    if (src_len < 2) return false;
+    if (src[0] == 0) {
+        uint64_t n = src[1];
+        if (2 + n > src_len) return false;
+        out.assign(src + 2, src + 2 + n);
+        *consumed = 2 + n;
+        return true;
+    }
+    if (src[0] != 1 || !zstd_load()) return false;
+    uint64_t clen = 0, o = 1;
+    unsigned sh = 0;
+    while (o < src_len) {
+        uint8_t c = src[o++];
+        clen |= (uint64_t)(c & 0x7f) << sh;
+        if (c < 0x80) break;
+        sh += 7;
+    }
+    if (o + clen > src_len) return false;
+    out.resize(dst_hint);
+    size_t dl = p_zstd_decompress(out.data(), out.size(), src + o, clen);
+    if (p_zstd_iserr(dl)) return false;
+    out.resize(dl);
+    *consumed = o + clen;
+    return true;
 }
 
 bool compress_append(std::vector<uint8_t> &dst, const uint8_t *src, size_t n) {
@@ -389,6 +423,45 @@ bool dictionary_append(std::vector<uint8_t> &dst, const uint8_t *data,
 }
 
 }  // namespace
+
+// Host-side normalization of a reference-form Plain tag column
+// (column.go:266-278 fallback: [EncodeTypePlain][bytes block], written when
+// the dictionary bails at >256 distinct values, dictionary.go:58) into the
+// device-parseable form [ENC_PLAIN][u32le nrows][wt][lens BE][payload] —
+// the same decompress-at-part-open the reference performs (zstd.go:49)
+// before row access; the GPU then evaluates the per-row predicate itself.
+bool bydb_normalize_plain_tag(const uint8_t *src, uint64_t src_len,
+                              uint64_t nrows, std::vector<uint8_t> &out) {
+    if (src_len < 1 || src[0] != BYDB_ENC_PLAIN) return false;
+    const uint8_t *p = src + 1;
+    uint64_t rem = src_len - 1, used = 0;
+    std::vector<uint8_t> lens;
+    if (!decompress_block_host(lens, p, rem, 1 + nrows * 8, &used)) return false;
+    p += used;
+    rem -= used;
+    if (lens.empty()) return false;
+    uint8_t wt = lens[0];
+    uint32_t wbytes = wt == 0 ? 1 : wt == 1 ? 2 : wt == 2 ? 4 : 8;
+    if (lens.size() != 1 + (size_t)nrows * wbytes) return false;
+    uint64_t total = 0;
+    for (uint64_t i = 0; i < nrows; i++) {
+        uint64_t ap1 = 0;
+        for (uint32_t b = 0; b < wbytes; b++)
+            ap1 = (ap1 << 8) | lens[1 + i * wbytes + b];
+        if (ap1) total += ap1 - 1;
+    }
+    std::vector<uint8_t> vals;
+    if (!decompress_block_host(vals, p, rem, total, &used)) return false;
+    if (used != rem || vals.size() != total) return false;
+    out.clear();
+    out.reserve(6 + (lens.size() - 1) + vals.size());
+    out.push_back(BYDB_ENC_PLAIN);
+    for (int b = 0; b < 4; b++) out.push_back((uint8_t)(nrows >> (8 * b)));
+    out.push_back(wt);
+    out.insert(out.end(), lens.begin() + 1, lens.end());
+    out.insert(out.end(), vals.begin(), vals.end());
+    return true;
+}
 
 // ===================== part builder =====================
 struct bydb_part_builder {

@@ -40,6 +40,15 @@
 
 #define WAVE 64
 
+// Plain (non-dictionary) tag columns — the >256-distinct-values fallback
+// (column.go:266-278, dictionary.go:58) — are zstd-compressed on disk; the
+// host normalizes them at part registration (encode.cpp) into a sidecar
+// arena, flagged in the stored descriptor by the top offset bit.
+#define TAG_SIDECAR_BIT (1ull << 63)
+#define PLAIN_BM_WORDS 128  // bitmap words per block (8192 rows max)
+bool bydb_normalize_plain_tag(const uint8_t *src, uint64_t src_len,
+                              uint64_t nrows, std::vector<uint8_t> &out);
+
 // ---------------- device helpers ----------------
 
 __device__ __forceinline__ int64_t zz_dec(uint64_t u) {
@@ -497,9 +506,11 @@ struct PredBlock {
     uint64_t rle_bit_off;  // absolute bit offset of packed RLE entries
     uint32_t nentries;     // packed entry count (2 per run: value, count)
     uint8_t width;         // bits per packed entry
-    uint8_t active;        // block has a dictionary tag column
-    uint8_t err;           // unparseable on device (zstd/plain) -> error
+    uint8_t active;        // block has a tag column on this slot
+    uint8_t err;           // unparseable on device -> error
     uint8_t uniform;       // single run covers the whole block
+    uint8_t plain;         // plain (non-dictionary) column: rle_bit_off is
+                           // the per-row match-bitmap WORD offset instead
 };
 
 // MSB-first bit read at an arbitrary bit offset (reader.go:39-79 order);
@@ -516,6 +527,7 @@ __device__ __forceinline__ uint64_t rd_bits_be(const uint8_t *p,
 
 struct PredWalk {
     const uint8_t *payload;
+    const uint64_t *bm;        // plain columns: per-row match bitmap (O(1))
     uint64_t bit0;
     uint32_t nentries, width;
     uint64_t mask[4];
@@ -525,8 +537,10 @@ struct PredWalk {
 };
 
 __device__ __forceinline__ void pred_init(PredWalk *pw, const uint8_t *payload,
-                                          const PredBlock *pb) {
+                                          const PredBlock *pb,
+                                          const uint64_t *bm_arena) {
     pw->payload = payload;
+    pw->bm = pb->plain ? bm_arena + pb->rle_bit_off : nullptr;
     pw->bit0 = pb->rle_bit_off;
     pw->nentries = pb->nentries;
     pw->width = pb->width;
@@ -561,6 +575,10 @@ __device__ __forceinline__ void pred_advance(PredWalk *pw) {
 // lanes get the run covering their row; the walker only moves forward.
 __device__ __forceinline__ bool pred_match_rows(PredWalk *pw, int64_t row,
                                                 bool need) {
+    if (pw->bm) {   // plain column: stateless bitmap lookup, no wave sync
+        if (!need) return false;
+        return (pw->bm[row >> 6] >> (row & 63)) & 1;
+    }
     bool match = false;
     while (true) {
         bool mine = need && row >= pw->run_lo && row < pw->run_hi;
@@ -653,6 +671,11 @@ __global__ void k_resolve_groups(const uint8_t *__restrict__ payload,
     gb.rle_bit_off = 0; gb.nentries = 0; gb.width = 0; gb.err = 0;
     gb._p[0] = gb._p[1] = 0;
     if (tlen == 0) { out[bi] = gb; return; }
+    if (toff & TAG_SIDECAR_BIT) {  // plain column: group-by unsupported (v1)
+        gb.err = 1;
+        out[bi] = gb;
+        return;
+    }
     const uint8_t *p = payload + toff;
     const uint8_t *end = p + tlen;
     if (*p != BYDB_ENC_DICTIONARY) { gb.err = 1; out[bi] = gb; return; }
@@ -716,10 +739,12 @@ __global__ void k_resolve_groups(const uint8_t *__restrict__ payload,
 // parseable on device (bytes.go:291-303); zstd-compressed dictionaries are
 // flagged and surfaced as a device error if a predicate needs them.
 __global__ void k_resolve_pred(const uint8_t *__restrict__ payload,
+                               const uint8_t *__restrict__ sidecar,
                                const bydb_block_desc *__restrict__ blocks,
                                int64_t n_blocks, const uint8_t *__restrict__ pred,
                                uint64_t pred_len, int slot,
-                               PredBlock *__restrict__ out) {
+                               PredBlock *__restrict__ out,
+                               uint32_t *__restrict__ plain_ctr) {
     int64_t bi = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (bi >= n_blocks) return;
     const bydb_block_desc *bd = &blocks[bi];
@@ -728,10 +753,21 @@ __global__ void k_resolve_pred(const uint8_t *__restrict__ payload,
     PredBlock pb;
     for (int i = 0; i < 4; i++) pb.mask[i] = 0;
     pb.rle_bit_off = 0; pb.nentries = 0; pb.width = 0;
-    pb.active = 0; pb.err = 0; pb.uniform = 0;
+    pb.active = 0; pb.err = 0; pb.uniform = 0; pb.plain = 0;
     if (tlen == 0) { out[bi] = pb; return; }
-    const uint8_t *p = payload + toff;
+    const uint8_t *p = (toff & TAG_SIDECAR_BIT)
+                           ? sidecar + (toff & ~TAG_SIDECAR_BIT)
+                           : payload + toff;
     const uint8_t *end = p + tlen;
+    if (*p == BYDB_ENC_PLAIN) {
+        // host-normalized plain column: a second pass (k_resolve_plain)
+        // fills this block's match bitmap; count arena slots here
+        pb.active = 1;
+        pb.plain = 1;
+        atomicAdd(plain_ctr, 1u);
+        out[bi] = pb;
+        return;
+    }
     if (*p != BYDB_ENC_DICTIONARY) { pb.err = 1; out[bi] = pb; return; }
     p++;
     pb.active = 1;
@@ -787,6 +823,69 @@ __global__ void k_resolve_pred(const uint8_t *__restrict__ payload,
         if (cnt >= bd->count) pb.uniform = 1;
     }
     out[bi] = pb;
+}
+
+// Second resolve pass for plain (non-dictionary) tag columns: one wave per
+// block evaluates the equality predicate on every row of the normalized
+// stream ([ENC_PLAIN][u32le n][wt][lens BE][payload]) and writes a per-row
+// match bitmap.  The residual-predicate pushdown the reference performs
+// row-by-row on decoded [][]byte (aggregation.go:310 over column.go:410
+// Plain columns) becomes 64 rows per step with a wave prefix-scan over the
+// value lengths for the payload offsets.
+__global__ __launch_bounds__(WAVE) void k_resolve_plain(
+    const uint8_t *__restrict__ payload, const uint8_t *__restrict__ sidecar,
+    const bydb_block_desc *__restrict__ blocks, int64_t n_blocks,
+    const uint8_t *__restrict__ pred, uint64_t pred_len, int slot,
+    PredBlock *__restrict__ out, uint64_t *__restrict__ bm,
+    uint32_t *__restrict__ ctr) {
+    const int lane = threadIdx.x;
+    for (int64_t bi = blockIdx.x; bi < n_blocks; bi += gridDim.x) {
+        PredBlock pb = out[bi];
+        if (!pb.plain || pb.err) continue;
+        const bydb_block_desc *bd = &blocks[bi];
+        uint64_t toff = slot == 0 ? bd->tag_off
+                                  : slot == 1 ? bd->tag2_off : bd->tag3_off;
+        const uint8_t *p = (toff & TAG_SIDECAR_BIT)
+                               ? sidecar + (toff & ~TAG_SIDECAR_BIT)
+                               : payload + toff;
+        p++;  // ENC_PLAIN
+        uint64_t n = (uint64_t)p[0] | ((uint64_t)p[1] << 8) |
+                     ((uint64_t)p[2] << 16) | ((uint64_t)p[3] << 24);
+        p += 4;
+        uint8_t wt = *p++;
+        uint32_t wbytes = wt == 0 ? 1 : wt == 1 ? 2 : wt == 2 ? 4 : 8;
+        const uint8_t *lens = p;
+        const uint8_t *vals = lens + n * wbytes;
+        if (n != (uint64_t)bd->count || n > (uint64_t)PLAIN_BM_WORDS * 64) {
+            if (lane == 0) out[bi].err = 1;
+            continue;
+        }
+        uint32_t slot_id = 0;
+        if (lane == 0) slot_id = atomicAdd(ctr, 1u);
+        slot_id = (uint32_t)__shfl((int)slot_id, 0);
+        uint64_t w0 = (uint64_t)slot_id * PLAIN_BM_WORDS;
+        uint64_t carry = 0;  // payload bytes consumed by prior chunks
+        for (uint64_t base = 0; base < n; base += WAVE) {
+            uint64_t row = base + (uint64_t)lane;
+            uint64_t lp1 = 0;
+            if (row < n)
+                for (uint32_t b = 0; b < wbytes; b++)
+                    lp1 = (lp1 << 8) | lens[row * wbytes + b];
+            uint64_t vlen = lp1 ? lp1 - 1 : 0;
+            uint64_t incl = wave_incl_scan(vlen, lane);
+            uint64_t myoff = carry + incl - vlen;
+            bool match = false;
+            if (row < n && lp1 && vlen == pred_len) {
+                match = true;
+                for (uint64_t k = 0; k < vlen; k++)
+                    if (vals[myoff + k] != pred[k]) { match = false; break; }
+            }
+            uint64_t word = __ballot(match);
+            if (lane == 0) bm[w0 + (base >> 6)] = word;
+            carry += (uint64_t)__shfl((long long)incl, WAVE - 1);
+        }
+        if (lane == 0) out[bi].rle_bit_off = w0;
+    }
 }
 
 // Fold arithmetic-progression blocks (Const dd=0 / DeltaConst) under a
@@ -1221,6 +1320,7 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
     const uint8_t *__restrict__ payload, const bydb_block_desc *__restrict__ blocks,
     int64_t n_blocks, int64_t min_ts, int64_t max_ts, int flags,
     const PredBlock *__restrict__ preds_in, int n_preds,
+    const uint64_t *__restrict__ pred_bm,
     const SegEntry *__restrict__ segs_in,
     const GroupBlock *__restrict__ groups_in,
     const uint16_t *__restrict__ gmap_in, int n_gslots, int64_t gm0,
@@ -1327,7 +1427,7 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
                 }
                 if (!pb.active) { skip_block = true; continue; }  // nil tag
                 PredWalk *w = sl == 0 ? &pw0 : sl == 1 ? &pw1 : &pw2;
-                pred_init(w, payload, &pb);
+                pred_init(w, payload, &pb, pred_bm);
                 if (pb.uniform) {
                     // one run covers the block: predicate is block-uniform
                     pred_advance(w);
@@ -1761,6 +1861,13 @@ struct bydb_session {
     uint32_t dom_n[3] = {};
     uint8_t *d_pred_bytes = nullptr;
     uint64_t pred_bytes_cap = 0;
+    // plain (non-dictionary) tag columns: host-normalized sidecar arena +
+    // per-row match bitmaps filled by k_resolve_plain
+    uint8_t *d_sidecar = nullptr;
+    uint64_t sidecar_len = 0, sidecar_cap = 0;
+    uint32_t *d_plain_ctr = nullptr;
+    uint64_t *d_pred_bm = nullptr;
+    uint64_t pred_bm_cap = 0;  // words
     float last_ms = 0.0f;
     bool consumed = false;
     int16_t float_exp = 0;  // shared decimal exponent for float64 restore
@@ -1792,12 +1899,14 @@ extern "C" bydb_session *bydb_session_create(int device) {
     if ((e = hipStreamCreate(&s->stream)) != hipSuccess ||
         (e = hipEventCreate(&s->ev_start)) != hipSuccess ||
         (e = hipEventCreate(&s->ev_stop)) != hipSuccess ||
-        (e = hipMalloc(&s->d_err, sizeof(DevErr))) != hipSuccess) {
+        (e = hipMalloc(&s->d_err, sizeof(DevErr))) != hipSuccess ||
+        (e = hipMalloc(&s->d_plain_ctr, sizeof(uint32_t))) != hipSuccess) {
         fprintf(stderr, "bydb_session_create: init failed: %s\n", hipGetErrorString(e));
         delete s;
         return nullptr;
     }
     (void)hipMemset(s->d_err, 0, sizeof(DevErr));
+    (void)hipMemset(s->d_plain_ctr, 0, sizeof(uint32_t));
     return s;
 }
 
@@ -1819,6 +1928,9 @@ extern "C" void bydb_session_destroy(bydb_session *s) {
         if (s->d_dom_gids[sl]) (void)hipFree(s->d_dom_gids[sl]);
     }
     if (s->d_err) (void)hipFree(s->d_err);
+    if (s->d_sidecar) (void)hipFree(s->d_sidecar);
+    if (s->d_plain_ctr) (void)hipFree(s->d_plain_ctr);
+    if (s->d_pred_bm) (void)hipFree(s->d_pred_bm);
     if (s->ev_start) (void)hipEventDestroy(s->ev_start);
     if (s->ev_stop) (void)hipEventDestroy(s->ev_stop);
     if (s->stream) (void)hipStreamDestroy(s->stream);
@@ -1843,6 +1955,7 @@ extern "C" int bydb_part_reserve(bydb_session *s, uint64_t payload_bytes,
     s->blocks_cap = n_blocks;
     s->payload_len = 0;
     s->n_blocks = 0;
+    s->sidecar_len = 0;
     s->segs_built = false;
     s->groups_built = false;
     return BYDB_OK;
@@ -1859,9 +1972,61 @@ extern "C" int bydb_part_append(bydb_session *s, const uint8_t *payload,
         s->err = "part_append exceeds reservation";
         return BYDB_ERR_BAD_ARG;
     }
+    // Plain (non-dictionary) tag columns are zstd-compressed in the part;
+    // normalize them here — where the reference decompresses at part open
+    // (zstd.go:49) — into the sidecar arena, and repoint the stored
+    // descriptor (TAG_SIDECAR_BIT).  The payload arena layout is untouched
+    // so later chunks' absolute offsets stay valid.
+    std::vector<bydb_block_desc> fixed;
+    std::vector<uint8_t> extra;
+    for (int64_t i = 0; i < n_blocks; i++) {
+        const bydb_block_desc *bd = &blocks[i];
+        for (int sl = 0; sl < 3; sl++) {
+            uint64_t toff = sl == 0 ? bd->tag_off
+                                    : sl == 1 ? bd->tag2_off : bd->tag3_off;
+            uint64_t tlen = sl == 0 ? bd->tag_len
+                                    : sl == 1 ? bd->tag2_len : bd->tag3_len;
+            if (tlen == 0 || (toff & TAG_SIDECAR_BIT)) continue;
+            if (toff < s->payload_len || toff + tlen > s->payload_len + len)
+                continue;  // stream not in this chunk: leave as-is
+            const uint8_t *src = payload + (toff - s->payload_len);
+            if (src[0] != BYDB_ENC_PLAIN) continue;
+            std::vector<uint8_t> norm;
+            if (!bydb_normalize_plain_tag(src, tlen, bd->count, norm)) {
+                s->err = "plain tag column normalization failed";
+                return BYDB_ERR_BAD_DATA;
+            }
+            if (fixed.empty()) fixed.assign(blocks, blocks + n_blocks);
+            uint64_t noff = TAG_SIDECAR_BIT | (s->sidecar_len + extra.size());
+            bydb_block_desc &fd = fixed[(size_t)i];
+            if (sl == 0) { fd.tag_off = noff; fd.tag_len = norm.size(); }
+            else if (sl == 1) { fd.tag2_off = noff; fd.tag2_len = norm.size(); }
+            else { fd.tag3_off = noff; fd.tag3_len = norm.size(); }
+            extra.insert(extra.end(), norm.begin(), norm.end());
+        }
+    }
+    if (!extra.empty()) {
+        uint64_t need = s->sidecar_len + extra.size() + 64;
+        if (need > s->sidecar_cap) {
+            uint64_t ncap = s->sidecar_cap ? s->sidecar_cap : 4096;
+            while (ncap < need) ncap *= 2;
+            uint8_t *nb = nullptr;
+            HIP_TRY(s, hipMalloc(&nb, ncap));
+            if (s->sidecar_len)
+                HIP_TRY(s, hipMemcpy(nb, s->d_sidecar, s->sidecar_len,
+                                     hipMemcpyDeviceToDevice));
+            if (s->d_sidecar) (void)hipFree(s->d_sidecar);
+            s->d_sidecar = nb;
+            s->sidecar_cap = ncap;
+        }
+        HIP_TRY(s, hipMemcpy(s->d_sidecar + s->sidecar_len, extra.data(),
+                             extra.size(), hipMemcpyHostToDevice));
+        s->sidecar_len += extra.size();
+    }
     HIP_TRY(s, hipMemcpy(s->d_payload + s->payload_len, payload, len,
                          hipMemcpyHostToDevice));
-    HIP_TRY(s, hipMemcpy(s->d_blocks + s->n_blocks, blocks,
+    HIP_TRY(s, hipMemcpy(s->d_blocks + s->n_blocks,
+                         fixed.empty() ? blocks : fixed.data(),
                          sizeof(bydb_block_desc) * (size_t)n_blocks,
                          hipMemcpyHostToDevice));
     s->payload_len += len;
@@ -1872,6 +2037,7 @@ extern "C" int bydb_part_append(bydb_session *s, const uint8_t *payload,
 extern "C" int bydb_part_clear(bydb_session *s) {
     s->payload_len = 0;
     s->n_blocks = 0;
+    s->sidecar_len = 0;
     return BYDB_OK;
 }
 
@@ -2039,16 +2205,48 @@ extern "C" int bydb_consume_multi(bydb_session *s, int64_t min_ts,
         uint64_t off = 0;
         int rthreads = 256;
         int rblocks = (int)((s->n_blocks + rthreads - 1) / rthreads);
+        HIP_TRY(s, hipMemsetAsync(s->d_plain_ctr, 0, sizeof(uint32_t),
+                                  s->stream));
         for (int i = 0; i < n_preds; i++) {
             HIP_TRY(s, hipMemcpyAsync(s->d_pred_bytes + off, preds_in[i],
                                       pred_lens[i], hipMemcpyHostToDevice,
                                       s->stream));
             hipLaunchKernelGGL(k_resolve_pred, dim3(rblocks), dim3(rthreads), 0,
-                               s->stream, s->d_payload, s->d_blocks, s->n_blocks,
+                               s->stream, s->d_payload, s->d_sidecar,
+                               s->d_blocks, s->n_blocks,
                                s->d_pred_bytes + off, pred_lens[i], i,
-                               s->d_preds + (int64_t)i * s->n_blocks);
+                               s->d_preds + (int64_t)i * s->n_blocks,
+                               s->d_plain_ctr);
             HIP_TRY(s, hipGetLastError());
             off += pred_lens[i];
+        }
+        // plain (non-dictionary) columns found: size the match-bitmap arena
+        // and run the per-row equality pass
+        uint32_t n_plain = 0;
+        HIP_TRY(s, hipMemcpyAsync(&n_plain, s->d_plain_ctr, sizeof(uint32_t),
+                                  hipMemcpyDeviceToHost, s->stream));
+        HIP_TRY(s, hipStreamSynchronize(s->stream));
+        if (n_plain > 0) {
+            uint64_t words = (uint64_t)n_plain * PLAIN_BM_WORDS;
+            if (words > s->pred_bm_cap) {
+                if (s->d_pred_bm) (void)hipFree(s->d_pred_bm);
+                HIP_TRY(s, hipMalloc(&s->d_pred_bm, words * sizeof(uint64_t)));
+                s->pred_bm_cap = words;
+            }
+            HIP_TRY(s, hipMemsetAsync(s->d_plain_ctr, 0, sizeof(uint32_t),
+                                      s->stream));
+            int pgrid = (int)(s->n_blocks < 65535 ? s->n_blocks : 65535);
+            off = 0;
+            for (int i = 0; i < n_preds; i++) {
+                hipLaunchKernelGGL(k_resolve_plain, dim3(pgrid), dim3(WAVE), 0,
+                                   s->stream, s->d_payload, s->d_sidecar,
+                                   s->d_blocks, s->n_blocks,
+                                   s->d_pred_bytes + off, pred_lens[i], i,
+                                   s->d_preds + (int64_t)i * s->n_blocks,
+                                   s->d_pred_bm, s->d_plain_ctr);
+                HIP_TRY(s, hipGetLastError());
+                off += pred_lens[i];
+            }
         }
         preds = s->d_preds;
     }
@@ -2121,9 +2319,9 @@ extern "C" int bydb_consume_multi(bydb_session *s, int64_t min_ts,
     const bool en_preds = n_preds > 0;
     const bool en_groups = groups != nullptr;
     void (*kfn)(const uint8_t *, const bydb_block_desc *, int64_t, int64_t,
-                int64_t, int, const PredBlock *, int, const SegEntry *,
-                const GroupBlock *, const uint16_t *, int, int64_t, int64_t,
-                int64_t, int64_t, bydb_partial *, DevErr *);
+                int64_t, int, const PredBlock *, int, const uint64_t *,
+                const SegEntry *, const GroupBlock *, const uint16_t *, int,
+                int64_t, int64_t, int64_t, int64_t, bydb_partial *, DevErr *);
     if (en_values) {
         if (en_preds) kfn = en_groups ? k_scan_agg_t<true, true, true>
                                       : k_scan_agg_t<true, true, false>;
@@ -2137,9 +2335,9 @@ extern "C" int bydb_consume_multi(bydb_session *s, int64_t min_ts,
     }
     hipLaunchKernelGGL(kfn, dim3(grid), dim3(threads), 0, s->stream,
                        s->d_payload, s->d_blocks, s->n_blocks, min_ts, max_ts,
-                       flags, preds, n_preds, segs, groups, s->d_gmap,
-                       s->n_gslots, s->gmul[0], s->gmul[1], s->gmul[2],
-                       (int64_t)s->n_groups, s->d_acc, s->d_err);
+                       flags, preds, n_preds, s->d_pred_bm, segs, groups,
+                       s->d_gmap, s->n_gslots, s->gmul[0], s->gmul[1],
+                       s->gmul[2], (int64_t)s->n_groups, s->d_acc, s->d_err);
     HIP_TRY(s, hipGetLastError());
     HIP_TRY(s, hipEventRecord(s->ev_stop, s->stream));
     s->consumed = true;

@@ -1108,12 +1108,23 @@ static int scan_block(const uint8_t *payload, const bo_block_desc *b,
         uint64_t tlen = sl == 0 ? b->tag_len : sl == 1 ? b->tag2_len : b->tag3_len;
         if (tlen == 0) return BO_OK;  /* nil tags: exclude whole block */
         uint8_t tag_type = payload[toff];
-        if (tag_type != BO_ENC_DICTIONARY) return BO_ERR_BAD_TYPE;
         size_t tagdata_len = 0;
-        rc = bo_dictionary_decode(tagdata_buf, (size_t)1 << 24, taglen_buf,
-                                  payload + toff + 1, tlen - 1, n,
-                                  &tagdata_len);
-        if (rc != BO_OK) return rc;
+        if (tag_type == BO_ENC_DICTIONARY) {
+            rc = bo_dictionary_decode(tagdata_buf, (size_t)1 << 24, taglen_buf,
+                                      payload + toff + 1, tlen - 1, n,
+                                      &tagdata_len);
+            if (rc != BO_OK) return rc;
+        } else if (tag_type == BO_ENC_PLAIN) {
+            /* residual pushdown on a plain bytes column — the >256-distinct
+             * dictionary bail (column.go:266-278, dictionary.go:58):
+             * decode the bytes block and compare per row */
+            rc = bo_bytes_block_decode(tagdata_buf, (size_t)1 << 24,
+                                       taglen_buf, payload + toff + 1,
+                                       tlen - 1, n, &tagdata_len);
+            if (rc != BO_OK) return rc;
+        } else {
+            return BO_ERR_BAD_TYPE;
+        }
         size_t tago = 0;
         for (int64_t i = 0; i < n; i++) {
             int64_t tl = taglen_buf[i];
