@@ -463,6 +463,75 @@ bool bydb_normalize_plain_tag(const uint8_t *src, uint64_t src_len,
     return true;
 }
 
+// Same normalization for dictionary tag columns whose compress_block
+// sections are zstd'd (>=128 B, bytes.go:291-303): decompress the lengths
+// and values sections and rewrite them with the raw framing
+// [2][u32le len][bytes] the device parsers accept; the bit-packed RLE
+// section is raw bits and is copied verbatim.  Returns false when the
+// stream needs no normalization (both sections already plain).
+bool bydb_normalize_dict_tag(const uint8_t *src, uint64_t src_len,
+                             bool *needed, std::vector<uint8_t> &out) {
+    *needed = false;
+    if (src_len < 2 || src[0] != BYDB_ENC_DICTIONARY) return false;
+    const uint8_t *p = src + 1;
+    const uint8_t *end = src + src_len;
+    // varuint count (int.go:152-199)
+    uint64_t count = 0;
+    unsigned sh = 0;
+    while (p < end) {
+        uint8_t c = *p++;
+        count |= (uint64_t)(c & 0x7f) << sh;
+        if (c < 0x80) break;
+        sh += 7;
+    }
+    if (p >= end) return false;
+    if (p[0] == 0) {
+        // lengths plain: values section decides
+        const uint8_t *v = p + 2 + p[1];
+        if (v >= end || v[0] == 0) return true;  // fully plain: no work
+    }
+    *needed = true;
+    std::vector<uint8_t> lens, vals;
+    uint64_t used = 0;
+    if (!decompress_block_host(lens, p, (uint64_t)(end - p),
+                               1 + count * 8, &used))
+        return false;
+    p += used;
+    // values payload upper bound: decompressed lens give the exact total
+    if (lens.empty()) return false;
+    uint8_t wt = lens[0];
+    uint32_t wbytes = wt == 0 ? 1 : wt == 1 ? 2 : wt == 2 ? 4 : 8;
+    if (lens.size() != 1 + (size_t)count * wbytes) return false;
+    uint64_t total = 0;
+    for (uint64_t i = 0; i < count; i++) {
+        uint64_t ap1 = 0;
+        for (uint32_t b = 0; b < wbytes; b++)
+            ap1 = (ap1 << 8) | lens[1 + i * wbytes + b];
+        if (ap1) total += ap1 - 1;
+    }
+    if (!decompress_block_host(vals, p, (uint64_t)(end - p), total, &used))
+        return false;
+    p += used;
+    if (vals.size() != total) return false;
+    out.clear();
+    out.push_back(BYDB_ENC_DICTIONARY);
+    {
+        uint64_t u = count;
+        while (u >= 0x80) { out.push_back((uint8_t)(u | 0x80)); u >>= 7; }
+        out.push_back((uint8_t)u);
+    }
+    auto raw_section = [&out](const std::vector<uint8_t> &sec) {
+        out.push_back(2);
+        uint32_t l = (uint32_t)sec.size();
+        for (int b = 0; b < 4; b++) out.push_back((uint8_t)(l >> (8 * b)));
+        out.insert(out.end(), sec.begin(), sec.end());
+    };
+    raw_section(lens);
+    raw_section(vals);
+    out.insert(out.end(), p, end);  // bit-packed RLE verbatim
+    return true;
+}
+
 // ===================== part builder =====================
 struct bydb_part_builder {
     std::vector<uint8_t> payload;

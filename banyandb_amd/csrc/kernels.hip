@@ -48,6 +48,8 @@
 #define PLAIN_BM_WORDS 128  // bitmap words per block (8192 rows max)
 bool bydb_normalize_plain_tag(const uint8_t *src, uint64_t src_len,
                               uint64_t nrows, std::vector<uint8_t> &out);
+bool bydb_normalize_dict_tag(const uint8_t *src, uint64_t src_len,
+                             bool *needed, std::vector<uint8_t> &out);
 
 // ---------------- device helpers ----------------
 
@@ -525,6 +527,29 @@ __device__ __forceinline__ uint64_t rd_bits_be(const uint8_t *p,
     return (acc >> (64 - sh - n)) & ((1ull << n) - 1);
 }
 
+// compress_block section header inside a dictionary stream: [0][u8 len]
+// plain (<128 B, bytes.go:291-295) or [2][u32le len] raw — the framing the
+// host normalization writes after decompressing a zstd'd section.  Returns
+// the section start, or nullptr on an unparseable marker (zstd reaching
+// the device unnormalized).
+__device__ __forceinline__ const uint8_t *dict_section(const uint8_t *p,
+                                                       const uint8_t *end,
+                                                       uint64_t *len_out) {
+    if (p >= end) return nullptr;
+    if (*p == 0) {
+        if (p + 2 > end) return nullptr;
+        *len_out = p[1];
+        return p + 2;
+    }
+    if (*p == 2) {
+        if (p + 5 > end) return nullptr;
+        *len_out = (uint64_t)p[1] | ((uint64_t)p[2] << 8) |
+                   ((uint64_t)p[3] << 16) | ((uint64_t)p[4] << 24);
+        return p + 5;
+    }
+    return nullptr;
+}
+
 struct PredWalk {
     const uint8_t *payload;
     const uint64_t *bm;        // plain columns: per-row match bitmap (O(1))
@@ -537,11 +562,14 @@ struct PredWalk {
 };
 
 __device__ __forceinline__ void pred_init(PredWalk *pw, const uint8_t *payload,
+                                          const uint8_t *sidecar,
                                           const PredBlock *pb,
                                           const uint64_t *bm_arena) {
-    pw->payload = payload;
+    // bit 63 of rle_bit_off: the dictionary stream was host-normalized
+    // into the sidecar arena (zstd'd sections); offsets are sidecar-based
+    pw->payload = (pb->rle_bit_off & TAG_SIDECAR_BIT) ? sidecar : payload;
     pw->bm = pb->plain ? bm_arena + pb->rle_bit_off : nullptr;
-    pw->bit0 = pb->rle_bit_off;
+    pw->bit0 = pb->rle_bit_off & ~TAG_SIDECAR_BIT;
     pw->nentries = pb->nentries;
     pw->width = pb->width;
 #pragma unroll
@@ -655,6 +683,7 @@ __device__ inline uint32_t domain_lookup(const GroupDomain *d, const uint8_t *v,
 
 // one thread per block: parse the tag dictionary, map codes to gids
 __global__ void k_resolve_groups(const uint8_t *__restrict__ payload,
+                                 const uint8_t *__restrict__ sidecar,
                                  const bydb_block_desc *__restrict__ blocks,
                                  int64_t n_blocks, int slot, GroupDomain dom,
                                  GroupBlock *__restrict__ out,
@@ -671,13 +700,11 @@ __global__ void k_resolve_groups(const uint8_t *__restrict__ payload,
     gb.rle_bit_off = 0; gb.nentries = 0; gb.width = 0; gb.err = 0;
     gb._p[0] = gb._p[1] = 0;
     if (tlen == 0) { out[bi] = gb; return; }
-    if (toff & TAG_SIDECAR_BIT) {  // plain column: group-by unsupported (v1)
-        gb.err = 1;
-        out[bi] = gb;
-        return;
-    }
-    const uint8_t *p = payload + toff;
+    const bool in_sidecar = (toff & TAG_SIDECAR_BIT) != 0;
+    const uint8_t *base = in_sidecar ? sidecar : payload;
+    const uint8_t *p = base + (toff & ~TAG_SIDECAR_BIT);
     const uint8_t *end = p + tlen;
+    // group-by on a plain (>256-distinct) column is unsupported in v1
     if (*p != BYDB_ENC_DICTIONARY) { gb.err = 1; out[bi] = gb; return; }
     p++;
     uint64_t count = 0;
@@ -688,16 +715,14 @@ __global__ void k_resolve_groups(const uint8_t *__restrict__ payload,
         if (c < 0x80) break;
         sh += 7;
     }
-    if (p >= end || *p != 0) { gb.err = 1; out[bi] = gb; return; }
-    p++;
-    uint64_t ll = *p++;
-    const uint8_t *lens_blk = p;
-    p += ll;
-    if (p >= end || *p != 0) { gb.err = 1; out[bi] = gb; return; }
-    p++;
-    uint64_t vl = *p++;
-    const uint8_t *vals = p;
-    p += vl;
+    uint64_t ll = 0;
+    const uint8_t *lens_blk = dict_section(p, end, &ll);
+    if (!lens_blk) { gb.err = 1; out[bi] = gb; return; }
+    p = lens_blk + ll;
+    uint64_t vl = 0;
+    const uint8_t *vals = dict_section(p, end, &vl);
+    if (!vals) { gb.err = 1; out[bi] = gb; return; }
+    p = vals + vl;
     uint8_t wt = lens_blk[0];
     uint32_t wbytes = wt == 0 ? 1 : wt == 1 ? 2 : wt == 2 ? 4 : 8;
     const uint8_t *lp = lens_blk + 1;
@@ -714,15 +739,15 @@ __global__ void k_resolve_groups(const uint8_t *__restrict__ payload,
         }
         map[v] = g == GID_NONE ? 0xFFFFu : (uint16_t)g;
     }
-    uint64_t bit0 = ((uint64_t)(p - payload)) * 8;
-    uint32_t nentries = (uint32_t)rd_bits_be(payload, bit0, 32);
-    uint32_t width = nentries ? (uint32_t)rd_bits_be(payload, bit0 + 32, 8) : 0;
+    uint64_t bit0 = ((uint64_t)(p - base)) * 8;
+    uint32_t nentries = (uint32_t)rd_bits_be(base, bit0, 32);
+    uint32_t width = nentries ? (uint32_t)rd_bits_be(base, bit0 + 32, 8) : 0;
     gb.nentries = nentries;
     gb.width = (uint8_t)width;
-    gb.rle_bit_off = bit0 + 40;
+    gb.rle_bit_off = (bit0 + 40) | (in_sidecar ? TAG_SIDECAR_BIT : 0);
     if (nentries == 2) {
-        uint64_t code = rd_bits_be(payload, gb.rle_bit_off, width);
-        uint64_t cnt = rd_bits_be(payload, gb.rle_bit_off + width, width);
+        uint64_t code = rd_bits_be(base, bit0 + 40, width);
+        uint64_t cnt = rd_bits_be(base, bit0 + 40 + width, width);
         if (cnt >= bd->count) {
             uint16_t m = map[code < 256 ? code : 0];
             gb.uniform_gid = m == 0xFFFFu ? GID_NONE : m;
@@ -755,9 +780,9 @@ __global__ void k_resolve_pred(const uint8_t *__restrict__ payload,
     pb.rle_bit_off = 0; pb.nentries = 0; pb.width = 0;
     pb.active = 0; pb.err = 0; pb.uniform = 0; pb.plain = 0;
     if (tlen == 0) { out[bi] = pb; return; }
-    const uint8_t *p = (toff & TAG_SIDECAR_BIT)
-                           ? sidecar + (toff & ~TAG_SIDECAR_BIT)
-                           : payload + toff;
+    const bool in_sidecar = (toff & TAG_SIDECAR_BIT) != 0;
+    const uint8_t *base = in_sidecar ? sidecar : payload;
+    const uint8_t *p = base + (toff & ~TAG_SIDECAR_BIT);
     const uint8_t *end = p + tlen;
     if (*p == BYDB_ENC_PLAIN) {
         // host-normalized plain column: a second pass (k_resolve_plain)
@@ -780,18 +805,16 @@ __global__ void k_resolve_pred(const uint8_t *__restrict__ payload,
         if (c < 0x80) break;
         sh += 7;
     }
-    // lengths block: compress_block(u64list) — plain only
-    if (p >= end || *p != 0) { pb.err = 1; out[bi] = pb; return; }
-    p++;
-    uint64_t ll = *p++;
-    const uint8_t *lens_blk = p;
-    p += ll;
-    // values payload block: plain only
-    if (p >= end || *p != 0) { pb.err = 1; out[bi] = pb; return; }
-    p++;
-    uint64_t vl = *p++;
-    const uint8_t *vals = p;
-    p += vl;
+    // lengths block: compress_block(u64list) — plain or host-normalized
+    uint64_t ll = 0;
+    const uint8_t *lens_blk = dict_section(p, end, &ll);
+    if (!lens_blk) { pb.err = 1; out[bi] = pb; return; }
+    p = lens_blk + ll;
+    // values payload block
+    uint64_t vl = 0;
+    const uint8_t *vals = dict_section(p, end, &vl);
+    if (!vals) { pb.err = 1; out[bi] = pb; return; }
+    p = vals + vl;
     // parse width-typed lengths (bytes.go:205-235)
     uint8_t wt = lens_blk[0];
     uint32_t wbytes = wt == 0 ? 1 : wt == 1 ? 2 : wt == 2 ? 4 : 8;
@@ -812,14 +835,14 @@ __global__ void k_resolve_pred(const uint8_t *__restrict__ payload,
         }
     }
     // bit-packed RLE: [32b entry count][8b width][entries...] MSB-first
-    uint64_t bit0 = ((uint64_t)(p - payload)) * 8;
-    uint32_t nentries = (uint32_t)rd_bits_be(payload, bit0, 32);
-    uint32_t width = nentries ? (uint32_t)rd_bits_be(payload, bit0 + 32, 8) : 0;
+    uint64_t bit0 = ((uint64_t)(p - base)) * 8;
+    uint32_t nentries = (uint32_t)rd_bits_be(base, bit0, 32);
+    uint32_t width = nentries ? (uint32_t)rd_bits_be(base, bit0 + 32, 8) : 0;
     pb.nentries = nentries;
     pb.width = (uint8_t)width;
-    pb.rle_bit_off = bit0 + 40;
+    pb.rle_bit_off = (bit0 + 40) | (in_sidecar ? TAG_SIDECAR_BIT : 0);
     if (nentries == 2) {
-        uint64_t cnt = rd_bits_be(payload, pb.rle_bit_off + width, width);
+        uint64_t cnt = rd_bits_be(base, bit0 + 40 + width, width);
         if (cnt >= bd->count) pb.uniform = 1;
     }
     out[bi] = pb;
@@ -1317,7 +1340,8 @@ __device__ void fold_range(const uint8_t *fstream, uint8_t fenc, int64_t first,
 // guards — dead branches are eliminated per instantiation.)
 template <bool EN_VALUES, bool EN_PREDS, bool EN_GROUPS>
 __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) void k_scan_agg_t(
-    const uint8_t *__restrict__ payload, const bydb_block_desc *__restrict__ blocks,
+    const uint8_t *__restrict__ payload, const uint8_t *__restrict__ sidecar,
+    const bydb_block_desc *__restrict__ blocks,
     int64_t n_blocks, int64_t min_ts, int64_t max_ts, int flags,
     const PredBlock *__restrict__ preds_in, int n_preds,
     const uint64_t *__restrict__ pred_bm,
@@ -1427,7 +1451,7 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
                 }
                 if (!pb.active) { skip_block = true; continue; }  // nil tag
                 PredWalk *w = sl == 0 ? &pw0 : sl == 1 ? &pw1 : &pw2;
-                pred_init(w, payload, &pb, pred_bm);
+                pred_init(w, payload, sidecar, &pb, pred_bm);
                 if (pb.uniform) {
                     // one run covers the block: predicate is block-uniform
                     pred_advance(w);
@@ -1476,18 +1500,21 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
                 if (seg != 0) continue;  // whole block folded at seg 0
                 const uint8_t *gstream = payload + bd->field_off;
                 // per-slot run cursors (uniform slots are one infinite run)
+                const uint8_t *gsrc[3];
                 uint64_t bit[3];
                 uint32_t ent[3];
                 int64_t run_hi_s[3];
                 int64_t gid_s[3];
                 for (int sl = 0; sl < n_gslots; sl++) {
+                    gsrc[sl] = (gb[sl].rle_bit_off & TAG_SIDECAR_BIT)
+                                   ? sidecar : payload;
                     if (gb[sl].uniform_gid != GID_VARYING) {
                         gid_s[sl] = (int64_t)gb[sl].uniform_gid;
                         run_hi_s[sl] = n;
                         ent[sl] = gb[sl].nentries;  // exhausted
                         bit[sl] = 0;
                     } else {
-                        bit[sl] = gb[sl].rle_bit_off;
+                        bit[sl] = gb[sl].rle_bit_off & ~TAG_SIDECAR_BIT;
                         ent[sl] = 0;
                         run_hi_s[sl] = 0;
                         gid_s[sl] = -1;
@@ -1499,9 +1526,9 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
                         gid_s[sl] = -1;  // RLE exhausted: no group
                         return;
                     }
-                    uint64_t code = rd_bits_be(payload, bit[sl], gb[sl].width);
+                    uint64_t code = rd_bits_be(gsrc[sl], bit[sl], gb[sl].width);
                     bit[sl] += gb[sl].width;
-                    uint64_t cnt = rd_bits_be(payload, bit[sl], gb[sl].width);
+                    uint64_t cnt = rd_bits_be(gsrc[sl], bit[sl], gb[sl].width);
                     bit[sl] += gb[sl].width;
                     ent[sl] += 2;
                     uint16_t m = (gmap_in + gb[sl].map_off)[code < 256 ? code : 0];
@@ -1979,8 +2006,20 @@ extern "C" int bydb_part_append(bydb_session *s, const uint8_t *payload,
     // so later chunks' absolute offsets stay valid.
     std::vector<bydb_block_desc> fixed;
     std::vector<uint8_t> extra;
+    const uint64_t part_end = s->payload_len + len;
     for (int64_t i = 0; i < n_blocks; i++) {
         const bydb_block_desc *bd = &blocks[i];
+        // loud host-side validation: every stream the descriptor names
+        // must lie inside the part appended so far (catches builders whose
+        // offset base drifted from this session's arena)
+        if (bd->ts_off + bd->ts_len > part_end ||
+            bd->field_off + bd->field_len > part_end ||
+            (bd->tag_len && bd->tag_off + bd->tag_len > part_end) ||
+            (bd->tag2_len && bd->tag2_off + bd->tag2_len > part_end) ||
+            (bd->tag3_len && bd->tag3_off + bd->tag3_len > part_end)) {
+            s->err = "descriptor stream offsets exceed appended payload";
+            return BYDB_ERR_BAD_ARG;
+        }
         for (int sl = 0; sl < 3; sl++) {
             uint64_t toff = sl == 0 ? bd->tag_off
                                     : sl == 1 ? bd->tag2_off : bd->tag3_off;
@@ -1990,11 +2029,27 @@ extern "C" int bydb_part_append(bydb_session *s, const uint8_t *payload,
             if (toff < s->payload_len || toff + tlen > s->payload_len + len)
                 continue;  // stream not in this chunk: leave as-is
             const uint8_t *src = payload + (toff - s->payload_len);
-            if (src[0] != BYDB_ENC_PLAIN) continue;
             std::vector<uint8_t> norm;
-            if (!bydb_normalize_plain_tag(src, tlen, bd->count, norm)) {
-                s->err = "plain tag column normalization failed";
-                return BYDB_ERR_BAD_DATA;
+            if (src[0] == BYDB_ENC_PLAIN) {
+                if (!bydb_normalize_plain_tag(src, tlen, bd->count, norm)) {
+                    s->err = "plain tag column normalization failed";
+                    return BYDB_ERR_BAD_DATA;
+                }
+            } else if (src[0] == BYDB_ENC_DICTIONARY) {
+                // dictionaries whose compress_block sections are zstd'd
+                // (>=128 B) get the same host decompress; small plain
+                // dictionaries parse on device untouched
+                bool needed = false;
+                if (!bydb_normalize_dict_tag(src, tlen, &needed, norm)) {
+                    if (needed) {
+                        s->err = "dictionary tag normalization failed";
+                        return BYDB_ERR_BAD_DATA;
+                    }
+                    continue;
+                }
+                if (!needed) continue;
+            } else {
+                continue;
             }
             if (fixed.empty()) fixed.assign(blocks, blocks + n_blocks);
             uint64_t noff = TAG_SIDECAR_BIT | (s->sidecar_len + extra.size());
@@ -2301,8 +2356,9 @@ extern "C" int bydb_consume_multi(bydb_session *s, int64_t min_ts,
                 dom.n = s->dom_n[i];
                 hipLaunchKernelGGL(
                     k_resolve_groups, dim3(rblocks), dim3(rthreads), 0,
-                    s->stream, s->d_payload, s->d_blocks, s->n_blocks,
-                    s->gslots[i], dom, s->d_groups + (int64_t)i * s->n_blocks,
+                    s->stream, s->d_payload, s->d_sidecar, s->d_blocks,
+                    s->n_blocks, s->gslots[i], dom,
+                    s->d_groups + (int64_t)i * s->n_blocks,
                     s->d_gmap, (uint32_t)((int64_t)i * 256 * s->n_blocks));
                 HIP_TRY(s, hipGetLastError());
             }
@@ -2318,10 +2374,11 @@ extern "C" int bydb_consume_multi(bydb_session *s, int64_t min_ts,
     const bool en_values = (flags & KF_NEED_VALUES) != 0;
     const bool en_preds = n_preds > 0;
     const bool en_groups = groups != nullptr;
-    void (*kfn)(const uint8_t *, const bydb_block_desc *, int64_t, int64_t,
-                int64_t, int, const PredBlock *, int, const uint64_t *,
-                const SegEntry *, const GroupBlock *, const uint16_t *, int,
-                int64_t, int64_t, int64_t, int64_t, bydb_partial *, DevErr *);
+    void (*kfn)(const uint8_t *, const uint8_t *, const bydb_block_desc *,
+                int64_t, int64_t, int64_t, int, const PredBlock *, int,
+                const uint64_t *, const SegEntry *, const GroupBlock *,
+                const uint16_t *, int, int64_t, int64_t, int64_t, int64_t,
+                bydb_partial *, DevErr *);
     if (en_values) {
         if (en_preds) kfn = en_groups ? k_scan_agg_t<true, true, true>
                                       : k_scan_agg_t<true, true, false>;
@@ -2334,10 +2391,11 @@ extern "C" int bydb_consume_multi(bydb_session *s, int64_t min_ts,
                              : k_scan_agg_t<false, false, false>;
     }
     hipLaunchKernelGGL(kfn, dim3(grid), dim3(threads), 0, s->stream,
-                       s->d_payload, s->d_blocks, s->n_blocks, min_ts, max_ts,
-                       flags, preds, n_preds, s->d_pred_bm, segs, groups,
-                       s->d_gmap, s->n_gslots, s->gmul[0], s->gmul[1],
-                       s->gmul[2], (int64_t)s->n_groups, s->d_acc, s->d_err);
+                       s->d_payload, s->d_sidecar, s->d_blocks, s->n_blocks,
+                       min_ts, max_ts, flags, preds, n_preds, s->d_pred_bm,
+                       segs, groups, s->d_gmap, s->n_gslots, s->gmul[0],
+                       s->gmul[1], s->gmul[2], (int64_t)s->n_groups, s->d_acc,
+                       s->d_err);
     HIP_TRY(s, hipGetLastError());
     HIP_TRY(s, hipEventRecord(s->ev_stop, s->stream));
     s->consumed = true;

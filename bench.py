@@ -52,11 +52,13 @@ def main():
                     default=os.environ.get("BYDB_SKIP_CPU_BASELINE") == "1")
     ap.add_argument("--workload", default="i64_sum",
                     choices=["i64_sum", "f64_pred", "group256",
-                             "group4096_pred3"],
+                             "group4096_pred3", "mixed"],
                     help="i64_sum = BASELINE configs[1] (default, the "
                          "headline metric); f64_pred = configs[2] shape; "
                          "group256 = configs[3] shape; group4096_pred3 = "
-                         "configs[4] shape (int64 leg)")
+                         "configs[4] shape (int64 leg); mixed = "
+                         "configs[5] shape (half int64 + half float64 "
+                         "series, two overlapped sessions)")
     args = ap.parse_args()
 
     import torch
@@ -106,10 +108,23 @@ def main():
                                 dtype="int64",
                                 agg="sum+count groupby(4096) 3-tag filter",
                                 enc_bpd=1.05),
+        # configs[5]: mixed int64/float64 fields — half the series carry an
+        # int64 field, half a float64 field; two sessions (one per value
+        # type, separate HIP streams, overlapped) fold into separate
+        # per-group partials, mirroring one column each of the reference's
+        # multi-field MeasureBatch
+        "mixed": dict(float="mixed", n_groups=4096, group_mod=4096,
+                      tags=[ENVS, REGIONS, SVCS],
+                      preds=[b"prod", b"r4", b"s4"],
+                      funcs=["sum", "count"], algo_bpd=19,
+                      dtype="int64+f64",
+                      agg="sum+count groupby(4096) 3-tag filter mixed",
+                      enc_bpd=1.6),
     }[args.workload]
     base_name = {"i64_sum": "int64_sum_count", "f64_pred": "f64_minmaxavg_pred",
                  "group256": "int64_group256",
-                 "group4096_pred3": "int64_group4096_pred3"}[args.workload]
+                 "group4096_pred3": "int64_group4096_pred3",
+                 "mixed": "mixed_group4096_pred3"}[args.workload]
     workload = (f"10k_series_x_1M_{base_name}" if full_config
                 else f"{n_series}_series_x_{n_dp}_{base_name}")
     if args.workload == "i64_sum" and full_config:
@@ -125,8 +140,14 @@ def main():
         + n_series * ((n_dp + 8191) // 8192) * tag_overhead
     est_blocks = n_series * ((n_dp + 8191) // 8192)
 
+    mixed = W["float"] == "mixed"
     sess = ba.Session(device)
-    sess.reserve(est_payload, est_blocks)
+    sess.reserve(est_payload if not mixed else est_payload // 2 + 1024,
+                 est_blocks if not mixed else est_blocks // 2 + 1)
+    sess2 = None
+    if mixed:
+        sess2 = ba.Session(device)
+        sess2.reserve(est_payload, est_blocks // 2 + 1)
 
     gen_t = time.perf_counter()
     chunk = 500
@@ -134,40 +155,68 @@ def main():
     for slot, table in enumerate(W["tags"]):
         b.set_tag_table(slot, table)
     uploaded_dp = 0
-    for s0 in range(0, n_series, chunk):
-        ns = min(chunk, n_series - s0)
-        if W["float"]:
-            b.gen_bulk_f64(s0, ns, n_dp, t0_r, STRIDE, 10.0, 0.01, seed_r,
-                           group_mod=W["group_mod"], threads=threads)
+    half = n_series // 2
+    # mixed: a separate builder per session — a builder's running base
+    # offset must track its own session's payload arena
+    b2 = None
+    if mixed:
+        b2 = ba.PartBuilder()
+        for slot, table in enumerate(W["tags"]):
+            b2.set_tag_table(slot, table)
+    s0 = 0
+    while s0 < n_series:
+        lim = half if (mixed and s0 < half) else n_series
+        ns = min(chunk, lim - s0)
+        use_float = W["float"] is True or (mixed and s0 >= half)
+        bb = b2 if (mixed and s0 >= half) else b
+        if use_float:
+            bb.gen_bulk_f64(s0, ns, n_dp, t0_r, STRIDE, 10.0, 0.01, seed_r,
+                            group_mod=W["group_mod"], threads=threads)
         else:
-            b.gen_bulk_i64(s0, ns, n_dp, t0_r, STRIDE, 1000, 1, seed_r,
-                           group_mod=W["group_mod"], threads=threads)
-        sess.append(b)
+            bb.gen_bulk_i64(s0, ns, n_dp, t0_r, STRIDE, 1000, 1, seed_r,
+                            group_mod=W["group_mod"], threads=threads)
+        (sess2 if (mixed and s0 >= half) else sess).append(bb)
         uploaded_dp += ns * n_dp
-        b.drain()
+        bb.drain()
+        s0 += ns
     gen_s = time.perf_counter() - gen_t
     log(f"rank {rank}: generated+uploaded {uploaded_dp} dp in {gen_s:.1f}s")
 
     n_groups = W["n_groups"]
     fmap = {"sum": ba.AGG_SUM, "count": ba.AGG_COUNT, "min": ba.AGG_MIN,
             "max": ba.AGG_MAX}
-    sess.configure(ba.VT_FLOAT64 if W["float"] else ba.VT_INT64,
+    sess.configure(ba.VT_FLOAT64 if W["float"] is True else ba.VT_INT64,
                    [fmap[f] for f in W["funcs"]], n_groups=n_groups,
-                   float_exp=-2 if W["float"] else 0)
+                   float_exp=-2 if W["float"] is True else 0)
+    if mixed:
+        sess2.configure(ba.VT_FLOAT64, [fmap[f] for f in W["funcs"]],
+                        n_groups=n_groups, float_exp=-2)
 
     # partials live in a torch CUDA tensor so the merge is RCCL over xGMI
-    part_t = None
+    part_t = part_t2 = None
     if world > 1 and backend == "nccl":
         part_t = torch.zeros(n_groups * 6, dtype=torch.int64,
                              device=f"cuda:{device}")
         sess.set_partials_buffer(part_t.data_ptr(), part_t.numel() * 8)
+        if mixed:
+            part_t2 = torch.zeros(n_groups * 6, dtype=torch.int64,
+                                  device=f"cuda:{device}")
+            sess2.set_partials_buffer(part_t2.data_ptr(),
+                                      part_t2.numel() * 8)
 
     preds = W["preds"] or None
     need_minmax = "min" in W["funcs"] or "max" in W["funcs"]
 
     def one_step():
         sess.reset()
+        if mixed:
+            sess2.reset()
+        # consume is asynchronous: with two sessions the int64 and float64
+        # scans overlap on their own HIP streams
         sess.consume(min_ts=t0_r, max_ts=t0_r + n_dp * STRIDE, preds=preds)
+        if mixed:
+            sess2.consume(min_ts=t0_r, max_ts=t0_r + n_dp * STRIDE,
+                          preds=preds)
         if world > 1:
             # sync the session stream, then merge the partials
             # (AggModeReduce Combine semantics): RCCL over xGMI on the
@@ -184,7 +233,21 @@ def main():
                 allreduce_partials(dist, t, n_groups,
                                    need_minmax=need_minmax,
                                    need_float=W["float"])
+            if mixed:
+                parts2 = sess2.finalize_partials()
+                if backend == "nccl":
+                    allreduce_partials(dist, part_t2, n_groups,
+                                       need_minmax=need_minmax,
+                                       need_float=True)
+                else:
+                    t2 = partials_from_structs(parts2)
+                    allreduce_partials(dist, t2, n_groups,
+                                       need_minmax=need_minmax,
+                                       need_float=True)
+                return parts, parts2
             return parts
+        if mixed:
+            return sess.finalize_partials(), sess2.finalize_partials()
         return sess.finalize_partials()
 
     # ---- warmup ----
@@ -198,7 +261,12 @@ def main():
     t_start = time.perf_counter()
     for _ in range(args.steps):
         one_step()
-        launch_ms.append(sess.last_consume_ms())
+        lm = sess.last_consume_ms()
+        if mixed:
+            # overlapped streams: the conservative roofline denominator is
+            # the SUM of the two launch durations
+            lm += sess2.last_consume_ms()
+        launch_ms.append(lm)
     torch.cuda.synchronize(device) if torch.cuda.is_available() else None
     if world > 1:
         dist.barrier()
@@ -213,6 +281,8 @@ def main():
     if world == 1:
         parts = sess.finalize_partials()
         total_count = sum(p.count for p in parts)
+        if mixed:
+            total_count += sum(p.count for p in sess2.finalize_partials())
         if not W["preds"]:
             assert total_count == total_dp_rank, \
                 f"count {total_count} != {total_dp_rank}"
@@ -285,6 +355,8 @@ def main():
     if world > 1:
         dist.destroy_process_group()
     sess.close()
+    if mixed:
+        sess2.close()
 
 
 def run_cpu_baseline(n_dp):
