@@ -768,8 +768,7 @@ __global__ void k_resolve_pred(const uint8_t *__restrict__ payload,
                                const bydb_block_desc *__restrict__ blocks,
                                int64_t n_blocks, const uint8_t *__restrict__ pred,
                                uint64_t pred_len, int slot,
-                               PredBlock *__restrict__ out,
-                               uint32_t *__restrict__ plain_ctr) {
+                               PredBlock *__restrict__ out) {
     int64_t bi = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (bi >= n_blocks) return;
     const bydb_block_desc *bd = &blocks[bi];
@@ -786,10 +785,12 @@ __global__ void k_resolve_pred(const uint8_t *__restrict__ payload,
     const uint8_t *end = p + tlen;
     if (*p == BYDB_ENC_PLAIN) {
         // host-normalized plain column: a second pass (k_resolve_plain)
-        // fills this block's match bitmap; count arena slots here
+        // fills this block's match bitmap.  A plain stream that did NOT
+        // go through host normalization (no sidecar bit) cannot be
+        // parsed here — flag it loudly.
+        if (!in_sidecar) { pb.err = 1; out[bi] = pb; return; }
         pb.active = 1;
         pb.plain = 1;
-        atomicAdd(plain_ctr, 1u);
         out[bi] = pb;
         return;
     }
@@ -846,6 +847,44 @@ __global__ void k_resolve_pred(const uint8_t *__restrict__ payload,
         if (cnt >= bd->count) pb.uniform = 1;
     }
     out[bi] = pb;
+}
+
+// Combine the per-slot PredBlocks into one byte per block so the scan's
+// hot loop prices a non-matching block at a single byte load instead of
+// three 56-B struct loads + RLE header reads:
+//   0 = every predicated slot is uniform-match -> fold unpredicated
+//   1 = some slot misses (uniform non-match or nil tag) -> skip block
+//   2 = some slot is row-varying -> full walker path
+//   3 = unparseable tag stream -> device error
+#define PF_CLEAR 0
+#define PF_SKIP 1
+#define PF_WALK 2
+#define PF_ERR 3
+__global__ void k_combine_preds(const uint8_t *__restrict__ payload,
+                                const uint8_t *__restrict__ sidecar,
+                                const PredBlock *__restrict__ preds,
+                                int n_preds, int64_t n_blocks,
+                                uint8_t *__restrict__ flags) {
+    int64_t bi = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (bi >= n_blocks) return;
+    uint8_t f = PF_CLEAR;
+    for (int sl = 0; sl < n_preds; sl++) {
+        PredBlock pb = preds[(int64_t)sl * n_blocks + bi];
+        if (pb.err) { f = PF_ERR; break; }
+        if (!pb.active) { f = PF_SKIP; break; }
+        if (pb.uniform) {
+            // resolve the block-uniform verdict here (first run's code)
+            const uint8_t *base =
+                (pb.rle_bit_off & TAG_SIDECAR_BIT) ? sidecar : payload;
+            uint64_t bit0 = pb.rle_bit_off & ~TAG_SIDECAR_BIT;
+            uint64_t code = rd_bits_be(base, bit0, pb.width);
+            bool match = (pb.mask[(code >> 6) & 3] >> (code & 63)) & 1;
+            if (!match) { f = PF_SKIP; break; }
+        } else {
+            f = PF_WALK;  // keep scanning: a later slot may still skip
+        }
+    }
+    flags[bi] = f;
 }
 
 // Second resolve pass for plain (non-dictionary) tag columns: one wave per
@@ -1345,6 +1384,7 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
     int64_t n_blocks, int64_t min_ts, int64_t max_ts, int flags,
     const PredBlock *__restrict__ preds_in, int n_preds,
     const uint64_t *__restrict__ pred_bm,
+    const uint8_t *__restrict__ pred_flags,
     const SegEntry *__restrict__ segs_in,
     const GroupBlock *__restrict__ groups_in,
     const uint16_t *__restrict__ gmap_in, int n_gslots, int64_t gm0,
@@ -1436,34 +1476,40 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
         const uint64_t nsel = (uint64_t)(r1 - r0 + 1);
 
         // ---- per-row tag predicates (conjunctive, dictionary codes) ----
+        // One combined flag byte decides the common cases; only PF_WALK
+        // blocks pay for walker setup.
         PredWalk pw0, pw1, pw2;
         PredWalk *wp0 = nullptr, *wp1 = nullptr, *wp2 = nullptr;
-        bool skip_block = false;
         if (preds != nullptr) {
-#pragma unroll
-            for (int sl = 0; sl < 3; sl++) {
-                if (sl >= n_preds || skip_block) continue;
-                PredBlock pb = preds[(int64_t)sl * n_blocks + bi];
-                if (pb.err) {
-                    dev_set_err(derr, DERR_BAD_ENC, (uint64_t)bi);
-                    skip_block = true;
-                    continue;
-                }
-                if (!pb.active) { skip_block = true; continue; }  // nil tag
-                PredWalk *w = sl == 0 ? &pw0 : sl == 1 ? &pw1 : &pw2;
-                pred_init(w, payload, sidecar, &pb, pred_bm);
-                if (pb.uniform) {
-                    // one run covers the block: predicate is block-uniform
-                    pred_advance(w);
-                    if (!w->run_match) skip_block = true;
-                    // uniform + match: drop this walker, fold unpredicated
-                } else {
-                    if (sl == 0) wp0 = w;
-                    else if (sl == 1) wp1 = w;
-                    else wp2 = w;
-                }
+            const uint8_t pf = pred_flags[bi];
+            if (pf == PF_SKIP) continue;
+            if (pf == PF_ERR) {
+                dev_set_err(derr, DERR_BAD_ENC, (uint64_t)bi);
+                continue;
             }
-            if (skip_block) continue;
+            if (pf == PF_WALK) {
+                bool skip_block = false;
+#pragma unroll
+                for (int sl = 0; sl < 3; sl++) {
+                    if (sl >= n_preds || skip_block) continue;
+                    PredBlock pb = preds[(int64_t)sl * n_blocks + bi];
+                    if (!pb.active) { skip_block = true; continue; }
+                    PredWalk *w = sl == 0 ? &pw0 : sl == 1 ? &pw1 : &pw2;
+                    pred_init(w, payload, sidecar, &pb, pred_bm);
+                    if (pb.uniform) {
+                        // block-uniform slot inside a varying conjunction
+                        pred_advance(w);
+                        if (!w->run_match) skip_block = true;
+                        // uniform + match: walker dropped, slot is a no-op
+                    } else {
+                        if (sl == 0) wp0 = w;
+                        else if (sl == 1) wp1 = w;
+                        else wp2 = w;
+                    }
+                }
+                if (skip_block) continue;
+            }
+            // PF_CLEAR: every slot uniform-match -> fold unpredicated
         }
         bool pred_on = wp0 || wp1 || wp2;
 
@@ -1892,9 +1938,12 @@ struct bydb_session {
     // per-row match bitmaps filled by k_resolve_plain
     uint8_t *d_sidecar = nullptr;
     uint64_t sidecar_len = 0, sidecar_cap = 0;
+    uint64_t n_plain_host = 0;  // (block,slot) plain streams normalized
     uint32_t *d_plain_ctr = nullptr;
     uint64_t *d_pred_bm = nullptr;
     uint64_t pred_bm_cap = 0;  // words
+    uint8_t *d_pred_flags = nullptr;
+    int64_t pred_flags_cap = 0;
     float last_ms = 0.0f;
     bool consumed = false;
     int16_t float_exp = 0;  // shared decimal exponent for float64 restore
@@ -1958,6 +2007,7 @@ extern "C" void bydb_session_destroy(bydb_session *s) {
     if (s->d_sidecar) (void)hipFree(s->d_sidecar);
     if (s->d_plain_ctr) (void)hipFree(s->d_plain_ctr);
     if (s->d_pred_bm) (void)hipFree(s->d_pred_bm);
+    if (s->d_pred_flags) (void)hipFree(s->d_pred_flags);
     if (s->ev_start) (void)hipEventDestroy(s->ev_start);
     if (s->ev_stop) (void)hipEventDestroy(s->ev_stop);
     if (s->stream) (void)hipStreamDestroy(s->stream);
@@ -1983,6 +2033,7 @@ extern "C" int bydb_part_reserve(bydb_session *s, uint64_t payload_bytes,
     s->payload_len = 0;
     s->n_blocks = 0;
     s->sidecar_len = 0;
+    s->n_plain_host = 0;
     s->segs_built = false;
     s->groups_built = false;
     return BYDB_OK;
@@ -2035,6 +2086,7 @@ extern "C" int bydb_part_append(bydb_session *s, const uint8_t *payload,
                     s->err = "plain tag column normalization failed";
                     return BYDB_ERR_BAD_DATA;
                 }
+                s->n_plain_host++;
             } else if (src[0] == BYDB_ENC_DICTIONARY) {
                 // dictionaries whose compress_block sections are zstd'd
                 // (>=128 B) get the same host decompress; small plain
@@ -2093,6 +2145,7 @@ extern "C" int bydb_part_clear(bydb_session *s) {
     s->payload_len = 0;
     s->n_blocks = 0;
     s->sidecar_len = 0;
+    s->n_plain_host = 0;
     return BYDB_OK;
 }
 
@@ -2250,6 +2303,11 @@ extern "C" int bydb_consume_multi(bydb_session *s, int64_t min_ts,
                                  sizeof(PredBlock) * (size_t)s->n_blocks * 3));
             s->preds_cap = s->n_blocks * 3;
         }
+        if (s->pred_flags_cap < s->n_blocks) {
+            if (s->d_pred_flags) (void)hipFree(s->d_pred_flags);
+            HIP_TRY(s, hipMalloc(&s->d_pred_flags, (size_t)s->n_blocks));
+            s->pred_flags_cap = s->n_blocks;
+        }
         uint64_t total = 0;
         for (int i = 0; i < n_preds; i++) total += pred_lens[i];
         if (s->pred_bytes_cap < total) {
@@ -2260,8 +2318,6 @@ extern "C" int bydb_consume_multi(bydb_session *s, int64_t min_ts,
         uint64_t off = 0;
         int rthreads = 256;
         int rblocks = (int)((s->n_blocks + rthreads - 1) / rthreads);
-        HIP_TRY(s, hipMemsetAsync(s->d_plain_ctr, 0, sizeof(uint32_t),
-                                  s->stream));
         for (int i = 0; i < n_preds; i++) {
             HIP_TRY(s, hipMemcpyAsync(s->d_pred_bytes + off, preds_in[i],
                                       pred_lens[i], hipMemcpyHostToDevice,
@@ -2270,19 +2326,15 @@ extern "C" int bydb_consume_multi(bydb_session *s, int64_t min_ts,
                                s->stream, s->d_payload, s->d_sidecar,
                                s->d_blocks, s->n_blocks,
                                s->d_pred_bytes + off, pred_lens[i], i,
-                               s->d_preds + (int64_t)i * s->n_blocks,
-                               s->d_plain_ctr);
+                               s->d_preds + (int64_t)i * s->n_blocks);
             HIP_TRY(s, hipGetLastError());
             off += pred_lens[i];
         }
-        // plain (non-dictionary) columns found: size the match-bitmap arena
-        // and run the per-row equality pass
-        uint32_t n_plain = 0;
-        HIP_TRY(s, hipMemcpyAsync(&n_plain, s->d_plain_ctr, sizeof(uint32_t),
-                                  hipMemcpyDeviceToHost, s->stream));
-        HIP_TRY(s, hipStreamSynchronize(s->stream));
-        if (n_plain > 0) {
-            uint64_t words = (uint64_t)n_plain * PLAIN_BM_WORDS;
+        // plain (non-dictionary) columns present: the host counted every
+        // normalized (block,slot) stream at part_append, so the bitmap
+        // arena is sized without a device round-trip
+        if (s->n_plain_host > 0) {
+            uint64_t words = s->n_plain_host * PLAIN_BM_WORDS;
             if (words > s->pred_bm_cap) {
                 if (s->d_pred_bm) (void)hipFree(s->d_pred_bm);
                 HIP_TRY(s, hipMalloc(&s->d_pred_bm, words * sizeof(uint64_t)));
@@ -2303,6 +2355,10 @@ extern "C" int bydb_consume_multi(bydb_session *s, int64_t min_ts,
                 off += pred_lens[i];
             }
         }
+        hipLaunchKernelGGL(k_combine_preds, dim3(rblocks), dim3(rthreads), 0,
+                           s->stream, s->d_payload, s->d_sidecar, s->d_preds,
+                           n_preds, s->n_blocks, s->d_pred_flags);
+        HIP_TRY(s, hipGetLastError());
         preds = s->d_preds;
     }
     int flags = 0;
@@ -2376,9 +2432,9 @@ extern "C" int bydb_consume_multi(bydb_session *s, int64_t min_ts,
     const bool en_groups = groups != nullptr;
     void (*kfn)(const uint8_t *, const uint8_t *, const bydb_block_desc *,
                 int64_t, int64_t, int64_t, int, const PredBlock *, int,
-                const uint64_t *, const SegEntry *, const GroupBlock *,
-                const uint16_t *, int, int64_t, int64_t, int64_t, int64_t,
-                bydb_partial *, DevErr *);
+                const uint64_t *, const uint8_t *, const SegEntry *,
+                const GroupBlock *, const uint16_t *, int, int64_t, int64_t,
+                int64_t, int64_t, bydb_partial *, DevErr *);
     if (en_values) {
         if (en_preds) kfn = en_groups ? k_scan_agg_t<true, true, true>
                                       : k_scan_agg_t<true, true, false>;
@@ -2393,9 +2449,9 @@ extern "C" int bydb_consume_multi(bydb_session *s, int64_t min_ts,
     hipLaunchKernelGGL(kfn, dim3(grid), dim3(threads), 0, s->stream,
                        s->d_payload, s->d_sidecar, s->d_blocks, s->n_blocks,
                        min_ts, max_ts, flags, preds, n_preds, s->d_pred_bm,
-                       segs, groups, s->d_gmap, s->n_gslots, s->gmul[0],
-                       s->gmul[1], s->gmul[2], (int64_t)s->n_groups, s->d_acc,
-                       s->d_err);
+                       s->d_pred_flags, segs, groups, s->d_gmap, s->n_gslots,
+                       s->gmul[0], s->gmul[1], s->gmul[2],
+                       (int64_t)s->n_groups, s->d_acc, s->d_err);
     HIP_TRY(s, hipGetLastError());
     HIP_TRY(s, hipEventRecord(s->ev_stop, s->stream));
     s->consumed = true;
