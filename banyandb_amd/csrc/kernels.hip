@@ -61,8 +61,50 @@ __device__ __forceinline__ uint64_t lanemask_lt(int lane) {
     return (lane == 0) ? 0ull : (~0ull >> (64 - lane));
 }
 
+// DPP row-shift: pure-VALU lane shift within 16-lane rows (bound_ctrl
+// zero-fills lanes whose source is outside the row).  __shfl_up compiles
+// to ds_bpermute (an LDS op, ~40-cycle latency); a 6-step shfl scan is a
+// dependent LDS chain that dominates the scan kernels' wait time, so the
+// scans below use 4 DPP row_shr steps + v_readlane cross-row fixup —
+// zero LDS traffic.
+template <int CTRL>
+__device__ __forceinline__ uint32_t dpp_mov32(uint32_t v) {
+    return (uint32_t)__builtin_amdgcn_update_dpp(0, (int)v, CTRL, 0xF, 0xF,
+                                                 true);
+}
+
+__device__ __forceinline__ uint64_t readlane64(uint64_t v, int l) {
+    uint32_t lo = (uint32_t)__builtin_amdgcn_readlane((int)(uint32_t)v, l);
+    uint32_t hi = (uint32_t)__builtin_amdgcn_readlane((int)(uint32_t)(v >> 32),
+                                                      l);
+    return ((uint64_t)hi << 32) | lo;
+}
+
 // wave-wide inclusive scan of a u64 (wrapping adds)
 __device__ __forceinline__ uint64_t wave_incl_scan(uint64_t v, int lane) {
+#define BYDB_SCAN_STEP64(C)                                                  \
+    {                                                                        \
+        uint64_t t = ((uint64_t)dpp_mov32<C>((uint32_t)(v >> 32)) << 32) |   \
+                     dpp_mov32<C>((uint32_t)v);                              \
+        v += t;                                                              \
+    }
+    BYDB_SCAN_STEP64(0x111)   // row_shr:1
+    BYDB_SCAN_STEP64(0x112)   // row_shr:2
+    BYDB_SCAN_STEP64(0x114)   // row_shr:4
+    BYDB_SCAN_STEP64(0x118)   // row_shr:8
+#undef BYDB_SCAN_STEP64
+    uint64_t s15 = readlane64(v, 15);
+    uint64_t s31 = readlane64(v, 31);
+    uint64_t s47 = readlane64(v, 47);
+    int r = lane >> 4;
+    uint64_t add = r == 0 ? 0
+                 : r == 1 ? s15
+                 : r == 2 ? s15 + s31
+                          : s15 + s31 + s47;
+    return v + add;
+}
+
+__device__ __forceinline__ uint64_t wave_incl_scan_shfl(uint64_t v, int lane) {
 #pragma unroll
     for (int off = 1; off < 64; off <<= 1) {
         uint64_t t = (uint64_t)__shfl_up((long long)v, off);
@@ -233,12 +275,19 @@ __device__ void dense_region(const uint8_t *stream, int64_t lo, int64_t hi,
 
 // wave-wide inclusive scan of an int32 (per-window lane sums)
 __device__ __forceinline__ int32_t wave_incl_scan32(int32_t v, int lane) {
-#pragma unroll
-    for (int off = 1; off < 64; off <<= 1) {
-        int32_t t = __shfl_up(v, off);
-        if (lane >= off) v += t;
-    }
-    return v;
+    v += (int32_t)dpp_mov32<0x111>((uint32_t)v);
+    v += (int32_t)dpp_mov32<0x112>((uint32_t)v);
+    v += (int32_t)dpp_mov32<0x114>((uint32_t)v);
+    v += (int32_t)dpp_mov32<0x118>((uint32_t)v);
+    int32_t s15 = __builtin_amdgcn_readlane(v, 15);
+    int32_t s31 = __builtin_amdgcn_readlane(v, 31);
+    int32_t s47 = __builtin_amdgcn_readlane(v, 47);
+    int r = lane >> 4;
+    int32_t add = r == 0 ? 0
+                : r == 1 ? s15
+                : r == 2 ? s15 + s31
+                         : s15 + s31 + s47;
+    return v + add;
 }
 
 // Dense min/max over an all-1-byte delta stream: fold min/max of the
@@ -282,7 +331,7 @@ __device__ void dense_minmax(const uint8_t *stream, int64_t b_lo, int64_t b_hi,
                     lmx = v > lmx ? v : lmx;
                 }
             }
-            carry += (uint64_t)(int64_t)__shfl(incl, 63);
+            carry += (uint64_t)(int64_t)__builtin_amdgcn_readlane(incl, 63);
             wp += 64;
             wbase += 256;
         }
@@ -944,7 +993,7 @@ __global__ __launch_bounds__(WAVE) void k_resolve_plain(
             }
             uint64_t word = __ballot(match);
             if (lane == 0) bm[w0 + (base >> 6)] = word;
-            carry += (uint64_t)__shfl((long long)incl, WAVE - 1);
+            carry += readlane64(incl, WAVE - 1);
         }
         if (lane == 0) out[bi].rle_bit_off = w0;
     }
@@ -1006,28 +1055,63 @@ __device__ void scan_stream(const uint8_t *stream, int64_t n_deltas, bool dod,
     uint64_t d1_carry = (uint64_t)d1_init;
     uint64_t l_sum = 0, l_nsel = 0, l_nlo = 0, l_nhi = 0;
     int64_t l_mn = INT64_MAX, l_mx = INT64_MIN;
+    // Fixed 64-B window advance: every window consumes exactly 64 bytes
+    // and a varint straddling the boundary is carried in registers
+    // (carry_u holds its low groups, carry_n its byte count).  This makes
+    // the window address an induction variable, so the next window's load
+    // (b_nxt) issues a full window of decode ahead of its s_waitcnt —
+    // without it the scan is latency-bound (63% of wave cycles parked on
+    // the serial pos -> load -> ballot -> pos chain).  Terminators past
+    // jmax (bytes of the following stream / the part's 1-KiB slack) fold
+    // as zeros, so the wave-wide scans read their totals at lane 63.
+    uint64_t carry_u = 0;
+    uint32_t carry_n = 0;
+    uint32_t b_cur = stream[lane];
     while (j <= jmax) {
-        uint8_t b = stream[pos + (uint64_t)lane];
+        uint32_t b_nxt = stream[pos + 64 + (uint64_t)lane];  // prefetch
+        uint8_t b = (uint8_t)b_cur;
         uint64_t emask = __ballot(b < 0x80);
         if (emask == 0) { dev_set_err(derr, DERR_BAD_STREAM, bi); break; }
         int rank = __popcll(emask & lanemask_lt(lane));
         int64_t myj = j + rank;
         bool is_term = (b < 0x80) && (myj <= jmax);
+        int nterm = __popcll(emask);
         // fast window: every varint is 1 or 2 bytes (no two consecutive
-        // continuation bits) and the whole window is consumed — decode via
-        // the left neighbour's byte (shfl) and scan in 32-bit (window
-        // delta sums fit: |d| <= 2^13 per value)
+        // continuation bits), not the tail window, and any carried byte
+        // completes at lane 0 — decode via the left neighbour's byte
+        // (shfl) and scan in 32-bit (window delta sums fit: <= 64 * 2^21)
         uint64_t cont = ~emask;
-        if (!dod && pw0 == nullptr && pw1 == nullptr && pw2 == nullptr &&
-            (cont & (cont << 1)) == 0 && (int64_t)__popcll(emask) <= jmax - j + 1) {
+        if (pw0 == nullptr && pw1 == nullptr && pw2 == nullptr &&
+            (cont & (cont << 1)) == 0 &&
+            (int64_t)nterm <= jmax - j + 1 &&
+            (carry_n == 0 || (emask & 1))) {
             uint32_t prev = (uint32_t)__shfl_up((int)b, 1);
             bool is2 = lane > 0 && (prev & 0x80u);
             uint32_t u = is2 ? ((prev & 0x7fu) | ((uint32_t)b << 7)) : b;
+            if (lane == 0 && carry_n)
+                u = (uint32_t)carry_u | ((uint32_t)b << 7);
             int32_t d32 = (int32_t)(u >> 1) ^ -(int32_t)(u & 1);
             if (!(b < 0x80)) d32 = 0;          // continuation lanes carry 0
             int32_t s32 = wave_incl_scan32(d32, lane);
+            uint64_t sv_u;
+            if (dod) {
+                // delta-of-delta in two 32-bit scans regardless of the
+                // 64-bit d1 carry:  v_j = v_carry + cnt_j*d1_carry + t_j
+                // with t_j = scan over terminators of (scan of d2)
+                int32_t t32 = wave_incl_scan32((b < 0x80) ? s32 : 0, lane);
+                sv_u = v_carry + (uint64_t)(int64_t)(rank + 1) * d1_carry +
+                       (uint64_t)(int64_t)t32;
+                int32_t t_tot = __builtin_amdgcn_readlane(t32, 63);
+                int32_t s_tot = __builtin_amdgcn_readlane(s32, 63);
+                v_carry += (uint64_t)(int64_t)nterm * d1_carry +
+                           (uint64_t)(int64_t)t_tot;
+                d1_carry += (uint64_t)(int64_t)s_tot;
+            } else {
+                sv_u = v_carry + (uint64_t)(int64_t)s32;
+                v_carry += (uint64_t)(int64_t)__builtin_amdgcn_readlane(s32, 63);
+            }
             if (b < 0x80) {
-                int64_t sv = (int64_t)(v_carry + (uint64_t)(int64_t)s32);
+                int64_t sv = (int64_t)sv_u;
                 if (myj >= r0 && myj <= r1) {
                     l_sum += (uint64_t)sv;
                     l_nsel++;
@@ -1037,81 +1121,64 @@ __device__ void scan_stream(const uint8_t *stream, int64_t n_deltas, bool dod,
                 if (sv < lo_bound) l_nlo++;
                 if (sv > hi_bound) l_nhi++;
             }
-            int last_lane = 63 - __clzll(emask);
-            v_carry += (uint64_t)(int64_t)__shfl(s32, last_lane);
-            int nterm = __popcll(emask);
-            if (j + nterm > jmax) break;
+            if (emask >> 63) {
+                carry_n = 0;
+                carry_u = 0;
+            } else {
+                carry_n = 1;
+                carry_u = (uint64_t)((uint32_t)__builtin_amdgcn_readlane((int)b, 63) & 0x7f);
+            }
             j += nterm;
-            pos += (uint64_t)(last_lane + 1);
+            pos += 64;
+            b_cur = b_nxt;
             continue;
         }
         uint64_t d = 0;
-        if (emask == ~0ull) {
+        if (emask == ~0ull && carry_n == 0) {
             d = (uint64_t)zz_dec(b);
-        } else if (is_term) {
+        } else {
+            // assemble each terminator's varint from the bytes already in
+            // registers: byte k back arrives by one __shfl_up at a
+            // wave-uniform distance (trip count = longest in-window varint
+            // from the widest terminator gap); the first terminator also
+            // prepends the carried bytes.  LSB-first groups
+            // (int.go:81-103): the byte k back holds bits (len-1-k)*7.
             uint64_t below = emask & lanemask_lt(lane);
             int start = below ? (64 - __clzll(below)) : 0;
-            uint64_t u = 0;
-            unsigned sh = 0;
-            for (int i = start; i < lane; ++i) {
-                u |= (uint64_t)(stream[pos + (uint64_t)i] & 0x7f) << sh;
-                sh += 7;
+            int mylen = lane - start + 1;        // in-window bytes
+            if (mylen > 10) mylen = 10;  // valid varints are <=10 B; longer
+                                         // gaps only occur on corrupt input
+            // wave-uniform max varint length over terminator lanes
+            int maxlen = (b < 0x80) ? mylen : 1;
+            for (int off2 = 32; off2 > 0; off2 >>= 1) {
+                int t = __shfl_xor(maxlen, off2);
+                maxlen = t > maxlen ? t : maxlen;
             }
-            u |= (uint64_t)b << sh;
-            d = (uint64_t)zz_dec(u);
+            uint64_t u = (uint64_t)(b & 0x7f) << (7 * (mylen - 1));
+            for (int k = 1; k < maxlen; ++k) {
+                uint32_t bk = (uint32_t)__shfl_up((int)b, k);
+                if (k < mylen)
+                    u |= (uint64_t)(bk & 0x7f) << (7 * (mylen - 1 - k));
+            }
+            if (below == 0 && carry_n)
+                u = carry_u | (u << (7 * carry_n));
+            if (is_term) d = (uint64_t)zz_dec(u);
         }
         if (!is_term) d = 0;
         uint64_t val;
-        uint64_t tot;
         if (dod) {
             // first scan: d1_j = d1_carry + incl_scan(d2)
             uint64_t s1 = wave_incl_scan(d, lane);
             uint64_t d1j = is_term ? (d1_carry + s1) : 0;
-            uint64_t s1_tot = (uint64_t)__shfl((long long)s1, 63);
             // second scan: v_j = v_carry + incl_scan(d1_j over values)
             uint64_t s2 = wave_incl_scan(d1j, lane);
             val = v_carry + s2;
-            // carry advance = scans at the last CONSUMED terminator lane;
-            // the window is fully consumed except at the stream tail, where
-            // the nvals-th set bit is searched (once per stream)
-            int nterm_all = __popcll(emask);
-            int64_t nvals = (jmax - j + 1) < (int64_t)nterm_all
-                                ? (jmax - j + 1) : (int64_t)nterm_all;
-            int last_lane;
-            if (nvals == (int64_t)nterm_all) {
-                last_lane = 63 - __clzll(emask);
-            } else {
-                uint64_t mm = emask;
-                last_lane = 0;
-                for (int t = 0; t < nvals; t++) {
-                    last_lane = __ffsll((unsigned long long)mm) - 1;
-                    mm &= mm - 1;
-                }
-            }
-            tot = (uint64_t)__shfl((long long)s2, last_lane);
-            v_carry += tot;
-            uint64_t d1tot = (uint64_t)__shfl((long long)s1, last_lane);
-            d1_carry += d1tot;
-            (void)s1_tot;
+            v_carry += readlane64(s2, 63);
+            d1_carry += readlane64(s1, 63);
         } else {
-            uint64_t s = wave_incl_scan(d, lane);
-            val = v_carry + s;
-            int nterm_all = __popcll(emask);
-            int64_t nvals = (jmax - j + 1) < (int64_t)nterm_all
-                                ? (jmax - j + 1) : (int64_t)nterm_all;
-            int last_lane;
-            if (nvals == (int64_t)nterm_all) {
-                last_lane = 63 - __clzll(emask);
-            } else {
-                uint64_t mm = emask;
-                last_lane = 0;
-                for (int t = 0; t < nvals; t++) {
-                    last_lane = __ffsll((unsigned long long)mm) - 1;
-                    mm &= mm - 1;
-                }
-            }
-            tot = (uint64_t)__shfl((long long)s, last_lane);
-            v_carry += tot;
+            uint64_t sscan = wave_incl_scan(d, lane);
+            val = v_carry + sscan;
+            v_carry += readlane64(sscan, 63);
         }
         bool in_sel = is_term && myj >= r0 && myj <= r1;
         if (pw0) in_sel = pred_match_rows(pw0, myj, in_sel) && in_sel;
@@ -1130,10 +1197,28 @@ __device__ void scan_stream(const uint8_t *stream, int64_t n_deltas, bool dod,
                 if (sv > hi_bound) l_nhi++;
             }
         }
-        int nterm = __popcll(emask);
         if (j + nterm > jmax) break;
+        // carry out: bytes after the last terminator head the next value
+        int last_lane = 63 - __clzll(emask);
+        int n_tail = 63 - last_lane;
+        if (n_tail) {
+            uint64_t tail = 0;
+            if (lane > last_lane) {
+                int sh2 = lane - last_lane - 1;
+                if (sh2 > 9) sh2 = 9;        // corrupt-input guard
+                tail = (uint64_t)(b & 0x7f) << (7 * sh2);
+            }
+            for (int off2 = 32; off2 > 0; off2 >>= 1)
+                tail |= (uint64_t)__shfl_xor((long long)tail, off2);
+            carry_u = tail;
+            carry_n = n_tail > 9 ? 9 : (uint32_t)n_tail;
+        } else {
+            carry_u = 0;
+            carry_n = 0;
+        }
         j += nterm;
-        pos += (uint64_t)(64 - __clzll(emask));
+        pos += 64;
+        b_cur = b_nxt;
     }
     f->sum = l_sum;
     f->mn = l_mn;
@@ -1161,6 +1246,7 @@ struct SegEntry {
     uint32_t byte_off;   // offset of the segment's first delta in the stream
     uint32_t _pad;
     int64_t v_start;     // value at row seg*SEG_ROWS (before the first delta)
+    int64_t d1_start;    // delta-of-delta streams: d1 at row seg*SEG_ROWS
 };
 
 __global__ __launch_bounds__(256) void k_build_seg_index(
@@ -1173,23 +1259,40 @@ __global__ __launch_bounds__(256) void k_build_seg_index(
         const bydb_block_desc *bd = &blocks[bi];
         SegEntry *bseg = segs + bi * MAX_SEGS;
         const int64_t n = (int64_t)bd->count;
-        bool eligible = bd->field_enc == BYDB_ENC_DELTA && n > SEG_ROWS &&
-                        bd->field_len != (uint64_t)(n - 1);
+        const bool dod = bd->field_enc == BYDB_ENC_DELTA_OF_DELTA;
+        bool eligible =
+            n > SEG_ROWS &&
+            ((bd->field_enc == BYDB_ENC_DELTA &&
+              bd->field_len != (uint64_t)(n - 1)) ||
+             dod);
         if (!eligible) {
             if (lane == 0) bseg[0].byte_off = SEG_INELIGIBLE;
             continue;
         }
-        if (lane == 0) {
-            bseg[0].byte_off = 0;
-            bseg[0].v_start = bd->field_first;
-        }
         const uint8_t *stream = payload + bd->field_off;
-        const int64_t n_deltas = n - 1;
+        uint64_t d2_off = 0;
+        uint64_t d1_carry = 0;
+        if (dod) {
+            int vl;
+            d1_carry = (uint64_t)decode_one_varint(stream, &vl);
+            d2_off = (uint64_t)vl;
+        }
+        if (lane == 0) {
+            bseg[0].byte_off = (uint32_t)d2_off;
+            bseg[0].v_start =
+                dod ? (int64_t)((uint64_t)bd->field_first + d1_carry)
+                    : bd->field_first;
+            bseg[0].d1_start = (int64_t)d1_carry;
+        }
+        const uint8_t *d2s = stream + d2_off;
+        const int64_t n_deltas = n - 1 - (dod ? 1 : 0);
+        // j here indexes the per-stream varint: absolute row = j (delta)
+        // or j + 1 (delta-of-delta)
         uint64_t pos = 0;
         int64_t j = 1;
-        uint64_t v_carry = (uint64_t)bd->field_first;
+        uint64_t v_carry = (uint64_t)bd->field_first + (dod ? d1_carry : 0);
         while (j <= n_deltas) {
-            uint8_t b = stream[pos + (uint64_t)lane];
+            uint8_t b = d2s[pos + (uint64_t)lane];
             uint64_t emask = __ballot(b < 0x80);
             if (emask == 0) { dev_set_err(derr, DERR_BAD_STREAM, (uint64_t)bi); break; }
             int rank = __popcll(emask & lanemask_lt(lane));
@@ -1204,22 +1307,34 @@ __global__ __launch_bounds__(256) void k_build_seg_index(
                 uint64_t u = 0;
                 unsigned sh = 0;
                 for (int i = start; i < lane; ++i) {
-                    u |= (uint64_t)(stream[pos + (uint64_t)i] & 0x7f) << sh;
+                    u |= (uint64_t)(d2s[pos + (uint64_t)i] & 0x7f) << sh;
                     sh += 7;
                 }
                 u |= (uint64_t)b << sh;
                 d = (uint64_t)zz_dec(u);
             }
             if (!is_term) d = 0;
-            uint64_t s = wave_incl_scan(d, lane);
-            // a lane whose value index is a SEG_ROWS multiple records the
-            // next segment's start (byte after my terminator; v = v_myj)
-            if (is_term && (myj % SEG_ROWS) == 0 && myj / SEG_ROWS < MAX_SEGS) {
+            uint64_t s1 = wave_incl_scan(d, lane);
+            uint64_t vinc;
+            uint64_t d1j = 0;
+            if (dod) {
+                // v advances by d1_j = d1_carry + scan(d2); scan the d1_j
+                d1j = is_term ? (d1_carry + s1) : 0;
+                vinc = wave_incl_scan(d1j, lane);
+            } else {
+                vinc = s1;
+            }
+            // a lane whose absolute row is a SEG_ROWS multiple records the
+            // next segment's start (byte after my terminator)
+            int64_t myrow = myj + (dod ? 1 : 0);
+            if (is_term && (myrow % SEG_ROWS) == 0 &&
+                myrow / SEG_ROWS < MAX_SEGS) {
                 SegEntry e;
-                e.byte_off = (uint32_t)(pos + (uint64_t)lane + 1);
+                e.byte_off = (uint32_t)(d2_off + pos + (uint64_t)lane + 1);
                 e._pad = 0;
-                e.v_start = (int64_t)(v_carry + s);
-                bseg[myj / SEG_ROWS] = e;
+                e.v_start = (int64_t)(v_carry + vinc);
+                e.d1_start = (int64_t)(d1_carry + s1);
+                bseg[myrow / SEG_ROWS] = e;
             }
             int nterm_all = __popcll(emask);
             int64_t nvals = (n_deltas - j + 1) < (int64_t)nterm_all
@@ -1235,7 +1350,8 @@ __global__ __launch_bounds__(256) void k_build_seg_index(
                     mm &= mm - 1;
                 }
             }
-            v_carry += (uint64_t)__shfl((long long)s, last_lane);
+            v_carry += (uint64_t)__shfl((long long)vinc, last_lane);
+            if (dod) d1_carry += (uint64_t)__shfl((long long)s1, last_lane);
             if (j + nterm_all > n_deltas) break;
             j += nterm_all;
             pos += (uint64_t)(64 - __clzll(emask));
@@ -1378,6 +1494,10 @@ __device__ void fold_range(const uint8_t *fstream, uint8_t fenc, int64_t first,
 // register pressure; the full one keeps everything.  (EN_* are constant
 // guards — dead branches are eliminated per instantiation.)
 template <bool EN_VALUES, bool EN_PREDS, bool EN_GROUPS>
+// Occupancy: the value-scan instantiations are scattered-load
+// latency-bound (63% of wave cycles parked), so they request 6 waves/SIMD
+// even at the cost of tighter register allocation; the closed-form-only
+// instantiation already fits 6.
 __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) void k_scan_agg_t(
     const uint8_t *__restrict__ payload, const uint8_t *__restrict__ sidecar,
     const bydb_block_desc *__restrict__ blocks,
@@ -1629,9 +1749,11 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
                 continue;
             }
         }
-        const bool scan_path = EN_VALUES && (flags & KF_NEED_VALUES) &&
-                               bd->field_enc == BYDB_ENC_DELTA &&
-                               bd->field_len != (uint64_t)(n - 1);
+        const bool scan_path =
+            EN_VALUES && (flags & KF_NEED_VALUES) &&
+            ((bd->field_enc == BYDB_ENC_DELTA &&
+              bd->field_len != (uint64_t)(n - 1)) ||
+             bd->field_enc == BYDB_ENC_DELTA_OF_DELTA);
         const bool use_seg = seg_eligible && scan_path && !pred_on;
         if (!use_seg && seg != 0) continue;
 
@@ -1758,11 +1880,15 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
                         bsum = (uint64_t)first * nsel + (uint64_t)d1 * si + acc;
                     }
                 }
-            } else if (!dod && use_seg) {
+            } else if (use_seg) {
                 // segment-parallel value scan: this wave folds only rows
-                // [seg*SEG_ROWS+1 .. min((seg+1)*SEG_ROWS, n-1)] (+ row 0
-                // on segment 0), starting from the indexed byte offset and
-                // carry value — no cross-segment dependency
+                // [seg*SEG_ROWS+1 .. min((seg+1)*SEG_ROWS, n-1)] (+ rows 0
+                // and, for delta-of-delta, 1 on segment 0), starting from
+                // the indexed byte offset and carries — no cross-segment
+                // dependency.  For delta-of-delta, segment k >= 1 relabels
+                // absolute row (k*SEG_ROWS - 1 + j'), so j' = 2 lands on
+                // the segment's first row and e.v_start/e.d1_start are the
+                // carries at row k*SEG_ROWS.
                 const SegEntry e = segs[bi * MAX_SEGS + seg];
                 int64_t sj_lo = (int64_t)seg * SEG_ROWS + 1;
                 int64_t sj_hi = ((int64_t)seg + 1) * SEG_ROWS;
@@ -1770,16 +1896,20 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
                 int64_t a = sj_lo > r0 ? sj_lo : r0;
                 int64_t b = sj_hi < r1 ? sj_hi : r1;
                 bool handle0 = seg == 0 && r0 <= 0 && 0 <= r1;
-                if (a > b && !handle0) continue;
+                bool handle1 = dod && seg == 0 && r0 <= 1 && 1 <= r1;
+                if (dod && seg == 0 && a < 2) a = 2;
+                if (a > b && !handle0 && !handle1) continue;
                 uint64_t lsum = 0, lcnt = 0;
                 int64_t lmn = INT64_MAX, lmx = INT64_MIN;
                 if (a <= b) {
                     ScanFold ff;
-                    int64_t rel0 = (int64_t)seg * SEG_ROWS;
-                    scan_stream(fstream + e.byte_off, b - rel0, false,
-                                e.v_start, 0, a - rel0, b - rel0, INT64_MAX,
-                                INT64_MIN, lane, &ff, derr, (uint64_t)bi,
-                                nullptr, nullptr, nullptr);
+                    int64_t rel0 = dod ? ((int64_t)seg * SEG_ROWS -
+                                          (seg ? 1 : 0))
+                                       : (int64_t)seg * SEG_ROWS;
+                    scan_stream(fstream + e.byte_off, b - rel0, dod,
+                                e.v_start, e.d1_start, a - rel0, b - rel0,
+                                INT64_MAX, INT64_MIN, lane, &ff, derr,
+                                (uint64_t)bi, nullptr, nullptr, nullptr);
                     lsum = ff.sum;
                     lcnt = ff.nsel;
                     lmn = ff.mn;
@@ -1790,6 +1920,13 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
                     lcnt++;
                     lmn = first < lmn ? first : lmn;
                     lmx = first > lmx ? first : lmx;
+                }
+                if (lane == 0 && handle1) {
+                    int64_t row1 = e.v_start;  // seg 0: v at row 1
+                    lsum += (uint64_t)row1;
+                    lcnt++;
+                    lmn = row1 < lmn ? row1 : lmn;
+                    lmx = row1 > lmx ? row1 : lmx;
                 }
                 bsum = wave_reduce_add(lsum);
                 nsel_eff = wave_reduce_add(lcnt);
