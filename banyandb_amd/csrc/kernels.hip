@@ -104,15 +104,6 @@ __device__ __forceinline__ uint64_t wave_incl_scan(uint64_t v, int lane) {
     return v + add;
 }
 
-__device__ __forceinline__ uint64_t wave_incl_scan_shfl(uint64_t v, int lane) {
-#pragma unroll
-    for (int off = 1; off < 64; off <<= 1) {
-        uint64_t t = (uint64_t)__shfl_up((long long)v, off);
-        if (lane >= off) v += t;
-    }
-    return v;
-}
-
 __device__ __forceinline__ uint64_t wave_reduce_add(uint64_t v) {
 #pragma unroll
     for (int off = 32; off > 0; off >>= 1) v += (uint64_t)__shfl_xor((long long)v, off);
