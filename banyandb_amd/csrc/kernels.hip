@@ -803,22 +803,26 @@ __global__ void k_resolve_groups(const uint8_t *__restrict__ payload,
 // builds the code mask.  Only plain (<128 B) compress_block sections are
 // parseable on device (bytes.go:291-303); zstd-compressed dictionaries are
 // flagged and surfaced as a device error if a predicate needs them.
-__global__ void k_resolve_pred(const uint8_t *__restrict__ payload,
-                               const uint8_t *__restrict__ sidecar,
-                               const bydb_block_desc *__restrict__ blocks,
-                               int64_t n_blocks, const uint8_t *__restrict__ pred,
-                               uint64_t pred_len, int slot,
-                               PredBlock *__restrict__ out) {
-    int64_t bi = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    if (bi >= n_blocks) return;
-    const bydb_block_desc *bd = &blocks[bi];
+#define PF_CLEAR 0
+#define PF_SKIP 1
+#define PF_WALK 2
+#define PF_ERR 3
+
+// Resolve one predicate slot for one block; returns the slot verdict
+// (PF_CLEAR uniform-match / PF_SKIP miss-or-nil / PF_WALK row-varying /
+// PF_ERR unparseable) and fills *out for the walker path.
+__device__ int resolve_one_pred(const uint8_t *__restrict__ payload,
+                                const uint8_t *__restrict__ sidecar,
+                                const bydb_block_desc *__restrict__ bd,
+                                int slot, const uint8_t *__restrict__ pred,
+                                uint64_t pred_len, PredBlock *out) {
     uint64_t toff = slot == 0 ? bd->tag_off : slot == 1 ? bd->tag2_off : bd->tag3_off;
     uint64_t tlen = slot == 0 ? bd->tag_len : slot == 1 ? bd->tag2_len : bd->tag3_len;
     PredBlock pb;
     for (int i = 0; i < 4; i++) pb.mask[i] = 0;
     pb.rle_bit_off = 0; pb.nentries = 0; pb.width = 0;
     pb.active = 0; pb.err = 0; pb.uniform = 0; pb.plain = 0;
-    if (tlen == 0) { out[bi] = pb; return; }
+    if (tlen == 0) { *out = pb; return PF_SKIP; }   // nil tag: never equal
     const bool in_sidecar = (toff & TAG_SIDECAR_BIT) != 0;
     const uint8_t *base = in_sidecar ? sidecar : payload;
     const uint8_t *p = base + (toff & ~TAG_SIDECAR_BIT);
@@ -828,13 +832,13 @@ __global__ void k_resolve_pred(const uint8_t *__restrict__ payload,
         // fills this block's match bitmap.  A plain stream that did NOT
         // go through host normalization (no sidecar bit) cannot be
         // parsed here — flag it loudly.
-        if (!in_sidecar) { pb.err = 1; out[bi] = pb; return; }
+        if (!in_sidecar) { pb.err = 1; *out = pb; return PF_ERR; }
         pb.active = 1;
         pb.plain = 1;
-        out[bi] = pb;
-        return;
+        *out = pb;
+        return PF_WALK;
     }
-    if (*p != BYDB_ENC_DICTIONARY) { pb.err = 1; out[bi] = pb; return; }
+    if (*p != BYDB_ENC_DICTIONARY) { pb.err = 1; *out = pb; return PF_ERR; }
     p++;
     pb.active = 1;
     // varuint count (int.go:152-199)
@@ -849,12 +853,12 @@ __global__ void k_resolve_pred(const uint8_t *__restrict__ payload,
     // lengths block: compress_block(u64list) — plain or host-normalized
     uint64_t ll = 0;
     const uint8_t *lens_blk = dict_section(p, end, &ll);
-    if (!lens_blk) { pb.err = 1; out[bi] = pb; return; }
+    if (!lens_blk) { pb.err = 1; *out = pb; return PF_ERR; }
     p = lens_blk + ll;
     // values payload block
     uint64_t vl = 0;
     const uint8_t *vals = dict_section(p, end, &vl);
-    if (!vals) { pb.err = 1; out[bi] = pb; return; }
+    if (!vals) { pb.err = 1; *out = pb; return PF_ERR; }
     p = vals + vl;
     // parse width-typed lengths (bytes.go:205-235)
     uint8_t wt = lens_blk[0];
@@ -882,47 +886,47 @@ __global__ void k_resolve_pred(const uint8_t *__restrict__ payload,
     pb.nentries = nentries;
     pb.width = (uint8_t)width;
     pb.rle_bit_off = (bit0 + 40) | (in_sidecar ? TAG_SIDECAR_BIT : 0);
+    int verdict = PF_WALK;
     if (nentries == 2) {
+        uint64_t code = rd_bits_be(base, bit0 + 40, width);
         uint64_t cnt = rd_bits_be(base, bit0 + 40 + width, width);
-        if (cnt >= bd->count) pb.uniform = 1;
+        if (cnt >= bd->count) {
+            pb.uniform = 1;
+            bool match = (pb.mask[(code >> 6) & 3] >> (code & 63)) & 1;
+            verdict = match ? PF_CLEAR : PF_SKIP;
+        }
     }
-    out[bi] = pb;
+    *out = pb;
+    return verdict;
 }
 
-// Combine the per-slot PredBlocks into one byte per block so the scan's
-// hot loop prices a non-matching block at a single byte load instead of
-// three 56-B struct loads + RLE header reads:
+// One fused launch resolves every predicate slot for a block (the
+// descriptor is read once) and writes the combined flag byte the scan's
+// hot loop prices a non-matching block at:
 //   0 = every predicated slot is uniform-match -> fold unpredicated
 //   1 = some slot misses (uniform non-match or nil tag) -> skip block
 //   2 = some slot is row-varying -> full walker path
 //   3 = unparseable tag stream -> device error
-#define PF_CLEAR 0
-#define PF_SKIP 1
-#define PF_WALK 2
-#define PF_ERR 3
-__global__ void k_combine_preds(const uint8_t *__restrict__ payload,
-                                const uint8_t *__restrict__ sidecar,
-                                const PredBlock *__restrict__ preds,
-                                int n_preds, int64_t n_blocks,
-                                uint8_t *__restrict__ flags) {
+__global__ void k_resolve_preds(
+    const uint8_t *__restrict__ payload, const uint8_t *__restrict__ sidecar,
+    const bydb_block_desc *__restrict__ blocks, int64_t n_blocks,
+    const uint8_t *__restrict__ pred_bytes, uint64_t o0, uint64_t l0,
+    uint64_t o1, uint64_t l1, uint64_t o2, uint64_t l2, int n_preds,
+    PredBlock *__restrict__ out, uint8_t *__restrict__ flags) {
     int64_t bi = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (bi >= n_blocks) return;
+    const bydb_block_desc *bd = &blocks[bi];
+    const uint64_t offs[3] = {o0, o1, o2};
+    const uint64_t lens[3] = {l0, l1, l2};
     uint8_t f = PF_CLEAR;
     for (int sl = 0; sl < n_preds; sl++) {
-        PredBlock pb = preds[(int64_t)sl * n_blocks + bi];
-        if (pb.err) { f = PF_ERR; break; }
-        if (!pb.active) { f = PF_SKIP; break; }
-        if (pb.uniform) {
-            // resolve the block-uniform verdict here (first run's code)
-            const uint8_t *base =
-                (pb.rle_bit_off & TAG_SIDECAR_BIT) ? sidecar : payload;
-            uint64_t bit0 = pb.rle_bit_off & ~TAG_SIDECAR_BIT;
-            uint64_t code = rd_bits_be(base, bit0, pb.width);
-            bool match = (pb.mask[(code >> 6) & 3] >> (code & 63)) & 1;
-            if (!match) { f = PF_SKIP; break; }
-        } else {
-            f = PF_WALK;  // keep scanning: a later slot may still skip
-        }
+        PredBlock pb;
+        int v = resolve_one_pred(payload, sidecar, bd, sl,
+                                 pred_bytes + offs[sl], lens[sl], &pb);
+        out[(int64_t)sl * n_blocks + bi] = pb;
+        if (v == PF_ERR) { f = PF_ERR; break; }
+        if (v == PF_SKIP) { f = PF_SKIP; break; }
+        if (v == PF_WALK) f = PF_WALK;
     }
     flags[bi] = f;
 }
@@ -2444,20 +2448,25 @@ extern "C" int bydb_consume_multi(bydb_session *s, int64_t min_ts,
             s->pred_bytes_cap = total;
         }
         uint64_t off = 0;
+        uint64_t offs[3] = {0, 0, 0};
+        uint64_t lens3[3] = {0, 0, 0};
         int rthreads = 256;
         int rblocks = (int)((s->n_blocks + rthreads - 1) / rthreads);
         for (int i = 0; i < n_preds; i++) {
             HIP_TRY(s, hipMemcpyAsync(s->d_pred_bytes + off, preds_in[i],
                                       pred_lens[i], hipMemcpyHostToDevice,
                                       s->stream));
-            hipLaunchKernelGGL(k_resolve_pred, dim3(rblocks), dim3(rthreads), 0,
-                               s->stream, s->d_payload, s->d_sidecar,
-                               s->d_blocks, s->n_blocks,
-                               s->d_pred_bytes + off, pred_lens[i], i,
-                               s->d_preds + (int64_t)i * s->n_blocks);
-            HIP_TRY(s, hipGetLastError());
+            offs[i] = off;
+            lens3[i] = pred_lens[i];
             off += pred_lens[i];
         }
+        // one fused launch resolves all slots + the combined flag byte
+        hipLaunchKernelGGL(k_resolve_preds, dim3(rblocks), dim3(rthreads), 0,
+                           s->stream, s->d_payload, s->d_sidecar, s->d_blocks,
+                           s->n_blocks, s->d_pred_bytes, offs[0], lens3[0],
+                           offs[1], lens3[1], offs[2], lens3[2], n_preds,
+                           s->d_preds, s->d_pred_flags);
+        HIP_TRY(s, hipGetLastError());
         // plain (non-dictionary) columns present: the host counted every
         // normalized (block,slot) stream at part_append, so the bitmap
         // arena is sized without a device round-trip
@@ -2483,10 +2492,6 @@ extern "C" int bydb_consume_multi(bydb_session *s, int64_t min_ts,
                 off += pred_lens[i];
             }
         }
-        hipLaunchKernelGGL(k_combine_preds, dim3(rblocks), dim3(rthreads), 0,
-                           s->stream, s->d_payload, s->d_sidecar, s->d_preds,
-                           n_preds, s->n_blocks, s->d_pred_flags);
-        HIP_TRY(s, hipGetLastError());
         preds = s->d_preds;
     }
     int flags = 0;
