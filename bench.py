@@ -304,6 +304,8 @@ def main():
             tj = json.load(open(tpath))
             if tj.get("workload") == workload:
                 traffic = tj.get("bytes_per_launch")
+            else:
+                traffic = (tj.get("workloads") or {}).get(workload)
         except Exception:
             pass
 
