@@ -152,6 +152,9 @@ def _load():
     lib.bydb_part_builder_destroy.argtypes = [C.c_void_p]
     lib.bydb_part_builder_error.restype = C.c_char_p
     lib.bydb_part_builder_error.argtypes = [C.c_void_p]
+    lib.bydb_part_builder_add_block_i64_nullable.restype = C.c_int
+    lib.bydb_part_builder_add_block_i64_nullable.argtypes = [
+        C.c_void_p, C.c_uint64, i64p, i64p, i64p, u8p, C.c_int64, C.c_uint32]
     lib.bydb_part_builder_add_block_i64.restype = C.c_int
     lib.bydb_part_builder_add_block_i64.argtypes = [C.c_void_p, C.c_uint64, i64p,
                                                     i64p, i64p, C.c_int64, C.c_uint32]
@@ -225,6 +228,18 @@ class PartBuilder:
         self._ck(_lib.bydb_part_builder_add_block_i64(
             self._h, series_id, (C.c_int64 * n)(*ts), (C.c_int64 * n)(*versions),
             (C.c_int64 * n)(*vals), n, group_code))
+
+    def add_block_i64_nullable(self, series_id, ts, versions, vals,
+                               group_code=0):
+        """Null-bearing int64 field column: vals[i] is None for null rows
+        (stored as the reference's Plain cell block; the fold skips
+        nulls)."""
+        n = len(ts)
+        valid = (C.c_uint8 * n)(*[0 if v is None else 1 for v in vals])
+        vv = (C.c_int64 * n)(*[0 if v is None else v for v in vals])
+        self._ck(_lib.bydb_part_builder_add_block_i64_nullable(
+            self._h, series_id, (C.c_int64 * n)(*ts),
+            (C.c_int64 * n)(*versions), vv, valid, n, group_code))
 
     def add_block_f64(self, series_id, ts, versions, vals, group_code=0):
         n = len(ts)

@@ -532,6 +532,57 @@ bool bydb_normalize_dict_tag(const uint8_t *src, uint64_t src_len,
     return true;
 }
 
+// Normalize a Plain (null-bearing) FIELD column — the 8-B sign-flip cell
+// block written above — into the device fold form:
+// [u32le n][u32le 0][validity bitmap ceil(n/64)*8 B, bit=1 valid]
+// [n x 8-B cells big-endian, null rows zeroed].  8-B aligned throughout
+// so the kernel loads cells as u64.
+bool bydb_normalize_plain_field(const uint8_t *src, uint64_t src_len,
+                                uint64_t nrows, std::vector<uint8_t> &out) {
+    if (src_len < 1 || src[0] != BYDB_ENC_PLAIN) return false;
+    const uint8_t *p = src + 1;
+    uint64_t rem = src_len - 1, used = 0;
+    std::vector<uint8_t> lens;
+    if (!decompress_block_host(lens, p, rem, 1 + nrows * 8, &used)) return false;
+    p += used;
+    rem -= used;
+    if (lens.empty()) return false;
+    uint8_t wt = lens[0];
+    uint32_t wbytes = wt == 0 ? 1 : wt == 1 ? 2 : wt == 2 ? 4 : 8;
+    if (lens.size() != 1 + (size_t)nrows * wbytes) return false;
+    uint64_t total = 0;
+    for (uint64_t i = 0; i < nrows; i++) {
+        uint64_t ap1 = 0;
+        for (uint32_t b = 0; b < wbytes; b++)
+            ap1 = (ap1 << 8) | lens[1 + i * wbytes + b];
+        if (ap1 == 0) continue;
+        if (ap1 != 9) return false;  // cells are exactly 8 bytes
+        total += 8;
+    }
+    std::vector<uint8_t> vals;
+    if (!decompress_block_host(vals, p, rem, total, &used)) return false;
+    if (used != rem || vals.size() != total) return false;
+    uint64_t nw = (nrows + 63) / 64;
+    out.assign(8 + nw * 8 + nrows * 8, 0);
+    out[0] = (uint8_t)nrows;
+    out[1] = (uint8_t)(nrows >> 8);
+    out[2] = (uint8_t)(nrows >> 16);
+    out[3] = (uint8_t)(nrows >> 24);
+    uint8_t *bm = out.data() + 8;
+    uint8_t *cells = out.data() + 8 + nw * 8;
+    uint64_t voff = 0;
+    for (uint64_t i = 0; i < nrows; i++) {
+        uint64_t ap1 = 0;
+        for (uint32_t b = 0; b < wbytes; b++)
+            ap1 = (ap1 << 8) | lens[1 + i * wbytes + b];
+        if (ap1 == 0) continue;
+        bm[(i >> 3)] |= (uint8_t)(1u << (i & 7));
+        memcpy(cells + i * 8, vals.data() + voff, 8);
+        voff += 8;
+    }
+    return true;
+}
+
 // ===================== part builder =====================
 struct bydb_part_builder {
     std::vector<uint8_t> payload;
@@ -560,7 +611,8 @@ extern "C" const char *bydb_part_builder_error(bydb_part_builder *b) {
 static int add_block_common(bydb_part_builder *b, uint64_t series_id,
                             const int64_t *ts, const int64_t *versions,
                             const int64_t *field_ints, int16_t exp,
-                            uint8_t vtype, int64_t n, uint32_t group_code) {
+                            uint8_t vtype, int64_t n, uint32_t group_code,
+                            const uint8_t *valid = nullptr) {
     if (n < 1 || n > 8192) {  // measure.go:41-46
         b->err = "block row count out of range";
         return BYDB_ERR_BAD_ARG;
@@ -599,14 +651,54 @@ static int add_block_common(bydb_part_builder *b, uint64_t series_id,
     // field column stream (header fields parsed into the desc; the payload
     // keeps only the varint stream, header-free — kernels read streams)
     size_t fstart = b->payload.size();
-    int64_t ffirst;
-    uint8_t fenc = int64_list_append(b->payload, field_ints, n, &ffirst);
-    d.field_enc = fenc;
-    d.field_first = ffirst;
+    if (valid) {
+        // null-bearing column: the int-list encodings cannot represent
+        // nulls, so the writer stores Plain — a bytes block of 8-B
+        // sign-flip cells (convert/number.go:33-46), nil rows zero-length
+        // (column.go:214-233 + :266-278 fallback).  The fold skips nulls
+        // (vectorized/measure/aggregation.go:310 null check).
+        b->payload.push_back(BYDB_ENC_PLAIN);
+        std::vector<uint8_t> cells;
+        std::vector<int64_t> clens((size_t)n);
+        cells.reserve((size_t)n * 8);
+        for (int64_t i = 0; i < n; i++) {
+            if (valid[i]) {
+                cell_append(cells, field_ints[i]);
+                clens[(size_t)i] = 8;
+            } else {
+                clens[(size_t)i] = -1;
+            }
+        }
+        if (!bytes_block_append(b->payload, cells.data(), clens.data(), n)) {
+            b->err = "cell block encode failed (zstd unavailable?)";
+            b->payload.resize(start);
+            return BYDB_ERR_BAD_DATA;
+        }
+        d.field_enc = BYDB_ENC_PLAIN;
+        d.field_first = 0;
+    } else {
+        int64_t ffirst;
+        uint8_t fenc = int64_list_append(b->payload, field_ints, n, &ffirst);
+        d.field_enc = fenc;
+        d.field_first = ffirst;
+    }
     d.field_off = b->base_off + fstart;
     d.field_len = b->payload.size() - fstart;
     b->blocks.push_back(d);
     return BYDB_OK;
+}
+
+// Null-bearing int64 column: valid[i] == 0 marks row i null.
+extern "C" int bydb_part_builder_add_block_i64_nullable(
+    bydb_part_builder *b, uint64_t series_id, const int64_t *ts,
+    const int64_t *versions, const int64_t *vals, const uint8_t *valid,
+    int64_t n, uint32_t group_code) {
+    if (!valid) {
+        b->err = "valid mask required";
+        return BYDB_ERR_BAD_ARG;
+    }
+    return add_block_common(b, series_id, ts, versions, vals, 0,
+                            BYDB_VT_INT64, n, group_code, valid);
 }
 
 extern "C" int bydb_part_builder_add_block_i64(bydb_part_builder *b,

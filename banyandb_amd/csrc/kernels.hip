@@ -50,6 +50,8 @@ bool bydb_normalize_plain_tag(const uint8_t *src, uint64_t src_len,
                               uint64_t nrows, std::vector<uint8_t> &out);
 bool bydb_normalize_dict_tag(const uint8_t *src, uint64_t src_len,
                              bool *needed, std::vector<uint8_t> &out);
+bool bydb_normalize_plain_field(const uint8_t *src, uint64_t src_len,
+                                uint64_t nrows, std::vector<uint8_t> &out);
 
 // ---------------- device helpers ----------------
 
@@ -1980,6 +1982,52 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
                 bmax = wave_reduce_max(lmx);
                 have_minmax = true;
             }
+        } else if (fenc == BYDB_ENC_PLAIN) {
+            // null-bearing Plain column, host-normalized into the sidecar
+            // ([u32 n][pad][validity bitmap][8-B sign-flip cells]): fold
+            // valid selected rows, decoding cells in-lane
+            // (convert/number.go:95-108); nulls drop from count and every
+            // aggregate (aggregation.go:310).
+            if (!(bd->field_off & TAG_SIDECAR_BIT)) {
+                dev_set_err(derr, DERR_BAD_ENC, (uint64_t)bi);
+                continue;
+            }
+            const uint8_t *fp = sidecar + (bd->field_off & ~TAG_SIDECAR_BIT);
+            uint32_t nn = (uint32_t)fp[0] | ((uint32_t)fp[1] << 8) |
+                          ((uint32_t)fp[2] << 16) | ((uint32_t)fp[3] << 24);
+            if (nn != (uint32_t)n) {
+                dev_set_err(derr, DERR_BAD_ENC, (uint64_t)bi);
+                continue;
+            }
+            const uint64_t *vbm = (const uint64_t *)(fp + 8);
+            const uint64_t *cells = vbm + ((uint64_t)nn + 63) / 64;
+            uint64_t lsum = 0, lcnt = 0;
+            int64_t lmn = INT64_MAX, lmx = INT64_MIN;
+            for (int64_t base = r0; base <= r1; base += WAVE) {
+                int64_t row = base + lane;
+                bool ok = row <= r1 &&
+                          ((vbm[row >> 6] >> (row & 63)) & 1);
+                if (wp0) ok = pred_match_rows(wp0, row, ok) && ok;
+                if (wp1) ok = pred_match_rows(wp1, row, ok) && ok;
+                if (wp2) ok = pred_match_rows(wp2, row, ok) && ok;
+                if (ok) {
+                    uint64_t c = cells[row];
+                    // cells are big-endian on the wire
+                    uint64_t u = __builtin_bswap64(c);
+                    int64_t v = (u >> 63) ? (int64_t)(u & ~(1ull << 63))
+                                          : -(int64_t)((1ull << 63) - u);
+                    lsum += (uint64_t)v;
+                    lcnt++;
+                    lmn = v < lmn ? v : lmn;
+                    lmx = v > lmx ? v : lmx;
+                }
+            }
+            bsum = wave_reduce_add(lsum);
+            nsel_eff = wave_reduce_add(lcnt);
+            bmin = wave_reduce_min(lmn);
+            bmax = wave_reduce_max(lmx);
+            have_minmax = nsel_eff > 0;
+            bsum = lane == 0 ? bsum : 0;
         } else {
             dev_set_err(derr, DERR_BAD_ENC, (uint64_t)bi);
             continue;
@@ -2241,6 +2289,29 @@ extern "C" int bydb_part_append(bydb_session *s, const uint8_t *payload,
             if (sl == 0) { fd.tag_off = noff; fd.tag_len = norm.size(); }
             else if (sl == 1) { fd.tag2_off = noff; fd.tag2_len = norm.size(); }
             else { fd.tag3_off = noff; fd.tag3_len = norm.size(); }
+            extra.insert(extra.end(), norm.begin(), norm.end());
+        }
+        // null-bearing Plain FIELD columns (8-B sign-flip cell blocks):
+        // normalize into the sidecar as bitmap + fixed cells so the fold
+        // can skip nulls per row (aggregation.go:310 null check)
+        if (bd->field_enc == BYDB_ENC_PLAIN &&
+            !(bd->field_off & TAG_SIDECAR_BIT)) {
+            if (bd->field_vtype != BYDB_VT_INT64) {
+                s->err = "nullable float64 columns not supported in v1";
+                return BYDB_ERR_BAD_ARG;
+            }
+            const uint8_t *fsrc = payload + (bd->field_off - s->payload_len);
+            std::vector<uint8_t> norm;
+            if (!bydb_normalize_plain_field(fsrc, bd->field_len, bd->count,
+                                            norm)) {
+                s->err = "plain field column normalization failed";
+                return BYDB_ERR_BAD_DATA;
+            }
+            if (fixed.empty()) fixed.assign(blocks, blocks + n_blocks);
+            while ((s->sidecar_len + extra.size()) & 7) extra.push_back(0);
+            bydb_block_desc &fd = fixed[(size_t)i];
+            fd.field_off = TAG_SIDECAR_BIT | (s->sidecar_len + extra.size());
+            fd.field_len = norm.size();
             extra.insert(extra.end(), norm.begin(), norm.end());
         }
     }

@@ -219,6 +219,14 @@ int bydb_part_builder_add_block_f64(bydb_part_builder *b, uint64_t series_id,
                                     const int64_t *ts, const int64_t *versions,
                                     const double *vals, int64_t n,
                                     uint32_t group_code);
+/* null-bearing int64 column (valid[i]==0 -> row i null): stored as the
+ * reference's Plain fallback — a bytes block of 8-B sign-flip cells with
+ * nil rows zero-length (column.go:214-233, convert/number.go:33-46);
+ * the fold skips nulls (vectorized/measure/aggregation.go:310) */
+int bydb_part_builder_add_block_i64_nullable(
+    bydb_part_builder *b, uint64_t series_id, const int64_t *ts,
+    const int64_t *versions, const int64_t *vals, const uint8_t *valid,
+    int64_t n, uint32_t group_code);
 /* attach a dictionary-encoded tag column to the block just added:
  * tag value of row i = tag_values[codes[i]] (code order = first-seen) */
 int bydb_part_builder_set_block_tag(bydb_part_builder *b, const uint8_t *data,

@@ -764,6 +764,33 @@ int bo_bytes_block_encode(uint8_t *dst, size_t cap, const uint8_t *data,
     return BO_OK;
 }
 
+/* Plain (null-bearing) int64 column: a bytes block of 8-B sign-flip
+ * cells (convert/number.go:33-46), nil rows zero-length
+ * (column.go:214-233 write fallback).  Fills dst and valid; the fold
+ * must skip invalid rows (aggregation.go:310 null check). */
+static int column_i64_decode_plain(int64_t *dst, uint8_t *valid,
+                                   const uint8_t *src, size_t len,
+                                   int64_t n) {
+    if (len < 1 || src[0] != BO_ENC_PLAIN) return BO_ERR_BAD_TYPE;
+    size_t cap = (size_t)n * 8 + 16;
+    uint8_t *data = (uint8_t *)malloc(cap);
+    int64_t *lens = (int64_t *)malloc(sizeof(int64_t) * (size_t)n);
+    size_t dl = 0;
+    int rc = bo_bytes_block_decode(data, cap, lens, src + 1, len - 1, n, &dl);
+    if (rc == BO_OK) {
+        size_t off = 0;
+        for (int64_t i = 0; i < n; i++) {
+            if (lens[i] < 0) { valid[i] = 0; dst[i] = 0; continue; }
+            if (lens[i] != 8) { rc = BO_ERR_BAD_DATA; break; }
+            valid[i] = 1;
+            dst[i] = bo_cell_bytes_to_i64(data + off);
+            off += 8;
+        }
+    }
+    free(data); free(lens);
+    return rc;
+}
+
 /* BytesBlockDecoder.Decode — bytes.go:84-130 */
 int bo_bytes_block_decode(uint8_t *data_out, size_t data_cap, int64_t *lens_out,
                           const uint8_t *src, size_t src_len, int64_t n,
@@ -1137,8 +1164,18 @@ static int scan_block(const uint8_t *payload, const bo_block_desc *b,
     }
     (void)have_pred;
     if (field_vtype == BO_VT_INT64) {
-        rc = bo_column_i64_decode(i64_buf, payload + b->col_off, b->col_len, n);
-        if (rc != BO_OK) return rc;
+        if (payload[b->col_off] == BO_ENC_PLAIN) {
+            uint8_t valid[8192];
+            rc = column_i64_decode_plain(i64_buf, valid,
+                                         payload + b->col_off, b->col_len, n);
+            if (rc != BO_OK) return rc;
+            for (int64_t i = 0; i < n; i++)
+                if (!valid[i]) rowmatch_buf[i] = 0;
+        } else {
+            rc = bo_column_i64_decode(i64_buf, payload + b->col_off,
+                                      b->col_len, n);
+            if (rc != BO_OK) return rc;
+        }
         for (int64_t i = r0; i <= r1; i++) {
             if (!rowmatch_buf[i]) continue;
             int64_t v = i64_buf[i];
@@ -1276,9 +1313,20 @@ int bo_scan_agg_bytags(const uint8_t *payload, const bo_block_desc *blocks,
         }
         if (rc != BO_OK) break;
         if (dropped) continue;
-        if (field_vtype == BO_VT_INT64)
-            rc = bo_column_i64_decode(i64_buf, payload + b->col_off, b->col_len, n);
-        else if (field_vtype == BO_VT_FLOAT64)
+        if (field_vtype == BO_VT_INT64) {
+            if (payload[b->col_off] == BO_ENC_PLAIN) {
+                uint8_t validp[8192];
+                rc = column_i64_decode_plain(i64_buf, validp,
+                                             payload + b->col_off,
+                                             b->col_len, n);
+                if (rc == BO_OK)
+                    for (int64_t q = 0; q < n; q++)
+                        if (!validp[q]) rowgid[q] = -1;
+            } else {
+                rc = bo_column_i64_decode(i64_buf, payload + b->col_off,
+                                          b->col_len, n);
+            }
+        } else if (field_vtype == BO_VT_FLOAT64)
             rc = bo_column_f64_decode(f64_buf, payload + b->col_off, b->col_len, n);
         else rc = BO_ERR_BAD_TYPE;
         if (rc != BO_OK) break;

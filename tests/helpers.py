@@ -13,7 +13,10 @@ def oracle_blocks(builder):
     blocks = []
     for d in builder.blocks():
         stream = bytes(payload[d.field_off: d.field_off + d.field_len])
-        if d.field_vtype == 3:  # float64: [type][exp BE][first cell] + stream
+        if d.field_enc == 9:    # Plain (null-bearing): the stream already
+            col = stream        # carries [type][bytes block], no firstValue
+                                # (column.go:266-278 encodeDefault)
+        elif d.field_vtype == 3:  # float64: [type][exp BE][first cell] + stream
             col = bytes([d.field_enc]) + int(d.exp).to_bytes(2, "big", signed=True) \
                 + o.cell_encode(d.field_first) + stream
         else:                   # int64: [type][first cell] + stream

@@ -30,17 +30,30 @@ def _upload_raw(payload: bytes, descs):
 
 
 def test_unsupported_encode_type_fails_loud():
+    """Unknown encode byte reaches the device and is flagged at finalize."""
     b = PartBuilder()
     ts = [T0 + i * MS for i in range(64)]
     b.add_block_i64(1, ts, [1] * 64, list(range(0, 6400, 100)))
     d = _desc_copy(b.blocks()[0])
-    d.field_enc = 9  # Plain — numeric fold does not support it
+    d.field_enc = 77  # not an encode type
     s = _upload_raw(b.payload, [d])
     s.configure(VT_INT64, [AGG_SUM, AGG_COUNT])
     s.consume()
     with pytest.raises(RuntimeError, match="decode error"):
         s.finalize()
     s.close()
+
+
+def test_fake_plain_field_fails_loud_at_upload():
+    """A descriptor claiming Plain over a non-Plain stream is rejected by
+    the host normalization at part registration."""
+    b = PartBuilder()
+    ts = [T0 + i * MS for i in range(64)]
+    b.add_block_i64(1, ts, [1] * 64, list(range(0, 6400, 100)))
+    d = _desc_copy(b.blocks()[0])
+    d.field_enc = 9  # Plain claimed, stream is a delta varint stream
+    with pytest.raises(RuntimeError, match="normalization failed"):
+        _upload_raw(b.payload, [d])
 
 
 def test_descending_timestamps_fail_loud():
@@ -82,7 +95,7 @@ def test_sticky_error_then_reset_recovers():
     ts = [T0 + i * MS for i in range(64)]
     b.add_block_i64(1, ts, [1] * 64, [7] * 64)
     d_bad = _desc_copy(b.blocks()[0])
-    d_bad.field_enc = 9
+    d_bad.field_enc = 77
     s = _upload_raw(b.payload, [d_bad])
     s.configure(VT_INT64, [AGG_SUM, AGG_COUNT])
     s.consume()
