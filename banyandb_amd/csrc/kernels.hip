@@ -1052,21 +1052,95 @@ __device__ void scan_stream(const uint8_t *stream, int64_t n_deltas, bool dod,
     uint64_t d1_carry = (uint64_t)d1_init;
     uint64_t l_sum = 0, l_nsel = 0, l_nlo = 0, l_nhi = 0;
     int64_t l_mn = INT64_MAX, l_mx = INT64_MIN;
-    // Fixed 64-B window advance: every window consumes exactly 64 bytes
-    // and a varint straddling the boundary is carried in registers
-    // (carry_u holds its low groups, carry_n its byte count).  This makes
-    // the window address an induction variable, so the next window's load
-    // (b_nxt) issues a full window of decode ahead of its s_waitcnt —
-    // without it the scan is latency-bound (63% of wave cycles parked on
-    // the serial pos -> load -> ballot -> pos chain).  Terminators past
+    // Fixed-size window advance with a cross-window varint carry
+    // (carry_u holds the carried low groups, carry_n its byte count), so
+    // window addresses are induction variables.  The primary window is
+    // 128 B with TWO bytes per lane (one u16 load): when every varint is
+    // 1-2 bytes — the dominant telemetry shape — each lane decodes up to
+    // two values and one 32-bit scan covers 128 B, halving per-byte
+    // instruction and load counts vs the 64-B window.  Windows that fail
+    // the pattern (longer varints, walkers, the stream tail, a multi-byte
+    // carry) fall back to the 64-B machinery below.  Terminators past
     // jmax (bytes of the following stream / the part's 1-KiB slack) fold
     // as zeros, so the wave-wide scans read their totals at lane 63.
     uint64_t carry_u = 0;
     uint32_t carry_n = 0;
-    uint32_t b_cur = stream[lane];
+    const bool no_walk = pw0 == nullptr && pw1 == nullptr && pw2 == nullptr;
     while (j <= jmax) {
-        uint32_t b_nxt = stream[pos + 64 + (uint64_t)lane];  // prefetch
-        uint8_t b = (uint8_t)b_cur;
+        if (!dod && no_walk && carry_n <= 1) {
+            uint16_t pr;
+            __builtin_memcpy(&pr, stream + pos + 2 * (uint64_t)lane, 2);
+            uint32_t b0 = pr & 0xffu, b1 = (uint32_t)pr >> 8;
+            uint64_t e0 = __ballot(b0 < 0x80);
+            uint64_t e1 = __ballot(b1 < 0x80);
+            uint64_t c0 = ~e0, c1 = ~e1;
+            int nt = __popcll(e0) + __popcll(e1);
+            if ((c0 & c1) == 0 && (c1 & (c0 >> 1)) == 0 &&
+                (int64_t)nt <= jmax - j + 1 &&
+                (carry_n == 0 || (e0 & 1))) {
+                // the previous lane's b1 heads a 2-byte value ending at my
+                // b0; row heads patch via v_readlane, lane 0 via the carry
+                uint32_t pb1 = dpp_mov32<0x111>(b1);
+                uint32_t q15 = (uint32_t)__builtin_amdgcn_readlane((int)b1, 15);
+                uint32_t q31 = (uint32_t)__builtin_amdgcn_readlane((int)b1, 31);
+                uint32_t q47 = (uint32_t)__builtin_amdgcn_readlane((int)b1, 47);
+                if ((lane & 15) == 0 && lane)
+                    pb1 = lane == 16 ? q15 : lane == 32 ? q31 : q47;
+                if (lane == 0)
+                    pb1 = carry_n ? ((uint32_t)carry_u | 0x80u) : 0;
+                bool t0 = b0 < 0x80, t1 = b1 < 0x80;
+                uint32_t u0 = (pb1 & 0x80u) ? ((pb1 & 0x7fu) | (b0 << 7)) : b0;
+                uint32_t u1 = (b0 & 0x80u) ? ((b0 & 0x7fu) | (b1 << 7)) : b1;
+                int32_t dA = t0 ? ((int32_t)(u0 >> 1) ^ -(int32_t)(u0 & 1)) : 0;
+                int32_t dB = t1 ? ((int32_t)(u1 >> 1) ^ -(int32_t)(u1 & 1)) : 0;
+                int32_t sl = dA + dB;
+                int32_t S = wave_incl_scan32(sl, lane);
+                int32_t pre = S - sl;
+                int rb = __popcll(e0 & lanemask_lt(lane)) +
+                         __popcll(e1 & lanemask_lt(lane));
+                if (t0) {
+                    int64_t idx = j + rb;
+                    int64_t sv =
+                        (int64_t)(v_carry + (uint64_t)(int64_t)(pre + dA));
+                    if (idx >= r0 && idx <= r1) {
+                        l_sum += (uint64_t)sv;
+                        l_nsel++;
+                        l_mn = sv < l_mn ? sv : l_mn;
+                        l_mx = sv > l_mx ? sv : l_mx;
+                    }
+                    if (sv < lo_bound) l_nlo++;
+                    if (sv > hi_bound) l_nhi++;
+                }
+                if (t1) {
+                    int64_t idx = j + rb + (t0 ? 1 : 0);
+                    int64_t sv = (int64_t)(v_carry +
+                                           (uint64_t)(int64_t)(pre + dA + dB));
+                    if (idx >= r0 && idx <= r1) {
+                        l_sum += (uint64_t)sv;
+                        l_nsel++;
+                        l_mn = sv < l_mn ? sv : l_mn;
+                        l_mx = sv > l_mx ? sv : l_mx;
+                    }
+                    if (sv < lo_bound) l_nlo++;
+                    if (sv > hi_bound) l_nhi++;
+                }
+                v_carry +=
+                    (uint64_t)(int64_t)__builtin_amdgcn_readlane(S, 63);
+                if (e1 >> 63) {
+                    carry_n = 0;
+                    carry_u = 0;
+                } else {
+                    carry_n = 1;
+                    carry_u = (uint64_t)(
+                        (uint32_t)__builtin_amdgcn_readlane((int)b1, 63) &
+                        0x7f);
+                }
+                j += nt;
+                pos += 128;
+                continue;
+            }
+        }
+        uint8_t b = stream[pos + (uint64_t)lane];
         uint64_t emask = __ballot(b < 0x80);
         if (emask == 0) { dev_set_err(derr, DERR_BAD_STREAM, bi); break; }
         int rank = __popcll(emask & lanemask_lt(lane));
@@ -1127,7 +1201,6 @@ __device__ void scan_stream(const uint8_t *stream, int64_t n_deltas, bool dod,
             }
             j += nterm;
             pos += 64;
-            b_cur = b_nxt;
             continue;
         }
         uint64_t d = 0;
@@ -1215,7 +1288,6 @@ __device__ void scan_stream(const uint8_t *stream, int64_t n_deltas, bool dod,
         }
         j += nterm;
         pos += 64;
-        b_cur = b_nxt;
     }
     f->sum = l_sum;
     f->mn = l_mn;
