@@ -1066,10 +1066,18 @@ __device__ void scan_stream(const uint8_t *stream, int64_t n_deltas, bool dod,
     uint64_t carry_u = 0;
     uint32_t carry_n = 0;
     const bool no_walk = pw0 == nullptr && pw1 == nullptr && pw2 == nullptr;
+    uint16_t pr_cur = 0;
+    bool pr_valid = false;
     while (j <= jmax) {
         if (!dod && no_walk && carry_n <= 1) {
-            uint16_t pr;
-            __builtin_memcpy(&pr, stream + pos + 2 * (uint64_t)lane, 2);
+            if (!pr_valid)
+                __builtin_memcpy(&pr_cur, stream + pos + 2 * (uint64_t)lane,
+                                 2);
+            uint16_t pr_nxt;   // issue the next window's load a full
+                               // window of decode early
+            __builtin_memcpy(&pr_nxt, stream + pos + 128 + 2 * (uint64_t)lane,
+                             2);
+            uint16_t pr = pr_cur;
             uint32_t b0 = pr & 0xffu, b1 = (uint32_t)pr >> 8;
             uint64_t e0 = __ballot(b0 < 0x80);
             uint64_t e1 = __ballot(b1 < 0x80);
@@ -1137,8 +1145,13 @@ __device__ void scan_stream(const uint8_t *stream, int64_t n_deltas, bool dod,
                 }
                 j += nt;
                 pos += 128;
+                pr_cur = pr_nxt;
+                pr_valid = true;
                 continue;
             }
+            pr_valid = false;
+        } else {
+            pr_valid = false;
         }
         uint8_t b = stream[pos + (uint64_t)lane];
         uint64_t emask = __ballot(b < 0x80);
