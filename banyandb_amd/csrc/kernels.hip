@@ -1615,11 +1615,22 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
     for (int64_t wi = wave_id; wi < n_items; wi += n_waves) {
         const int64_t bi = segs ? wi / MAX_SEGS : wi;
         const int seg = segs ? (int)(wi % MAX_SEGS) : 0;
-        const bool seg_eligible =
-            segs && segs[bi * MAX_SEGS].byte_off != SEG_INELIGIBLE;
-        if (seg != 0 && !seg_eligible) continue;
-        // wave-uniform descriptor loads
+        // per-item cold-start: issue the independent scalar loads
+        // (descriptor, predicate flag, segment entries) back-to-back so
+        // the item pays ONE memory latency, not a chain of them
         const bydb_block_desc *bd = &blocks[bi];
+        const uint8_t pf_pre =
+            (EN_PREDS && preds != nullptr) ? pred_flags[bi] : PF_CLEAR;
+        uint32_t seg0_off = SEG_INELIGIBLE;
+        SegEntry e_pre;
+        e_pre.byte_off = 0; e_pre._pad = 0; e_pre.v_start = 0;
+        e_pre.d1_start = 0;
+        if (segs) {
+            seg0_off = segs[bi * MAX_SEGS].byte_off;
+            e_pre = segs[bi * MAX_SEGS + seg];
+        }
+        const bool seg_eligible = segs && seg0_off != SEG_INELIGIBLE;
+        if (seg != 0 && !seg_eligible) continue;
         const int64_t n = (int64_t)bd->count;
         if (seg != 0 && (int64_t)seg * SEG_ROWS + 1 > n - 1) continue;
         const int64_t ts_min = bd->ts_min, ts_max = bd->ts_max;
@@ -1683,7 +1694,7 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
         PredWalk pw0, pw1, pw2;
         PredWalk *wp0 = nullptr, *wp1 = nullptr, *wp2 = nullptr;
         if (preds != nullptr) {
-            const uint8_t pf = pred_flags[bi];
+            const uint8_t pf = pf_pre;
             if (pf == PF_SKIP) continue;
             if (pf == PF_ERR) {
                 dev_set_err(derr, DERR_BAD_ENC, (uint64_t)bi);
@@ -1963,6 +1974,7 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
                     }
                 }
             } else if (use_seg) {
+                // (SegEntry preloaded at item start)
                 // segment-parallel value scan: this wave folds only rows
                 // [seg*SEG_ROWS+1 .. min((seg+1)*SEG_ROWS, n-1)] (+ rows 0
                 // and, for delta-of-delta, 1 on segment 0), starting from
@@ -1971,7 +1983,7 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
                 // absolute row (k*SEG_ROWS - 1 + j'), so j' = 2 lands on
                 // the segment's first row and e.v_start/e.d1_start are the
                 // carries at row k*SEG_ROWS.
-                const SegEntry e = segs[bi * MAX_SEGS + seg];
+                const SegEntry e = e_pre;
                 int64_t sj_lo = (int64_t)seg * SEG_ROWS + 1;
                 int64_t sj_hi = ((int64_t)seg + 1) * SEG_ROWS;
                 if (sj_hi > n - 1) sj_hi = n - 1;
