@@ -106,28 +106,65 @@ __device__ __forceinline__ uint64_t wave_incl_scan(uint64_t v, int lane) {
     return v + add;
 }
 
+// reductions, DPP-style like the scans: 4 row_shr steps leave each
+// 16-lane row's total at its last lane; v_readlane combines the rows.
+// All lanes return the wave total (as the shfl_xor butterfly did).
 __device__ __forceinline__ uint64_t wave_reduce_add(uint64_t v) {
-#pragma unroll
-    for (int off = 32; off > 0; off >>= 1) v += (uint64_t)__shfl_xor((long long)v, off);
-    return v;
+#define BYDB_RSTEP64(C)                                                      \
+    {                                                                        \
+        uint64_t t = ((uint64_t)dpp_mov32<C>((uint32_t)(v >> 32)) << 32) |   \
+                     dpp_mov32<C>((uint32_t)v);                              \
+        v += t;                                                              \
+    }
+    BYDB_RSTEP64(0x111) BYDB_RSTEP64(0x112)
+    BYDB_RSTEP64(0x114) BYDB_RSTEP64(0x118)
+#undef BYDB_RSTEP64
+    return readlane64(v, 15) + readlane64(v, 31) + readlane64(v, 47) +
+           readlane64(v, 63);
 }
 
 __device__ __forceinline__ int64_t wave_reduce_min(int64_t v) {
-#pragma unroll
-    for (int off = 32; off > 0; off >>= 1) {
-        int64_t t = (int64_t)__shfl_xor((long long)v, off);
-        v = t < v ? t : v;
+    // row_shr moves are zero-filled out of row; feed the identity by
+    // selecting against the shifted VALIDITY instead: use bound_ctrl 0's
+    // zero fill and compare via a mask-free trick — shift lanes carry a
+    // zero, so compare only where the source lane exists (lane%16 >= off)
+    const int lane15 = (int)(threadIdx.x & 15);
+#define BYDB_RMIN(C, OFF)                                                    \
+    {                                                                        \
+        int64_t t = (int64_t)(((uint64_t)dpp_mov32<C>((uint32_t)((uint64_t)v >> 32)) << 32) | \
+                              dpp_mov32<C>((uint32_t)(uint64_t)v));          \
+        if (lane15 >= OFF && t < v) v = t;                                   \
     }
-    return v;
+    BYDB_RMIN(0x111, 1) BYDB_RMIN(0x112, 2)
+    BYDB_RMIN(0x114, 4) BYDB_RMIN(0x118, 8)
+#undef BYDB_RMIN
+    int64_t a = (int64_t)readlane64((uint64_t)v, 15);
+    int64_t b = (int64_t)readlane64((uint64_t)v, 31);
+    int64_t c = (int64_t)readlane64((uint64_t)v, 47);
+    int64_t d = (int64_t)readlane64((uint64_t)v, 63);
+    int64_t m = a < b ? a : b;
+    m = c < m ? c : m;
+    return d < m ? d : m;
 }
 
 __device__ __forceinline__ int64_t wave_reduce_max(int64_t v) {
-#pragma unroll
-    for (int off = 32; off > 0; off >>= 1) {
-        int64_t t = (int64_t)__shfl_xor((long long)v, off);
-        v = t > v ? t : v;
+    const int lane15 = (int)(threadIdx.x & 15);
+#define BYDB_RMAX(C, OFF)                                                    \
+    {                                                                        \
+        int64_t t = (int64_t)(((uint64_t)dpp_mov32<C>((uint32_t)((uint64_t)v >> 32)) << 32) | \
+                              dpp_mov32<C>((uint32_t)(uint64_t)v));          \
+        if (lane15 >= OFF && t > v) v = t;                                   \
     }
-    return v;
+    BYDB_RMAX(0x111, 1) BYDB_RMAX(0x112, 2)
+    BYDB_RMAX(0x114, 4) BYDB_RMAX(0x118, 8)
+#undef BYDB_RMAX
+    int64_t a = (int64_t)readlane64((uint64_t)v, 15);
+    int64_t b = (int64_t)readlane64((uint64_t)v, 31);
+    int64_t c = (int64_t)readlane64((uint64_t)v, 47);
+    int64_t d = (int64_t)readlane64((uint64_t)v, 63);
+    int64_t m = a > b ? a : b;
+    m = c > m ? c : m;
+    return d > m ? d : m;
 }
 
 // Decode one varint serially starting at p (used for wave-uniform headers:
