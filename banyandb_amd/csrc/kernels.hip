@@ -1103,16 +1103,18 @@ __device__ void scan_stream(const uint8_t *stream, int64_t n_deltas, bool dod,
     uint64_t carry_u = 0;
     uint32_t carry_n = 0;
     const bool no_walk = pw0 == nullptr && pw1 == nullptr && pw2 == nullptr;
-    uint16_t pr_cur = 0;
-    bool pr_valid = false;
+    uint16_t pr_cur = 0, pr_nx1 = 0;
+    int pr_valid = 0;   // how many upcoming windows are already loaded
     while (j <= jmax) {
         if (!dod && no_walk && carry_n <= 1) {
-            if (!pr_valid)
+            if (pr_valid < 1)
                 __builtin_memcpy(&pr_cur, stream + pos + 2 * (uint64_t)lane,
                                  2);
-            uint16_t pr_nxt;   // issue the next window's load a full
-                               // window of decode early
-            __builtin_memcpy(&pr_nxt, stream + pos + 128 + 2 * (uint64_t)lane,
+            if (pr_valid < 2)
+                __builtin_memcpy(&pr_nx1,
+                                 stream + pos + 128 + 2 * (uint64_t)lane, 2);
+            uint16_t pr_nxt;   // two windows of load lead
+            __builtin_memcpy(&pr_nxt, stream + pos + 256 + 2 * (uint64_t)lane,
                              2);
             uint16_t pr = pr_cur;
             uint32_t b0 = pr & 0xffu, b1 = (uint32_t)pr >> 8;
@@ -1182,13 +1184,14 @@ __device__ void scan_stream(const uint8_t *stream, int64_t n_deltas, bool dod,
                 }
                 j += nt;
                 pos += 128;
-                pr_cur = pr_nxt;
-                pr_valid = true;
+                pr_cur = pr_nx1;
+                pr_nx1 = pr_nxt;
+                pr_valid = 2;
                 continue;
             }
-            pr_valid = false;
+            pr_valid = 0;
         } else {
-            pr_valid = false;
+            pr_valid = 0;
         }
         uint8_t b = stream[pos + (uint64_t)lane];
         uint64_t emask = __ballot(b < 0x80);
