@@ -1413,7 +1413,100 @@ __global__ __launch_bounds__(256) void k_build_seg_index(
         uint64_t pos = 0;
         int64_t j = 1;
         uint64_t v_carry = (uint64_t)bd->field_first + (dod ? d1_carry : 0);
+        // delta streams walk 128-B paired-lane windows (the scan_stream
+        // fast shape) and record boundaries per value; non-qualifying
+        // windows rewind any carried byte and use the 64-B machinery
+        uint64_t carry_u2 = 0;
+        uint32_t carry_n2 = 0;
         while (j <= n_deltas) {
+            if (!dod && carry_n2 <= 1) {
+                uint16_t pr;
+                __builtin_memcpy(&pr, d2s + pos + 2 * (uint64_t)lane, 2);
+                uint32_t b0 = pr & 0xffu, b1 = (uint32_t)pr >> 8;
+                uint64_t e0 = __ballot(b0 < 0x80);
+                uint64_t e1 = __ballot(b1 < 0x80);
+                uint64_t c0 = ~e0, c1 = ~e1;
+                int nt = __popcll(e0) + __popcll(e1);
+                if ((c0 & c1) == 0 && (c1 & (c0 >> 1)) == 0 &&
+                    (int64_t)nt <= n_deltas - j + 1 &&
+                    (carry_n2 == 0 || (e0 & 1))) {
+                    uint32_t pb1 = dpp_mov32<0x111>(b1);
+                    uint32_t q15 =
+                        (uint32_t)__builtin_amdgcn_readlane((int)b1, 15);
+                    uint32_t q31 =
+                        (uint32_t)__builtin_amdgcn_readlane((int)b1, 31);
+                    uint32_t q47 =
+                        (uint32_t)__builtin_amdgcn_readlane((int)b1, 47);
+                    if ((lane & 15) == 0 && lane)
+                        pb1 = lane == 16 ? q15 : lane == 32 ? q31 : q47;
+                    if (lane == 0)
+                        pb1 = carry_n2 ? ((uint32_t)carry_u2 | 0x80u) : 0;
+                    bool t0 = b0 < 0x80, t1 = b1 < 0x80;
+                    uint32_t u0 =
+                        (pb1 & 0x80u) ? ((pb1 & 0x7fu) | (b0 << 7)) : b0;
+                    uint32_t u1 =
+                        (b0 & 0x80u) ? ((b0 & 0x7fu) | (b1 << 7)) : b1;
+                    int32_t dA =
+                        t0 ? ((int32_t)(u0 >> 1) ^ -(int32_t)(u0 & 1)) : 0;
+                    int32_t dB =
+                        t1 ? ((int32_t)(u1 >> 1) ^ -(int32_t)(u1 & 1)) : 0;
+                    int32_t sl = dA + dB;
+                    int32_t S = wave_incl_scan32(sl, lane);
+                    int32_t pre = S - sl;
+                    int rb = __popcll(e0 & lanemask_lt(lane)) +
+                             __popcll(e1 & lanemask_lt(lane));
+                    if (t0) {
+                        int64_t row = j + rb;
+                        if ((row % SEG_ROWS) == 0 &&
+                            row / SEG_ROWS < MAX_SEGS) {
+                            SegEntry E;
+                            E.byte_off =
+                                (uint32_t)(d2_off + pos + 2 * (uint64_t)lane +
+                                           1);
+                            E._pad = 0;
+                            E.v_start = (int64_t)(v_carry +
+                                                  (uint64_t)(int64_t)(pre + dA));
+                            E.d1_start = 0;
+                            bseg[row / SEG_ROWS] = E;
+                        }
+                    }
+                    if (t1) {
+                        int64_t row = j + rb + (t0 ? 1 : 0);
+                        if ((row % SEG_ROWS) == 0 &&
+                            row / SEG_ROWS < MAX_SEGS) {
+                            SegEntry E;
+                            E.byte_off =
+                                (uint32_t)(d2_off + pos + 2 * (uint64_t)lane +
+                                           2);
+                            E._pad = 0;
+                            E.v_start =
+                                (int64_t)(v_carry +
+                                          (uint64_t)(int64_t)(pre + dA + dB));
+                            E.d1_start = 0;
+                            bseg[row / SEG_ROWS] = E;
+                        }
+                    }
+                    v_carry +=
+                        (uint64_t)(int64_t)__builtin_amdgcn_readlane(S, 63);
+                    if (e1 >> 63) {
+                        carry_n2 = 0;
+                        carry_u2 = 0;
+                    } else {
+                        carry_n2 = 1;
+                        carry_u2 = (uint64_t)(
+                            (uint32_t)__builtin_amdgcn_readlane((int)b1, 63) &
+                            0x7f);
+                    }
+                    j += nt;
+                    pos += 128;
+                    continue;
+                }
+            }
+            if (carry_n2) {  // rewind the carried byte for the 64-B walk
+                pos -= carry_n2;
+                carry_n2 = 0;
+                carry_u2 = 0;
+            }
             uint8_t b = d2s[pos + (uint64_t)lane];
             uint64_t emask = __ballot(b < 0x80);
             if (emask == 0) { dev_set_err(derr, DERR_BAD_STREAM, (uint64_t)bi); break; }
