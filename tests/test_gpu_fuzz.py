@@ -142,7 +142,7 @@ def check_bytag(rng, b, is_float):
                 assert g.min_i == oc.min_i and g.max_i == oc.max_i
 
 
-@pytest.mark.parametrize("seed", range(12))
+@pytest.mark.parametrize("seed", range(20))
 def test_fuzz_scenario(seed):
     rng = random.Random(0xF0 + seed)
     b, is_float, tag_kind = build_scenario(rng)
@@ -151,7 +151,7 @@ def test_fuzz_scenario(seed):
         check_bytag(rng, b, is_float)
 
 
-@pytest.mark.parametrize("seed", range(6))
+@pytest.mark.parametrize("seed", range(10))
 def test_fuzz_grouped_by_code(seed):
     rng = random.Random(0x1F0 + seed)
     b, is_float, _ = build_scenario(rng, n_groups=3)
