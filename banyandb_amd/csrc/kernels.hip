@@ -1103,95 +1103,103 @@ __device__ void scan_stream(const uint8_t *stream, int64_t n_deltas, bool dod,
     uint64_t carry_u = 0;
     uint32_t carry_n = 0;
     const bool no_walk = pw0 == nullptr && pw1 == nullptr && pw2 == nullptr;
-    uint16_t pr_cur = 0, pr_nx1 = 0;
-    int pr_valid = 0;   // how many upcoming windows are already loaded
+    uint32_t w_cur = 0, w_nx1 = 0;
+    int w_valid = 0;   // how many upcoming quad windows are already loaded
     while (j <= jmax) {
         if (!dod && no_walk && carry_n <= 1) {
-            if (pr_valid < 1)
-                __builtin_memcpy(&pr_cur, stream + pos + 2 * (uint64_t)lane,
-                                 2);
-            if (pr_valid < 2)
-                __builtin_memcpy(&pr_nx1,
-                                 stream + pos + 128 + 2 * (uint64_t)lane, 2);
-            uint16_t pr_nxt;   // two windows of load lead
-            __builtin_memcpy(&pr_nxt, stream + pos + 256 + 2 * (uint64_t)lane,
-                             2);
-            uint16_t pr = pr_cur;
-            uint32_t b0 = pr & 0xffu, b1 = (uint32_t)pr >> 8;
+            // 256-B quad window: FOUR bytes per lane (one u32 load), each
+            // lane decoding up to four 1-2-byte varints; one 32-bit scan
+            // covers 256 B
+            if (w_valid < 1)
+                __builtin_memcpy(&w_cur, stream + pos + 4 * (uint64_t)lane,
+                                 4);
+            if (w_valid < 2)
+                __builtin_memcpy(&w_nx1,
+                                 stream + pos + 256 + 4 * (uint64_t)lane, 4);
+            uint32_t w_nxt;    // two windows of load lead
+            __builtin_memcpy(&w_nxt, stream + pos + 512 + 4 * (uint64_t)lane,
+                             4);
+            uint32_t b0 = w_cur & 0xffu, b1 = (w_cur >> 8) & 0xffu,
+                     b2 = (w_cur >> 16) & 0xffu, b3 = w_cur >> 24;
             uint64_t e0 = __ballot(b0 < 0x80);
             uint64_t e1 = __ballot(b1 < 0x80);
-            uint64_t c0 = ~e0, c1 = ~e1;
-            int nt = __popcll(e0) + __popcll(e1);
-            if ((c0 & c1) == 0 && (c1 & (c0 >> 1)) == 0 &&
+            uint64_t e2 = __ballot(b2 < 0x80);
+            uint64_t e3 = __ballot(b3 < 0x80);
+            uint64_t c0 = ~e0, c1 = ~e1, c2 = ~e2, c3 = ~e3;
+            int nt = __popcll(e0) + __popcll(e1) + __popcll(e2) +
+                     __popcll(e3);
+            if (((c0 & c1) | (c1 & c2) | (c2 & c3)) == 0 &&
+                (c3 & (c0 >> 1)) == 0 &&
                 (int64_t)nt <= jmax - j + 1 &&
                 (carry_n == 0 || (e0 & 1))) {
-                // the previous lane's b1 heads a 2-byte value ending at my
-                // b0; row heads patch via v_readlane, lane 0 via the carry
-                uint32_t pb1 = dpp_mov32<0x111>(b1);
-                uint32_t q15 = (uint32_t)__builtin_amdgcn_readlane((int)b1, 15);
-                uint32_t q31 = (uint32_t)__builtin_amdgcn_readlane((int)b1, 31);
-                uint32_t q47 = (uint32_t)__builtin_amdgcn_readlane((int)b1, 47);
+                // previous lane's b3 heads a value ending at my b0
+                uint32_t pb3 = dpp_mov32<0x111>(b3);
+                uint32_t q15 = (uint32_t)__builtin_amdgcn_readlane((int)b3, 15);
+                uint32_t q31 = (uint32_t)__builtin_amdgcn_readlane((int)b3, 31);
+                uint32_t q47 = (uint32_t)__builtin_amdgcn_readlane((int)b3, 47);
                 if ((lane & 15) == 0 && lane)
-                    pb1 = lane == 16 ? q15 : lane == 32 ? q31 : q47;
+                    pb3 = lane == 16 ? q15 : lane == 32 ? q31 : q47;
                 if (lane == 0)
-                    pb1 = carry_n ? ((uint32_t)carry_u | 0x80u) : 0;
-                bool t0 = b0 < 0x80, t1 = b1 < 0x80;
-                uint32_t u0 = (pb1 & 0x80u) ? ((pb1 & 0x7fu) | (b0 << 7)) : b0;
+                    pb3 = carry_n ? ((uint32_t)carry_u | 0x80u) : 0;
+                bool t0 = b0 < 0x80, t1 = b1 < 0x80, t2 = b2 < 0x80,
+                     t3 = b3 < 0x80;
+                uint32_t u0 = (pb3 & 0x80u) ? ((pb3 & 0x7fu) | (b0 << 7)) : b0;
                 uint32_t u1 = (b0 & 0x80u) ? ((b0 & 0x7fu) | (b1 << 7)) : b1;
+                uint32_t u2 = (b1 & 0x80u) ? ((b1 & 0x7fu) | (b2 << 7)) : b2;
+                uint32_t u3 = (b2 & 0x80u) ? ((b2 & 0x7fu) | (b3 << 7)) : b3;
                 int32_t dA = t0 ? ((int32_t)(u0 >> 1) ^ -(int32_t)(u0 & 1)) : 0;
                 int32_t dB = t1 ? ((int32_t)(u1 >> 1) ^ -(int32_t)(u1 & 1)) : 0;
-                int32_t sl = dA + dB;
+                int32_t dC = t2 ? ((int32_t)(u2 >> 1) ^ -(int32_t)(u2 & 1)) : 0;
+                int32_t dD = t3 ? ((int32_t)(u3 >> 1) ^ -(int32_t)(u3 & 1)) : 0;
+                int32_t sl = dA + dB + dC + dD;
                 int32_t S = wave_incl_scan32(sl, lane);
-                int32_t pre = S - sl;
+                int32_t cum = S - sl;
                 int rb = __popcll(e0 & lanemask_lt(lane)) +
-                         __popcll(e1 & lanemask_lt(lane));
-                if (t0) {
-                    int64_t idx = j + rb;
-                    int64_t sv =
-                        (int64_t)(v_carry + (uint64_t)(int64_t)(pre + dA));
-                    if (idx >= r0 && idx <= r1) {
-                        l_sum += (uint64_t)sv;
-                        l_nsel++;
-                        l_mn = sv < l_mn ? sv : l_mn;
-                        l_mx = sv > l_mx ? sv : l_mx;
-                    }
-                    if (sv < lo_bound) l_nlo++;
-                    if (sv > hi_bound) l_nhi++;
+                         __popcll(e1 & lanemask_lt(lane)) +
+                         __popcll(e2 & lanemask_lt(lane)) +
+                         __popcll(e3 & lanemask_lt(lane));
+                int64_t idx = j + rb;
+#define BYDB_QVAL(tk, dk)                                                    \
+                if (tk) {                                                    \
+                    cum += dk;                                               \
+                    int64_t sv =                                             \
+                        (int64_t)(v_carry + (uint64_t)(int64_t)cum);         \
+                    if (idx >= r0 && idx <= r1) {                            \
+                        l_sum += (uint64_t)sv;                               \
+                        l_nsel++;                                            \
+                        l_mn = sv < l_mn ? sv : l_mn;                        \
+                        l_mx = sv > l_mx ? sv : l_mx;                        \
+                    }                                                        \
+                    if (sv < lo_bound) l_nlo++;                              \
+                    if (sv > hi_bound) l_nhi++;                              \
+                    idx++;                                                   \
                 }
-                if (t1) {
-                    int64_t idx = j + rb + (t0 ? 1 : 0);
-                    int64_t sv = (int64_t)(v_carry +
-                                           (uint64_t)(int64_t)(pre + dA + dB));
-                    if (idx >= r0 && idx <= r1) {
-                        l_sum += (uint64_t)sv;
-                        l_nsel++;
-                        l_mn = sv < l_mn ? sv : l_mn;
-                        l_mx = sv > l_mx ? sv : l_mx;
-                    }
-                    if (sv < lo_bound) l_nlo++;
-                    if (sv > hi_bound) l_nhi++;
-                }
+                BYDB_QVAL(t0, dA)
+                BYDB_QVAL(t1, dB)
+                BYDB_QVAL(t2, dC)
+                BYDB_QVAL(t3, dD)
+#undef BYDB_QVAL
                 v_carry +=
                     (uint64_t)(int64_t)__builtin_amdgcn_readlane(S, 63);
-                if (e1 >> 63) {
+                if (e3 >> 63) {
                     carry_n = 0;
                     carry_u = 0;
                 } else {
                     carry_n = 1;
                     carry_u = (uint64_t)(
-                        (uint32_t)__builtin_amdgcn_readlane((int)b1, 63) &
+                        (uint32_t)__builtin_amdgcn_readlane((int)b3, 63) &
                         0x7f);
                 }
                 j += nt;
-                pos += 128;
-                pr_cur = pr_nx1;
-                pr_nx1 = pr_nxt;
-                pr_valid = 2;
+                pos += 256;
+                w_cur = w_nx1;
+                w_nx1 = w_nxt;
+                w_valid = 2;
                 continue;
             }
-            pr_valid = 0;
+            w_valid = 0;
         } else {
-            pr_valid = 0;
+            w_valid = 0;
         }
         uint8_t b = stream[pos + (uint64_t)lane];
         uint64_t emask = __ballot(b < 0x80);
