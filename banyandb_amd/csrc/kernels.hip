@@ -1368,8 +1368,8 @@ __device__ void scan_stream(const uint8_t *stream, int64_t n_deltas, bool dod,
 // parallel units and no cross-segment dependency.  The index is built
 // once per part upload + agg shape (like the reference's block metadata,
 // it is derived state over immutable part bytes).
-#define SEG_ROWS 1024
-#define MAX_SEGS 8
+#define SEG_ROWS 8192
+#define MAX_SEGS 1
 #define SEG_INELIGIBLE 0xFFFFFFFFu
 
 struct SegEntry {
