@@ -98,12 +98,13 @@ __device__ __forceinline__ uint64_t wave_incl_scan(uint64_t v, int lane) {
     uint64_t s15 = readlane64(v, 15);
     uint64_t s31 = readlane64(v, 31);
     uint64_t s47 = readlane64(v, 47);
+    // branchless: nested ternaries here compile to exec-mask branch
+    // cascades; mask-ANDs keep it pure VALU
     int r = lane >> 4;
-    uint64_t add = r == 0 ? 0
-                 : r == 1 ? s15
-                 : r == 2 ? s15 + s31
-                          : s15 + s31 + s47;
-    return v + add;
+    uint64_t m1 = (uint64_t)-(int64_t)(r >= 1);
+    uint64_t m2 = (uint64_t)-(int64_t)(r >= 2);
+    uint64_t m3 = (uint64_t)-(int64_t)(r >= 3);
+    return v + (s15 & m1) + (s31 & m2) + (s47 & m3);
 }
 
 // reductions, DPP-style like the scans: 4 row_shr steps leave each
@@ -312,12 +313,12 @@ __device__ __forceinline__ int32_t wave_incl_scan32(int32_t v, int lane) {
     int32_t s15 = __builtin_amdgcn_readlane(v, 15);
     int32_t s31 = __builtin_amdgcn_readlane(v, 31);
     int32_t s47 = __builtin_amdgcn_readlane(v, 47);
+    // branchless cross-row fixup (see the 64-bit variant)
     int r = lane >> 4;
-    int32_t add = r == 0 ? 0
-                : r == 1 ? s15
-                : r == 2 ? s15 + s31
-                         : s15 + s31 + s47;
-    return v + add;
+    int32_t m1 = -(int32_t)(r >= 1);
+    int32_t m2 = -(int32_t)(r >= 2);
+    int32_t m3 = -(int32_t)(r >= 3);
+    return v + (s15 & m1) + (s31 & m2) + (s47 & m3);
 }
 
 // Dense min/max over an all-1-byte delta stream: fold min/max of the
