@@ -200,6 +200,10 @@ enum {
     DERR_BAD_ENC = 2,       // unsupported encode type in kernel
     DERR_DESC_TS = 3,       // descending timestamps (reference never writes them)
     DERR_GROUP_RANGE = 4,   // block group_code >= configured n_groups
+    DERR_EXP_MISMATCH = 5,  // float64 block's decimal exponent differs from
+                            // the configured session exponent (per-block
+                            // rescale is a later row; silent mixing would
+                            // corrupt sums and make min/max incomparable)
 };
 
 __device__ __forceinline__ void dev_set_err(DevErr *e, unsigned code, uint64_t bi) {
@@ -1588,6 +1592,8 @@ __global__ __launch_bounds__(256) void k_build_seg_index(
 enum {
     KF_NEED_VALUES = 1,  // min/max requested -> reconstruct values
     KF_FLOAT = 2,        // float64 field: also accumulate mantissa sum as double
+                         // (bits 16-31 of flags carry the configured
+                         // decimal exponent for the per-block guard)
 };
 
 __device__ __forceinline__ void flush_partial(bydb_partial *partials,
@@ -1776,6 +1782,14 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
         const int64_t n = (int64_t)bd->count;
         if (seg != 0 && (int64_t)seg * SEG_ROWS + 1 > n - 1) continue;
         const int64_t ts_min = bd->ts_min, ts_max = bd->ts_max;
+        // float64 blocks fold in the decimal-int domain; a block whose
+        // exponent differs from the session's would mix incomparable
+        // mantissas — loud error (the oracle handles mixed exponents)
+        if ((flags & KF_FLOAT) &&
+            bd->exp != (int16_t)(uint16_t)((uint32_t)flags >> 16)) {
+            dev_set_err(derr, DERR_EXP_MISMATCH, (uint64_t)bi);
+            continue;
+        }
 
         // ---- row clamp (timestamp.FindRange, range.go:143-170) ----
         int64_t r0 = 0, r1 = n - 1;
@@ -2807,7 +2821,8 @@ extern "C" int bydb_consume_multi(bydb_session *s, int64_t min_ts,
     int flags = 0;
     if (s->func_mask & ((1u << BYDB_AGG_MIN) | (1u << BYDB_AGG_MAX)))
         flags |= KF_NEED_VALUES;
-    if (s->field_vtype == BYDB_VT_FLOAT64) flags |= KF_FLOAT;
+    if (s->field_vtype == BYDB_VT_FLOAT64)
+        flags |= KF_FLOAT | ((int)(uint16_t)s->float_exp << 16);
     const int threads = 256;                       // 4 waves per workgroup
     SegEntry *segs = nullptr;
     if (flags & KF_NEED_VALUES) {

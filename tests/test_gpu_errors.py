@@ -111,3 +111,25 @@ def test_sticky_error_then_reset_recovers():
     g = s.finalize()[0]
     assert g.count == 64 and g.sum_i == 7 * 64
     s.close()
+
+
+def test_float_exp_mismatch_fails_loud():
+    """A float64 block whose per-block decimal exponent differs from the
+    session's configured exponent must be a loud device error — mantissas
+    at different scales are incomparable (the oracle decodes each block
+    with its own exponent; per-block rescale on device is a later row)."""
+    from banyandb_amd import VT_FLOAT64
+    b = PartBuilder()
+    n = 128
+    ts = [T0 + i * MS for i in range(n)]
+    # every value has cents divisible by 10 -> block encodes at exp=-1
+    b.add_block_f64(1, ts, [1] * n, [i / 10.0 for i in range(1, n + 1)])
+    d = b.blocks()[0]
+    assert d.exp == -1
+    s = Session(0)
+    s.upload_part(b)
+    s.configure(VT_FLOAT64, [AGG_SUM, AGG_COUNT], float_exp=-2)
+    s.consume()
+    with pytest.raises(RuntimeError, match="decode error 5"):
+        s.finalize()
+    s.close()
