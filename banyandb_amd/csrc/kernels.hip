@@ -359,8 +359,9 @@ __device__ void dense_minmax(const uint8_t *stream, int64_t b_lo, int64_t b_hi,
             (const uint32_t *)((uintptr_t)(stream + b_lo) & ~(uintptr_t)3);
         int64_t wbase = (int64_t)((uintptr_t)wp - (uintptr_t)stream);
         uint64_t carry = (uint64_t)base_val;  // wrap-safe (Go int64 wraps)
+        uint32_t w = wp[lane];
         while (wbase < b_hi) {
-            uint32_t w = wp[lane];
+            uint32_t w_nxt = wp[64 + lane];   // next window, one ahead
             uint32_t sb = swar_zigzag(w);
             // mask bytes outside [b_lo, b_hi): zero their deltas and skip fold
             int64_t g0 = wbase + 4 * lane;
@@ -389,6 +390,7 @@ __device__ void dense_minmax(const uint8_t *stream, int64_t b_lo, int64_t b_hi,
             carry += (uint64_t)(int64_t)__builtin_amdgcn_readlane(incl, 63);
             wp += 64;
             wbase += 256;
+            w = w_nxt;
         }
     }
     *out_mn = wave_reduce_min(lmn);
