@@ -12,10 +12,12 @@
 //    configs is ~150 blocks per wave slot at full occupancy; no intra-block
 //    sync, no LDS;
 //  * the varint stream (zigzag, 7-bit groups LSB-first, terminator = high
-//    bit clear — pkg/encoding/int.go:81-148) is decoded 64 bytes per step:
-//    each lane loads one byte, a 64-bit __ballot of "high bit clear" marks
-//    value ends, each terminator lane reconstructs its value (1-byte fast
-//    path covers the dominant case);
+//    bit clear — pkg/encoding/int.go:81-148) decodes in 256-byte quad
+//    windows: four bytes per lane from one u32 load, __ballot terminator
+//    masks, up to four 1-2-byte values decoded per lane with one 32-bit
+//    DPP scan per window (shfl compiles to ds_bpermute — an LDS op — so
+//    the scans use DPP row_shr + v_readlane instead); wider varints fall
+//    back to a 64-byte ballot window;
 //  * SUM/COUNT need no prefix reconstruction at all: for delta streams
 //    sum(v_i, i=r0..r1) = nsel*first + sum_j w_j * d_j with closed-form
 //    weights, so the kernel does a weighted fold of raw deltas (exact mod

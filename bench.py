@@ -42,8 +42,8 @@ def log(msg):
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=3)
-    ap.add_argument("--warmup", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=6)
+    ap.add_argument("--warmup", type=int, default=2)
     ap.add_argument("--series", type=int,
                     default=int(os.environ.get("BYDB_BENCH_SERIES", 10000)))
     ap.add_argument("--dp", type=int,
