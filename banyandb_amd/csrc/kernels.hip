@@ -42,6 +42,14 @@
 
 #define WAVE 64
 
+// exact powers of ten for per-block decimal-exponent rescale
+__device__ __constant__ int64_t c_pow10i[19] = {
+    1LL, 10LL, 100LL, 1000LL, 10000LL, 100000LL, 1000000LL, 10000000LL,
+    100000000LL, 1000000000LL, 10000000000LL, 100000000000LL,
+    1000000000000LL, 10000000000000LL, 100000000000000LL,
+    1000000000000000LL, 10000000000000000LL, 100000000000000000LL,
+    1000000000000000000LL};
+
 // Plain (non-dictionary) tag columns — the >256-distinct-values fallback
 // (column.go:266-278, dictionary.go:58) — are zstd-compressed on disk; the
 // host normalizes them at part registration (encode.cpp) into a sidecar
@@ -1807,12 +1815,19 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
         if (seg != 0 && (int64_t)seg * SEG_ROWS + 1 > n - 1) continue;
         const int64_t ts_min = bd->ts_min, ts_max = bd->ts_max;
         // float64 blocks fold in the decimal-int domain; a block whose
-        // exponent differs from the session's would mix incomparable
-        // mantissas — loud error (the oracle handles mixed exponents)
-        if ((flags & KF_FLOAT) &&
-            bd->exp != (int16_t)(uint16_t)((uint32_t)flags >> 16)) {
-            dev_set_err(derr, DERR_EXP_MISMATCH, (uint64_t)bi);
-            continue;
+        // exponent EXCEEDS the session's rescales its partials by
+        // 10^(block_exp - cfg_exp) at the merge below (exact for the
+        // monotone min/max mantissas; the sum factor is a double).  The
+        // session exponent must be the part's minimum exponent — a block
+        // BELOW it cannot be represented and errors loudly.
+        int fdiff = 0;
+        if (flags & KF_FLOAT) {
+            fdiff = (int)bd->exp -
+                    (int)(int16_t)(uint16_t)((uint32_t)flags >> 16);
+            if (fdiff < 0 || fdiff > 18) {
+                dev_set_err(derr, DERR_EXP_MISMATCH, (uint64_t)bi);
+                continue;
+            }
         }
 
         // ---- row clamp (timestamp.FindRange, range.go:143-170) ----
@@ -1998,6 +2013,23 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
                                    EN_VALUES && (flags & KF_NEED_VALUES), lane,
                                    &rsum, &rmn, &rmx, &rhave, derr,
                                    (uint64_t)bi);
+                        double rfs = 1.0;
+                        if ((flags & KF_FLOAT) && fdiff) {
+                            int64_t mfac = c_pow10i[fdiff];
+                            rfs = (double)mfac;
+                            if (rhave) {
+                                int64_t smin, smax;
+                                if (__builtin_mul_overflow(rmn, mfac, &smin) ||
+                                    __builtin_mul_overflow(rmx, mfac, &smax)) {
+                                    dev_set_err(derr, DERR_EXP_MISMATCH,
+                                                (uint64_t)bi);
+                                    rhave = false;
+                                } else {
+                                    rmn = smin;
+                                    rmx = smax;
+                                }
+                            }
+                        }
                         if (comp != cur_group) {
                             flush_partial(partials, cur_group, wsum, wcnt,
                                           wmin, wmax, wsumf, lane);
@@ -2011,7 +2043,8 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
                             wmin = rmn < wmin ? rmn : wmin;
                             wmax = rmx > wmax ? rmx : wmax;
                         }
-                        if (flags & KF_FLOAT) wsumf += (double)(int64_t)rsum;
+                        if (flags & KF_FLOAT)
+                            wsumf += (double)(int64_t)rsum * rfs;
                     }
                     for (int sl = 0; sl < n_gslots; sl++)
                         if (run_hi_s[sl] == hi && gb[sl].uniform_gid == GID_VARYING)
@@ -2319,6 +2352,21 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
         // scan; instead fold block totals into per-wave registers and
         // flush only when the group changes or the wave is done.
         bsum = (uint64_t)__shfl((long long)bsum, 0);  // lane0 holds closed forms
+        double fscale = 1.0;
+        if ((flags & KF_FLOAT) && fdiff) {
+            int64_t mfac = c_pow10i[fdiff];
+            fscale = (double)mfac;
+            if (have_minmax) {
+                int64_t smin, smax;
+                if (__builtin_mul_overflow(bmin, mfac, &smin) ||
+                    __builtin_mul_overflow(bmax, mfac, &smax)) {
+                    dev_set_err(derr, DERR_EXP_MISMATCH, (uint64_t)bi);
+                    continue;
+                }
+                bmin = smin;
+                bmax = smax;
+            }
+        }
         if (block_group != cur_group) {
             flush_partial(partials, cur_group, wsum, wcnt, wmin, wmax, wsumf,
                           lane);
@@ -2331,7 +2379,7 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
             wmin = bmin < wmin ? bmin : wmin;
             wmax = bmax > wmax ? bmax : wmax;
         }
-        if (flags & KF_FLOAT) wsumf += (double)(int64_t)bsum;
+        if (flags & KF_FLOAT) wsumf += (double)(int64_t)bsum * fscale;
     }
     flush_partial(partials, cur_group, wsum, wcnt, wmin, wmax, wsumf, lane);
 }

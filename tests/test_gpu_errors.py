@@ -113,23 +113,56 @@ def test_sticky_error_then_reset_recovers():
     s.close()
 
 
-def test_float_exp_mismatch_fails_loud():
-    """A float64 block whose per-block decimal exponent differs from the
-    session's configured exponent must be a loud device error — mantissas
-    at different scales are incomparable (the oracle decodes each block
-    with its own exponent; per-block rescale on device is a later row)."""
+def test_float_exp_below_session_fails_loud():
+    """A block whose decimal exponent is BELOW the session's cannot be
+    represented in the session's mantissa domain — loud device error.
+    (Blocks ABOVE the session exponent rescale correctly; see
+    test_gpu_parity-style mixed-exponent coverage below.)"""
     from banyandb_amd import VT_FLOAT64
     b = PartBuilder()
     n = 128
     ts = [T0 + i * MS for i in range(n)]
-    # every value has cents divisible by 10 -> block encodes at exp=-1
-    b.add_block_f64(1, ts, [1] * n, [i / 10.0 for i in range(1, n + 1)])
+    b.add_block_f64(1, ts, [1] * n,
+                    [i / 100.0 for i in range(1, 2 * n, 2)])  # exp -2
     d = b.blocks()[0]
-    assert d.exp == -1
+    assert d.exp == -2
     s = Session(0)
     s.upload_part(b)
-    s.configure(VT_FLOAT64, [AGG_SUM, AGG_COUNT], float_exp=-2)
+    s.configure(VT_FLOAT64, [AGG_SUM, AGG_COUNT], float_exp=-1)
     s.consume()
     with pytest.raises(RuntimeError, match="decode error 5"):
         s.finalize()
     s.close()
+
+
+def test_float_mixed_exponent_blocks_rescale():
+    """Blocks at different decimal exponents fold correctly against a
+    session configured at the part's minimum exponent: the kernel
+    rescales each block's mantissa partials by 10^(block_exp - cfg_exp)
+    (the oracle restores per block natively)."""
+    import math
+    from banyandb_amd import VT_FLOAT64, AGG_MIN, AGG_MAX
+    from helpers import oracle_scan
+    b = PartBuilder()
+    n = 200
+    ts = [T0 + i * MS for i in range(n)]
+    b.add_block_f64(1, ts, [1] * n,
+                    [i / 100.0 for i in range(1, 2 * n, 2)])   # exp -2
+    b.add_block_f64(2, ts, [1] * n,
+                    [i / 10.0 for i in range(1, n + 1)])       # exp -1
+    b.add_block_f64(3, ts, [1] * n,
+                    [float(i + 5) for i in range(n)])          # exp 0
+    exps = sorted(d.exp for d in b.blocks())
+    assert exps == [-2, -1, 0]
+    orc = oracle_scan(b, VT_FLOAT64)[0]
+    s = Session(0)
+    s.upload_part(b)
+    s.configure(VT_FLOAT64, [AGG_SUM, AGG_COUNT, AGG_MIN, AGG_MAX],
+                float_exp=-2)
+    s.consume()
+    g = s.finalize()[0]
+    s.close()
+    assert g.count == orc.count == 3 * n
+    assert math.isclose(g.sum_f, orc.sum_f, rel_tol=1e-9)
+    assert math.isclose(g.min_f, orc.min_f, rel_tol=1e-12)
+    assert math.isclose(g.max_f, orc.max_f, rel_tol=1e-12)
