@@ -17,6 +17,7 @@ VT_FLOAT64 = 3
 AGG_SUM, AGG_COUNT, AGG_MIN, AGG_MAX, AGG_MEAN = range(5)
 MODE_ALL, MODE_MAP, MODE_REDUCE = range(3)
 INT64_MIN = -(2 ** 63)
+FLOAT_RAW_EXP = -32768  # sessions over nullable float64 columns
 INT64_MAX = 2 ** 63 - 1
 
 
@@ -152,6 +153,10 @@ def _load():
     lib.bydb_part_builder_destroy.argtypes = [C.c_void_p]
     lib.bydb_part_builder_error.restype = C.c_char_p
     lib.bydb_part_builder_error.argtypes = [C.c_void_p]
+    lib.bydb_part_builder_add_block_f64_nullable.restype = C.c_int
+    lib.bydb_part_builder_add_block_f64_nullable.argtypes = [
+        C.c_void_p, C.c_uint64, i64p, i64p, C.POINTER(C.c_double), u8p,
+        C.c_int64, C.c_uint32]
     lib.bydb_part_builder_add_block_i64_nullable.restype = C.c_int
     lib.bydb_part_builder_add_block_i64_nullable.argtypes = [
         C.c_void_p, C.c_uint64, i64p, i64p, i64p, u8p, C.c_int64, C.c_uint32]
@@ -238,6 +243,17 @@ class PartBuilder:
         valid = (C.c_uint8 * n)(*[0 if v is None else 1 for v in vals])
         vv = (C.c_int64 * n)(*[0 if v is None else v for v in vals])
         self._ck(_lib.bydb_part_builder_add_block_i64_nullable(
+            self._h, series_id, (C.c_int64 * n)(*ts),
+            (C.c_int64 * n)(*versions), vv, valid, n, group_code))
+
+    def add_block_f64_nullable(self, series_id, ts, versions, vals,
+                               group_code=0):
+        """Null-bearing float64 column (raw IEEE-754 cells; configure the
+        session with float_exp=FLOAT_RAW_EXP)."""
+        n = len(ts)
+        valid = (C.c_uint8 * n)(*[0 if v is None else 1 for v in vals])
+        vv = (C.c_double * n)(*[0.0 if v is None else v for v in vals])
+        self._ck(_lib.bydb_part_builder_add_block_f64_nullable(
             self._h, series_id, (C.c_int64 * n)(*ts),
             (C.c_int64 * n)(*versions), vv, valid, n, group_code))
 

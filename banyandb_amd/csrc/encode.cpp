@@ -688,6 +688,62 @@ static int add_block_common(bydb_part_builder *b, uint64_t series_id,
     return BYDB_OK;
 }
 
+// Null-bearing float64 column: 8-B big-endian IEEE-754 cells
+// (convert/number.go:128-132 Float64ToBytes), nil rows zero-length.
+extern "C" int bydb_part_builder_add_block_f64_nullable(
+    bydb_part_builder *b, uint64_t series_id, const int64_t *ts,
+    const int64_t *versions, const double *vals, const uint8_t *valid,
+    int64_t n, uint32_t group_code) {
+    if (!valid) {
+        b->err = "valid mask required";
+        return BYDB_ERR_BAD_ARG;
+    }
+    // reuse the common path with a pre-built cell list: encode the cells
+    // here and hand add_block_common a sentinel that writes them
+    if (n < 1 || n > 8192) {
+        b->err = "block row count out of range";
+        return BYDB_ERR_BAD_ARG;
+    }
+    // ts/version/desc handling matches add_block_common; the field stream
+    // is [ENC_PLAIN][bytes block of float cells]
+    b->scratch_i64.resize((size_t)n);
+    for (int64_t i = 0; i < n; i++) b->scratch_i64[(size_t)i] = 0;
+    int rc = add_block_common(b, series_id, ts, versions,
+                              b->scratch_i64.data(), 0, BYDB_VT_FLOAT64, n,
+                              group_code, valid);
+    if (rc != BYDB_OK) return rc;
+    // rewrite the just-written Plain cell payload with float bits: the
+    // common path wrote i64 sign-flip cells of zeros; redo the field
+    // stream properly
+    bydb_block_desc &d = b->blocks.back();
+    size_t fstart = (size_t)(d.field_off - b->base_off);
+    b->payload.resize(fstart);
+    b->payload.push_back(BYDB_ENC_PLAIN);
+    std::vector<uint8_t> cells;
+    std::vector<int64_t> clens((size_t)n);
+    cells.reserve((size_t)n * 8);
+    for (int64_t i = 0; i < n; i++) {
+        if (valid[i]) {
+            uint64_t bits;
+            memcpy(&bits, &vals[i], 8);
+            for (int k = 7; k >= 0; k--)
+                cells.push_back((uint8_t)(bits >> (8 * k)));
+            clens[(size_t)i] = 8;
+        } else {
+            clens[(size_t)i] = -1;
+        }
+    }
+    if (!bytes_block_append(b->payload, cells.data(), clens.data(), n)) {
+        b->err = "cell block encode failed (zstd unavailable?)";
+        b->blocks.pop_back();
+        b->payload.resize(fstart);
+        return BYDB_ERR_BAD_DATA;
+    }
+    d.field_len = (b->base_off + b->payload.size()) - d.field_off;
+    d.exp = 0;
+    return BYDB_OK;
+}
+
 // Null-bearing int64 column: valid[i] == 0 marks row i null.
 extern "C" int bydb_part_builder_add_block_i64_nullable(
     bydb_part_builder *b, uint64_t series_id, const int64_t *ts,

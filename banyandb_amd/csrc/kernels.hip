@@ -134,6 +134,30 @@ __device__ __forceinline__ uint64_t wave_reduce_add(uint64_t v) {
            readlane64(v, 63);
 }
 
+__device__ __forceinline__ double wave_reduce_addf(double v) {
+    uint64_t u;
+    __builtin_memcpy(&u, &v, 8);
+#define BYDB_RSTEPF(C)                                                       \
+    {                                                                        \
+        uint64_t tb = ((uint64_t)dpp_mov32<C>((uint32_t)(u >> 32)) << 32) |  \
+                      dpp_mov32<C>((uint32_t)u);                             \
+        double t;                                                            \
+        __builtin_memcpy(&t, &tb, 8);                                        \
+        v += t;                                                              \
+        __builtin_memcpy(&u, &v, 8);                                         \
+    }
+    BYDB_RSTEPF(0x111) BYDB_RSTEPF(0x112)
+    BYDB_RSTEPF(0x114) BYDB_RSTEPF(0x118)
+#undef BYDB_RSTEPF
+    double a, b, c, d;
+    uint64_t x;
+    x = readlane64(u, 15); __builtin_memcpy(&a, &x, 8);
+    x = readlane64(u, 31); __builtin_memcpy(&b, &x, 8);
+    x = readlane64(u, 47); __builtin_memcpy(&c, &x, 8);
+    x = readlane64(u, 63); __builtin_memcpy(&d, &x, 8);
+    return a + b + c + d;
+}
+
 __device__ __forceinline__ int64_t wave_reduce_min(int64_t v) {
     // row_shr moves are zero-filled out of row; feed the identity by
     // selecting against the shifted VALIDITY instead: use bound_ctrl 0's
@@ -1821,12 +1845,27 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
         // session exponent must be the part's minimum exponent — a block
         // BELOW it cannot be represented and errors loudly.
         int fdiff = 0;
+        bool raw_f64 = false;
         if (flags & KF_FLOAT) {
-            fdiff = (int)bd->exp -
-                    (int)(int16_t)(uint16_t)((uint32_t)flags >> 16);
-            if (fdiff < 0 || fdiff > 18) {
+            int16_t cfge = (int16_t)(uint16_t)((uint32_t)flags >> 16);
+            if (cfge == INT16_MIN) {
+                // raw-float session (nullable f64, IEEE-754 cells):
+                // every block must be a Plain cell block
+                raw_f64 = true;
+                if (bd->field_enc != BYDB_ENC_PLAIN) {
+                    dev_set_err(derr, DERR_EXP_MISMATCH, (uint64_t)bi);
+                    continue;
+                }
+            } else if (bd->field_enc == BYDB_ENC_PLAIN) {
+                // raw cells inside a decimal session: domains clash
                 dev_set_err(derr, DERR_EXP_MISMATCH, (uint64_t)bi);
                 continue;
+            } else {
+                fdiff = (int)bd->exp - (int)cfge;
+                if (fdiff < 0 || fdiff > 18) {
+                    dev_set_err(derr, DERR_EXP_MISMATCH, (uint64_t)bi);
+                    continue;
+                }
             }
         }
 
@@ -2312,6 +2351,7 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
             const uint64_t *vbm = (const uint64_t *)(fp + 8);
             const uint64_t *cells = vbm + ((uint64_t)nn + 63) / 64;
             uint64_t lsum = 0, lcnt = 0;
+            double lsumf = 0.0;
             int64_t lmn = INT64_MAX, lmx = INT64_MIN;
             for (int64_t base = r0; base <= r1; base += WAVE) {
                 int64_t row = base + lane;
@@ -2324,20 +2364,46 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
                     uint64_t c = cells[row];
                     // cells are big-endian on the wire
                     uint64_t u = __builtin_bswap64(c);
-                    int64_t v = (u >> 63) ? (int64_t)(u & ~(1ull << 63))
-                                          : -(int64_t)((1ull << 63) - u);
-                    lsum += (uint64_t)v;
+                    if (raw_f64) {
+                        // IEEE-754 bits: double sum + ordered-bits
+                        // min/max (Float64ToOrderedBytes map,
+                        // convert/number.go:148-157)
+                        double dv;
+                        __builtin_memcpy(&dv, &u, 8);
+                        lsumf += dv;
+                        // order-preserving map INTO THE SIGNED domain
+                        // (min/max compare as int64): positives keep raw
+                        // bits, negatives flip magnitude and sign
+                        int64_t kv =
+                            (u >> 63)
+                                ? (int64_t)(~u ^ 0x8000000000000000ull)
+                                : (int64_t)u;
+                        lmn = kv < lmn ? kv : lmn;
+                        lmx = kv > lmx ? kv : lmx;
+                    } else {
+                        int64_t v = (u >> 63)
+                                        ? (int64_t)(u & ~(1ull << 63))
+                                        : -(int64_t)((1ull << 63) - u);
+                        lsum += (uint64_t)v;
+                        lmn = v < lmn ? v : lmn;
+                        lmx = v > lmx ? v : lmx;
+                    }
                     lcnt++;
-                    lmn = v < lmn ? v : lmn;
-                    lmx = v > lmx ? v : lmx;
                 }
             }
-            bsum = wave_reduce_add(lsum);
+            if (raw_f64) {
+                double ds = wave_reduce_addf(lsumf);
+                uint64_t db;
+                __builtin_memcpy(&db, &ds, 8);
+                bsum = lane == 0 ? db : 0;
+            } else {
+                bsum = wave_reduce_add(lsum);
+                bsum = lane == 0 ? bsum : 0;
+            }
             nsel_eff = wave_reduce_add(lcnt);
             bmin = wave_reduce_min(lmn);
             bmax = wave_reduce_max(lmx);
             have_minmax = nsel_eff > 0;
-            bsum = lane == 0 ? bsum : 0;
         } else {
             dev_set_err(derr, DERR_BAD_ENC, (uint64_t)bi);
             continue;
@@ -2379,7 +2445,16 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
             wmin = bmin < wmin ? bmin : wmin;
             wmax = bmax > wmax ? bmax : wmax;
         }
-        if (flags & KF_FLOAT) wsumf += (double)(int64_t)bsum * fscale;
+        if (flags & KF_FLOAT) {
+            if (raw_f64) {
+                double db;
+                uint64_t ub = bsum;
+                __builtin_memcpy(&db, &ub, 8);
+                wsumf += db;
+            } else {
+                wsumf += (double)(int64_t)bsum * fscale;
+            }
+        }
     }
     flush_partial(partials, cur_group, wsum, wcnt, wmin, wmax, wsumf, lane);
 }
@@ -2621,10 +2696,6 @@ extern "C" int bydb_part_append(bydb_session *s, const uint8_t *payload,
         // can skip nulls per row (aggregation.go:310 null check)
         if (bd->field_enc == BYDB_ENC_PLAIN &&
             !(bd->field_off & TAG_SIDECAR_BIT)) {
-            if (bd->field_vtype != BYDB_VT_INT64) {
-                s->err = "nullable float64 columns not supported in v1";
-                return BYDB_ERR_BAD_ARG;
-            }
             const uint8_t *fsrc = payload + (bd->field_off - s->payload_len);
             std::vector<uint8_t> norm;
             if (!bydb_normalize_plain_field(fsrc, bd->field_len, bd->count,
@@ -3086,7 +3157,24 @@ static void partial_to_result(const bydb_partial *p, int field_vtype,
     r->count = p->count;
     r->min_i = p->min_i;
     r->max_i = p->max_i;
-    if (field_vtype == BYDB_VT_FLOAT64) {
+    if (field_vtype == BYDB_VT_FLOAT64 && float_exp == INT16_MIN) {
+        // raw IEEE-754 domain (nullable float columns): min/max carried
+        // as order-preserving bit keys (convert/number.go:148-157), sum
+        // carried directly as a double
+        auto key_to_f64 = [](int64_t k) {
+            uint64_t bits = k >= 0 ? (uint64_t)k
+                                   : ~((uint64_t)k ^ 0x8000000000000000ull);
+            double d;
+            memcpy(&d, &bits, 8);
+            return d;
+        };
+        r->min_f = p->count ? key_to_f64(p->min_i) : 1.7976931348623157e308;
+        r->max_f = p->count ? key_to_f64(p->max_i) : -1.7976931348623157e308;
+        r->sum_f = p->sum_f;
+        double c = (double)p->count;
+        r->mean_f = c == 0 ? 0 : (r->sum_f / c < 1 ? 1 : r->sum_f / c);
+        r->mean_i = 0;
+    } else if (field_vtype == BYDB_VT_FLOAT64) {
         // decimal-int domain -> float64 (monotone restore; float.go:69-102)
         r->min_f = p->count ? restore_f64(p->min_i, float_exp) : 1.7976931348623157e308;
         r->max_f = p->count ? restore_f64(p->max_i, float_exp) : -1.7976931348623157e308;

@@ -219,6 +219,17 @@ int bydb_part_builder_add_block_f64(bydb_part_builder *b, uint64_t series_id,
                                     const int64_t *ts, const int64_t *versions,
                                     const double *vals, int64_t n,
                                     uint32_t group_code);
+/* Raw-float sentinel for bydb_set_float_exp / bydb_reduce_partials2:
+ * sessions over null-bearing FLOAT64 columns (Plain IEEE-754 cell blocks,
+ * convert/number.go:128-132) fold in the raw-bits domain — ordered-bit
+ * min/max keys, direct double sums — selected by this exponent value.
+ * Mixing raw and decimal float blocks in one session is a loud error. */
+#define BYDB_FLOAT_RAW_EXP ((int16_t)-32768)
+
+int bydb_part_builder_add_block_f64_nullable(
+    bydb_part_builder *b, uint64_t series_id, const int64_t *ts,
+    const int64_t *versions, const double *vals, const uint8_t *valid,
+    int64_t n, uint32_t group_code);
 /* null-bearing int64 column (valid[i]==0 -> row i null): stored as the
  * reference's Plain fallback — a bytes block of 8-B sign-flip cells with
  * nil rows zero-length (column.go:214-233, convert/number.go:33-46);

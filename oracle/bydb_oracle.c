@@ -791,6 +791,33 @@ static int column_i64_decode_plain(int64_t *dst, uint8_t *valid,
     return rc;
 }
 
+/* Plain (null-bearing) float64 column: 8-B big-endian IEEE-754 cells
+ * (convert/number.go:128-132), nil rows zero-length. */
+static int column_f64_decode_plain(double *dst, uint8_t *valid,
+                                   const uint8_t *src, size_t len,
+                                   int64_t n) {
+    if (len < 1 || src[0] != BO_ENC_PLAIN) return BO_ERR_BAD_TYPE;
+    size_t cap = (size_t)n * 8 + 16;
+    uint8_t *data = (uint8_t *)malloc(cap);
+    int64_t *lens = (int64_t *)malloc(sizeof(int64_t) * (size_t)n);
+    size_t dl = 0;
+    int rc = bo_bytes_block_decode(data, cap, lens, src + 1, len - 1, n, &dl);
+    if (rc == BO_OK) {
+        size_t off = 0;
+        for (int64_t i = 0; i < n; i++) {
+            if (lens[i] < 0) { valid[i] = 0; dst[i] = 0; continue; }
+            if (lens[i] != 8) { rc = BO_ERR_BAD_DATA; break; }
+            valid[i] = 1;
+            uint64_t bits = 0;
+            for (int k = 0; k < 8; k++) bits = (bits << 8) | data[off + k];
+            memcpy(&dst[i], &bits, 8);
+            off += 8;
+        }
+    }
+    free(data); free(lens);
+    return rc;
+}
+
 /* BytesBlockDecoder.Decode — bytes.go:84-130 */
 int bo_bytes_block_decode(uint8_t *data_out, size_t data_cap, int64_t *lens_out,
                           const uint8_t *src, size_t src_len, int64_t n,
@@ -1185,8 +1212,18 @@ static int scan_block(const uint8_t *payload, const bo_block_desc *b,
             if (v > r->max_i) r->max_i = v;
         }
     } else if (field_vtype == BO_VT_FLOAT64) {
-        rc = bo_column_f64_decode(f64_buf, payload + b->col_off, b->col_len, n);
-        if (rc != BO_OK) return rc;
+        if (payload[b->col_off] == BO_ENC_PLAIN) {
+            uint8_t validf[8192];
+            rc = column_f64_decode_plain(f64_buf, validf,
+                                         payload + b->col_off, b->col_len, n);
+            if (rc != BO_OK) return rc;
+            for (int64_t i = 0; i < n; i++)
+                if (!validf[i]) rowmatch_buf[i] = 0;
+        } else {
+            rc = bo_column_f64_decode(f64_buf, payload + b->col_off,
+                                      b->col_len, n);
+            if (rc != BO_OK) return rc;
+        }
         for (int64_t i = r0; i <= r1; i++) {
             if (!rowmatch_buf[i]) continue;
             double v = f64_buf[i];

@@ -112,3 +112,55 @@ def test_all_null_block_gpu():
     b.add_block_i64_nullable(1, ts, [1] * n, [None] * n)
     gs, orc = run_both(b, [AGG_SUM, AGG_COUNT])
     assert gs[0].count == 0 == orc[0].count
+
+
+def test_nullable_f64_parity():
+    """Raw IEEE-754 cell blocks: session configured with FLOAT_RAW_EXP;
+    double sums, ordered-bit min/max keys."""
+    import math
+    from banyandb_amd import VT_FLOAT64, FLOAT_RAW_EXP
+    rng = random.Random(42)
+    b = PartBuilder()
+    for sid in range(6):
+        n = 2500
+        ts = [T0 + i * MS for i in range(n)]
+        vals = [None if rng.random() < 0.25
+                else rng.uniform(-1e9, 1e9) for _ in range(n)]
+        b.add_block_f64_nullable(sid + 1, ts, [1] * n, vals)
+    orc = oracle_scan(b, VT_FLOAT64)[0]
+    s = Session(0)
+    s.upload_part(b)
+    s.configure(VT_FLOAT64, [AGG_SUM, AGG_COUNT, AGG_MIN, AGG_MAX],
+                float_exp=FLOAT_RAW_EXP)
+    s.consume()
+    g = s.finalize()[0]
+    s.close()
+    assert orc.count > 0
+    assert g.count == orc.count
+    assert g.min_f == orc.min_f and g.max_f == orc.max_f
+    assert math.isclose(g.sum_f, orc.sum_f, rel_tol=1e-9)
+
+
+def test_nullable_f64_mixed_domain_fails_loud():
+    """A decimal float block inside a raw-float session (or vice versa)
+    is a loud device error — the min/max domains are incomparable."""
+    from banyandb_amd import VT_FLOAT64, FLOAT_RAW_EXP
+    b = PartBuilder()
+    n = 100
+    ts = [T0 + i * MS for i in range(n)]
+    b.add_block_f64_nullable(1, ts, [1] * n, [float(i) for i in range(n)])
+    b.add_block_f64(2, ts, [1] * n, [i / 100.0 for i in range(1, 2 * n, 2)])
+    s = Session(0)
+    s.upload_part(b)
+    s.configure(VT_FLOAT64, [AGG_SUM, AGG_COUNT], float_exp=FLOAT_RAW_EXP)
+    s.consume()
+    with pytest.raises(RuntimeError, match="decode error 5"):
+        s.finalize()
+    s.close()
+    s2 = Session(0)
+    s2.upload_part(b)
+    s2.configure(VT_FLOAT64, [AGG_SUM, AGG_COUNT], float_exp=-2)
+    s2.consume()
+    with pytest.raises(RuntimeError, match="decode error 5"):
+        s2.finalize()
+    s2.close()

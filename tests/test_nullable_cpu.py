@@ -56,3 +56,24 @@ def test_all_null_block():
     b.add_block_i64_nullable(1, ts, [1] * n, [None] * n)
     g = oracle_scan(b, VT_INT64)[0]
     assert g.count == 0
+
+
+def test_oracle_nullable_f64_matches_python():
+    import math
+    from banyandb_amd import VT_FLOAT64
+    rng = random.Random(41)
+    b = PartBuilder()
+    raw = []
+    for sid in range(3):
+        n = 1200
+        ts = [T0 + i * MS for i in range(n)]
+        vals = [None if rng.random() < 0.3
+                else rng.uniform(-1e6, 1e6) for _ in range(n)]
+        b.add_block_f64_nullable(sid + 1, ts, [1] * n, vals)
+        raw.append(vals)
+    g = oracle_scan(b, VT_FLOAT64)[0]
+    flat = [v for vals in raw for v in vals if v is not None]
+    assert g.count == len(flat)
+    assert math.isclose(g.sum_f, sum(flat), rel_tol=1e-12)
+    assert g.min_f == min(flat)
+    assert g.max_f == max(flat)
