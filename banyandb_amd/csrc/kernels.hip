@@ -653,6 +653,7 @@ struct PredBlock {
     uint8_t uniform;       // single run covers the whole block
     uint8_t plain;         // plain (non-dictionary) column: rle_bit_off is
                            // the per-row match-bitmap WORD offset instead
+    uint8_t disabled;      // slot carries no predicate (empty value)
 };
 
 // MSB-first bit read at an arbitrary bit offset (reader.go:39-79 order);
@@ -922,6 +923,12 @@ __device__ int resolve_one_pred(const uint8_t *__restrict__ payload,
     for (int i = 0; i < 4; i++) pb.mask[i] = 0;
     pb.rle_bit_off = 0; pb.nentries = 0; pb.width = 0;
     pb.active = 0; pb.err = 0; pb.uniform = 0; pb.plain = 0;
+    pb.disabled = 0;
+    if (pred_len == 0) {     // slot disabled: no constraint
+        pb.disabled = 1;
+        *out = pb;
+        return PF_CLEAR;
+    }
     if (tlen == 0) { *out = pb; return PF_SKIP; }   // nil tag: never equal
     const bool in_sidecar = (toff & TAG_SIDECAR_BIT) != 0;
     const uint8_t *base = in_sidecar ? sidecar : payload;
@@ -1940,6 +1947,7 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
                 for (int sl = 0; sl < 3; sl++) {
                     if (sl >= n_preds || skip_block) continue;
                     PredBlock pb = preds[(int64_t)sl * n_blocks + bi];
+                    if (pb.disabled) continue;   // empty slot: no constraint
                     if (!pb.active) { skip_block = true; continue; }
                     PredWalk *w = sl == 0 ? &pw0 : sl == 1 ? &pw1 : &pw2;
                     pred_init(w, payload, sidecar, &pb, pred_bm);
@@ -1986,7 +1994,11 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
                 for (int sl = 0; sl < n_gslots; sl++)
                     block_group += (int64_t)gb[sl].uniform_gid * gmul[sl];
             } else {
-                if (pred_on) {  // v1: row-varying groups + row-varying preds
+                // row-varying predicates join the run merge below as
+                // additional RLE cursors; bitmap-mode (plain-tag) walkers
+                // have no runs and stay a loud v1 limit here
+                if (pred_on && ((wp0 && wp0->bm) || (wp1 && wp1->bm) ||
+                                (wp2 && wp2->bm))) {
                     dev_set_err(derr, DERR_BAD_ENC, (uint64_t)bi);
                     continue;
                 }
@@ -2032,9 +2044,16 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
                     if (gb[sl].uniform_gid == GID_VARYING) advance(sl);
                 int64_t lo = 0;
                 while (lo < n) {
+                    // predicate cursors advance lazily to cover `lo`
+                    if (wp0) while (wp0->run_hi <= lo) pred_advance(wp0);
+                    if (wp1) while (wp1->run_hi <= lo) pred_advance(wp1);
+                    if (wp2) while (wp2->run_hi <= lo) pred_advance(wp2);
                     int64_t hi = n;
                     for (int sl = 0; sl < n_gslots; sl++)
                         hi = run_hi_s[sl] < hi ? run_hi_s[sl] : hi;
+                    if (wp0 && wp0->run_hi < hi) hi = wp0->run_hi;
+                    if (wp1 && wp1->run_hi < hi) hi = wp1->run_hi;
+                    if (wp2 && wp2->run_hi < hi) hi = wp2->run_hi;
                     int64_t aa = lo > r0 ? lo : r0;
                     int64_t bb2 = (hi - 1) < r1 ? (hi - 1) : r1;
                     int64_t comp = 0;
@@ -2043,6 +2062,9 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
                         if (gid_s[sl] < 0) ok = false;
                         else comp += gid_s[sl] * gmul[sl];
                     }
+                    if (wp0) ok = ok && wp0->run_match;
+                    if (wp1) ok = ok && wp1->run_match;
+                    if (wp2) ok = ok && wp2->run_match;
                     if (ok && aa <= bb2) {
                         uint64_t rsum;
                         int64_t rmn, rmx;
@@ -2886,16 +2908,11 @@ extern "C" int bydb_consume_multi(bydb_session *s, int64_t min_ts,
     if (!s->d_acc) { s->err = "configure first"; return BYDB_ERR_STATE; }
     if (s->n_blocks == 0) { s->err = "no part resident"; return BYDB_ERR_STATE; }
     if (n_preds < 0 || n_preds > 3) { s->err = "n_preds must be 0..3"; return BYDB_ERR_BAD_ARG; }
-    // drop trailing disabled slots
+    // drop trailing disabled slots; interior empty slots mean "no
+    // constraint on that tag"
     while (n_preds > 0 && pred_lens[n_preds - 1] == 0) n_preds--;
     PredBlock *preds = nullptr;
     if (n_preds > 0) {
-        for (int i = 0; i < n_preds; i++) {
-            if (pred_lens[i] == 0) {
-                s->err = "empty predicate in the middle of the slot list";
-                return BYDB_ERR_BAD_ARG;
-            }
-        }
         if (s->preds_cap < s->n_blocks * 3) {
             if (s->d_preds) (void)hipFree(s->d_preds);
             HIP_TRY(s, hipMalloc(&s->d_preds,
@@ -2949,6 +2966,7 @@ extern "C" int bydb_consume_multi(bydb_session *s, int64_t min_ts,
             int pgrid = (int)(s->n_blocks < 65535 ? s->n_blocks : 65535);
             off = 0;
             for (int i = 0; i < n_preds; i++) {
+                if (pred_lens[i] == 0) continue;
                 hipLaunchKernelGGL(k_resolve_plain, dim3(pgrid), dim3(WAVE), 0,
                                    s->stream, s->d_payload, s->d_sidecar,
                                    s->d_blocks, s->n_blocks,

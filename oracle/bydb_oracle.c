@@ -1287,6 +1287,7 @@ int bo_scan_agg_bytags(const uint8_t *payload, const bo_block_desc *blocks,
                        int64_t max_ts, const int *slots, int n_slots,
                        const uint8_t *const *dom_blobs,
                        const int64_t *const *dom_lens, const int64_t *n_doms,
+                       const uint8_t *preds_concat, const int64_t *pred_lens,
                        bo_agg_result *out) {
     int64_t total = 1;
     for (int i = 0; i < n_slots; i++) total *= n_doms[i];
@@ -1309,6 +1310,48 @@ int bo_scan_agg_bytags(const uint8_t *payload, const bo_block_desc *blocks,
         int64_t r0, r1;
         if (!bo_find_range(ts_buf, n, min_ts, max_ts, &r0, &r1)) continue;
         for (int64_t r = 0; r < n; r++) rowgid[r] = 0;
+        /* conjunctive tag-equality predicates (same contract as the
+         * scalar fold: nil-tag block excluded; dictionary or plain) */
+        int pdropped = 0;
+        if (pred_lens) {
+            int64_t pred_off = 0;
+            for (int sl = 0; sl < 3 && rc == BO_OK && !pdropped; sl++) {
+                int64_t plen = pred_lens[sl];
+                if (plen == 0) continue;
+                const uint8_t *pred = preds_concat + pred_off;
+                pred_off += plen;
+                uint64_t toff2 = sl == 0 ? b->tag_off
+                                 : sl == 1 ? b->tag2_off : b->tag3_off;
+                uint64_t tlen2 = sl == 0 ? b->tag_len
+                                 : sl == 1 ? b->tag2_len : b->tag3_len;
+                if (tlen2 == 0) { pdropped = 1; break; }
+                uint8_t tt = payload[toff2];
+                size_t tdl = 0;
+                if (tt == BO_ENC_DICTIONARY)
+                    rc = bo_dictionary_decode(tagdata, (size_t)1 << 24,
+                                              taglen, payload + toff2 + 1,
+                                              tlen2 - 1, n, &tdl);
+                else if (tt == BO_ENC_PLAIN)
+                    rc = bo_bytes_block_decode(tagdata, (size_t)1 << 24,
+                                               taglen, payload + toff2 + 1,
+                                               tlen2 - 1, n, &tdl);
+                else
+                    rc = BO_ERR_BAD_TYPE;
+                if (rc != BO_OK) break;
+                size_t tago = 0;
+                for (int64_t r = 0; r < n; r++) {
+                    int64_t tl = taglen[r];
+                    size_t my = tago;
+                    if (tl > 0) tago += (size_t)tl;
+                    if (tl != plen ||
+                        (tl > 0 &&
+                         memcmp(tagdata + my, pred, (size_t)tl) != 0))
+                        rowgid[r] = -1;
+                }
+            }
+        }
+        if (rc != BO_OK) break;
+        if (pdropped) continue;
         int64_t mul = 1;
         int dropped = 0;
         for (int si = 0; si < n_slots && rc == BO_OK && !dropped; si++) {
@@ -1400,7 +1443,8 @@ int bo_scan_agg_bytag(const uint8_t *payload, const bo_block_desc *blocks,
     const int64_t *lens[1] = {dom_lens};
     const int64_t nd[1] = {n_dom};
     return bo_scan_agg_bytags(payload, blocks, n_blocks, field_vtype, min_ts,
-                              max_ts, slots, 1, blobs, lens, nd, out);
+                              max_ts, slots, 1, blobs, lens, nd, NULL, NULL,
+                              out);
 }
 
 /* ===================== xxhash64 =====================

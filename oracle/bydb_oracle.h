@@ -203,6 +203,7 @@ int bo_scan_agg_bytags(const uint8_t *payload, const bo_block_desc *blocks,
                        int64_t max_ts, const int *slots, int n_slots,
                        const uint8_t *const *dom_blobs,
                        const int64_t *const *dom_lens, const int64_t *n_doms,
+                       const uint8_t *preds_concat, const int64_t *pred_lens,
                        bo_agg_result *out);
 int bo_scan_agg_bytag(const uint8_t *payload, const bo_block_desc *blocks,
                       int64_t n_blocks, int field_vtype, int64_t min_ts,

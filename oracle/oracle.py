@@ -134,7 +134,7 @@ _L.bo_scan_agg_bytags.restype = C.c_int
 _L.bo_scan_agg_bytags.argtypes = [u8p, C.POINTER(BlockDesc), C.c_int64, C.c_int,
                                   C.c_int64, C.c_int64, C.POINTER(C.c_int),
                                   C.c_int, C.POINTER(u8p), C.POINTER(i64p),
-                                  i64p, C.POINTER(AggResult)]
+                                  i64p, u8p, i64p, C.POINTER(AggResult)]
 _L.bo_scan_agg_bytag.restype = C.c_int
 _L.bo_scan_agg_bytag.argtypes = [u8p, C.POINTER(BlockDesc), C.c_int64, C.c_int,
                                  C.c_int64, C.c_int64, C.c_int, u8p, i64p,
@@ -414,7 +414,7 @@ def scan_agg_bytag(payload: bytes, blocks, field_vtype, slot, domain,
 
 
 def scan_agg_bytags(payload: bytes, blocks, field_vtype, slots, domains,
-                    min_ts=INT64_MIN, max_ts=INT64_MAX):
+                    min_ts=INT64_MIN, max_ts=INT64_MAX, preds=None):
     """Composite group-by over multiple tag slots; gid = g0 + n0*g1 + ..."""
     descs = (BlockDesc * len(blocks))()
     for i, b in enumerate(blocks):
@@ -435,11 +435,21 @@ def scan_agg_bytags(payload: bytes, blocks, field_vtype, slots, domains,
         total *= len(dom)
     out = (AggResult * total)()
     src = (C.c_uint8 * max(len(payload), 1)).from_buffer_copy(payload or b"\0")
+    predbuf = None
+    plens = None
+    if preds is not None:
+        concat = b"".join(p or b"" for p in preds)
+        lens3 = [len(p or b"") for p in preds] + [0, 0, 0]
+        predbuf = (C.c_uint8 * max(len(concat), 1)).from_buffer_copy(
+            concat or b"\0")
+        plens = (C.c_int64 * 3)(*lens3[:3])
     rc = _L.bo_scan_agg_bytags(
         src, descs, len(blocks), field_vtype, min_ts, max_ts,
         (C.c_int * len(slots))(*slots), len(slots),
         (u8p * len(domains))(*blobs), (i64p * len(domains))(*lens_arrs),
-        (C.c_int64 * len(domains))(*nds), out)
+        (C.c_int64 * len(domains))(*nds),
+        C.cast(predbuf, u8p) if predbuf else None,
+        C.cast(plens, i64p) if plens else None, out)
     if rc != 0:
         raise ValueError(f"scan_agg_bytags rc={rc}")
     return list(out)

@@ -216,3 +216,86 @@ def test_composite_both_rowvarying():
     for g, oc in zip(gs, orc):
         assert g.count == oc.count
         assert g.sum_i == oc.sum_i
+
+
+def test_rowvarying_groups_with_rowvarying_predicate():
+    """Row-varying group-by slot + row-varying predicate slot: the
+    predicate's RLE runs join the group-run merge as extra cursors."""
+    rng = random.Random(63)
+    regions = [b"r0", b"r1", b"r2", b"r3"]
+    b = PartBuilder()
+    for sid in range(8):
+        n = 4000
+        ts = [T0 + i * MS for i in range(n)]
+        b.add_block_i64(sid + 1, ts, [1] * n,
+                        [rng.randint(-10**9, 10**9) for _ in range(n)])
+        tags = []
+        while len(tags) < n:
+            run = min(rng.randint(1, 90), n - len(tags))
+            v = None if rng.random() < 0.05 else ENVS[rng.randrange(4)]
+            tags.extend([v] * run)
+        b.set_block_tag(tags)                    # slot 0: grouped
+        tags2 = []
+        while len(tags2) < n:
+            run = min(rng.randint(1, 140), n - len(tags2))
+            tags2.extend([regions[rng.randrange(4)]] * run)
+        b.set_block_tag(tags2)                   # slot 1: predicated
+    import oracle as o
+    from helpers import oracle_blocks
+    payload, blocks = oracle_blocks(b)
+    orc = o.scan_agg_bytags(payload, blocks, VT_INT64, [0], [ENVS],
+                            preds=[b"", b"r2", b""])
+    s = Session(0)
+    s.upload_part(b)
+    s.configure_by_tag(VT_INT64, [AGG_SUM, AGG_COUNT, AGG_MIN, AGG_MAX], 0,
+                       ENVS)
+    s.consume(preds=[b"", b"r2", b""])
+    gs = s.finalize()
+    s.close()
+    assert sum(oc.count for oc in orc) > 0
+    for g, oc in zip(gs, orc):
+        assert g.count == oc.count
+        assert g.sum_i == oc.sum_i
+        if oc.count:
+            assert g.min_i == oc.min_i and g.max_i == oc.max_i
+
+
+def test_composite_groups_with_predicate_and_clamp():
+    rng = random.Random(64)
+    regions = [b"x", b"yy"]
+    svcs = [b"s0", b"s1", b"s2"]
+    b = PartBuilder()
+    for sid in range(6):
+        n = 3000
+        ts = [T0 + i * MS for i in range(n)]
+        b.add_block_i64(sid + 1, ts, [1] * n,
+                        [rng.randint(0, 99999) for _ in range(n)])
+        for table, maxrun in ((ENVS, 70), (regions, 110)):
+            tags = []
+            while len(tags) < n:
+                run = min(rng.randint(1, maxrun), n - len(tags))
+                tags.extend([table[rng.randrange(len(table))]] * run)
+            b.set_block_tag(tags)
+        tags3 = []
+        while len(tags3) < n:
+            run = min(rng.randint(1, 60), n - len(tags3))
+            tags3.extend([svcs[rng.randrange(3)]] * run)
+        b.set_block_tag(tags3)                   # slot 2: predicated
+    import oracle as o
+    from helpers import oracle_blocks
+    payload, blocks = oracle_blocks(b)
+    lo, hi = T0 + 200 * MS, T0 + 2600 * MS
+    orc = o.scan_agg_bytags(payload, blocks, VT_INT64, [0, 1],
+                            [ENVS, regions], min_ts=lo, max_ts=hi,
+                            preds=[b"", b"", b"s1"])
+    s = Session(0)
+    s.upload_part(b)
+    s.configure_by_tags(VT_INT64, [AGG_SUM, AGG_COUNT], [0, 1],
+                        [ENVS, regions])
+    s.consume(min_ts=lo, max_ts=hi, preds=[b"", b"", b"s1"])
+    gs = s.finalize()
+    s.close()
+    assert sum(oc.count for oc in orc) > 0
+    for g, oc in zip(gs, orc):
+        assert g.count == oc.count
+        assert g.sum_i == oc.sum_i
