@@ -1680,11 +1680,72 @@ __device__ void fold_range(const uint8_t *fstream, uint8_t fenc, int64_t first,
                            uint64_t field_len, int64_t n, int64_t a, int64_t b,
                            bool need_values, int lane, uint64_t *out_sum,
                            int64_t *out_mn, int64_t *out_mx, bool *out_have_mm,
-                           DevErr *derr, uint64_t bi) {
+                           uint64_t *out_cnt, const uint8_t *plain_norm,
+                           bool raw_f64, DevErr *derr, uint64_t bi) {
     uint64_t nsel = (uint64_t)(b - a + 1);
     uint64_t bsum = 0;
     int64_t bmn = INT64_MAX, bmx = INT64_MIN;
     bool have = false;
+    *out_cnt = nsel;
+    if (fenc == BYDB_ENC_PLAIN) {
+        // null-bearing cell block (host-normalized; see the main Plain
+        // fold branch): count excludes nulls
+        if (!plain_norm) {
+            dev_set_err(derr, DERR_BAD_ENC, bi);
+            *out_sum = 0; *out_mn = bmn; *out_mx = bmx; *out_have_mm = false;
+            *out_cnt = 0;
+            return;
+        }
+        uint32_t nn = (uint32_t)plain_norm[0] | ((uint32_t)plain_norm[1] << 8) |
+                      ((uint32_t)plain_norm[2] << 16) |
+                      ((uint32_t)plain_norm[3] << 24);
+        const uint64_t *vbm = (const uint64_t *)(plain_norm + 8);
+        const uint64_t *cells = vbm + ((uint64_t)nn + 63) / 64;
+        uint64_t lsum = 0, lcnt = 0;
+        double lsumf = 0.0;
+        int64_t lmn = INT64_MAX, lmx = INT64_MIN;
+        for (int64_t base = a; base <= b; base += WAVE) {
+            int64_t row = base + lane;
+            bool ok = row <= b && ((vbm[row >> 6] >> (row & 63)) & 1);
+            if (ok) {
+                uint64_t u = __builtin_bswap64(cells[row]);
+                if (raw_f64) {
+                    double dv;
+                    __builtin_memcpy(&dv, &u, 8);
+                    lsumf += dv;
+                    int64_t kv = (u >> 63)
+                                     ? (int64_t)(~u ^ 0x8000000000000000ull)
+                                     : (int64_t)u;
+                    lmn = kv < lmn ? kv : lmn;
+                    lmx = kv > lmx ? kv : lmx;
+                } else {
+                    int64_t v = (u >> 63) ? (int64_t)(u & ~(1ull << 63))
+                                          : -(int64_t)((1ull << 63) - u);
+                    lsum += (uint64_t)v;
+                    lmn = v < lmn ? v : lmn;
+                    lmx = v > lmx ? v : lmx;
+                }
+                lcnt++;
+            }
+        }
+        if (raw_f64) {
+            double ds = wave_reduce_addf(lsumf);
+            uint64_t db;
+            __builtin_memcpy(&db, &ds, 8);
+            bsum = db;
+        } else {
+            bsum = wave_reduce_add(lsum);
+        }
+        *out_cnt = wave_reduce_add(lcnt);
+        bmn = wave_reduce_min(lmn);
+        bmx = wave_reduce_max(lmx);
+        have = *out_cnt > 0;
+        *out_sum = bsum;
+        *out_mn = bmn;
+        *out_mx = bmx;
+        *out_have_mm = have;
+        return;
+    }
     if (fenc == BYDB_ENC_CONST) {
         if (lane == 0) bsum = (uint64_t)first * nsel;
         bmn = bmx = first;
@@ -2004,6 +2065,11 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
                 }
                 if (seg != 0) continue;  // whole block folded at seg 0
                 const uint8_t *gstream = payload + bd->field_off;
+                const uint8_t *plain_norm =
+                    (bd->field_enc == BYDB_ENC_PLAIN &&
+                     (bd->field_off & TAG_SIDECAR_BIT))
+                        ? sidecar + (bd->field_off & ~TAG_SIDECAR_BIT)
+                        : nullptr;
                 // per-slot run cursors (uniform slots are one infinite run)
                 const uint8_t *gsrc[3];
                 uint64_t bit[3];
@@ -2066,14 +2132,14 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
                     if (wp1) ok = ok && wp1->run_match;
                     if (wp2) ok = ok && wp2->run_match;
                     if (ok && aa <= bb2) {
-                        uint64_t rsum;
+                        uint64_t rsum, rcnt;
                         int64_t rmn, rmx;
                         bool rhave;
                         fold_range(gstream, bd->field_enc, bd->field_first,
                                    bd->field_len, n, aa, bb2,
                                    EN_VALUES && (flags & KF_NEED_VALUES), lane,
-                                   &rsum, &rmn, &rmx, &rhave, derr,
-                                   (uint64_t)bi);
+                                   &rsum, &rmn, &rmx, &rhave, &rcnt,
+                                   plain_norm, raw_f64, derr, (uint64_t)bi);
                         double rfs = 1.0;
                         if ((flags & KF_FLOAT) && fdiff) {
                             int64_t mfac = c_pow10i[fdiff];
@@ -2099,13 +2165,21 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
                             wmax = INT64_MIN; wsumf = 0;
                         }
                         wsum += rsum;
-                        wcnt += (uint64_t)(bb2 - aa + 1);
+                        wcnt += rcnt;
                         if (rhave) {
                             wmin = rmn < wmin ? rmn : wmin;
                             wmax = rmx > wmax ? rmx : wmax;
                         }
-                        if (flags & KF_FLOAT)
-                            wsumf += (double)(int64_t)rsum * rfs;
+                        if (flags & KF_FLOAT) {
+                            if (raw_f64) {
+                                double db;
+                                uint64_t ub = rsum;
+                                __builtin_memcpy(&db, &ub, 8);
+                                wsumf += db;
+                            } else {
+                                wsumf += (double)(int64_t)rsum * rfs;
+                            }
+                        }
                     }
                     for (int sl = 0; sl < n_gslots; sl++)
                         if (run_hi_s[sl] == hi && gb[sl].uniform_gid == GID_VARYING)

@@ -1406,9 +1406,20 @@ int bo_scan_agg_bytags(const uint8_t *payload, const bo_block_desc *blocks,
                 rc = bo_column_i64_decode(i64_buf, payload + b->col_off,
                                           b->col_len, n);
             }
-        } else if (field_vtype == BO_VT_FLOAT64)
-            rc = bo_column_f64_decode(f64_buf, payload + b->col_off, b->col_len, n);
-        else rc = BO_ERR_BAD_TYPE;
+        } else if (field_vtype == BO_VT_FLOAT64) {
+            if (payload[b->col_off] == BO_ENC_PLAIN) {
+                uint8_t validf[8192];
+                rc = column_f64_decode_plain(f64_buf, validf,
+                                             payload + b->col_off,
+                                             b->col_len, n);
+                if (rc == BO_OK)
+                    for (int64_t q = 0; q < n; q++)
+                        if (!validf[q]) rowgid[q] = -1;
+            } else {
+                rc = bo_column_f64_decode(f64_buf, payload + b->col_off,
+                                          b->col_len, n);
+            }
+        } else rc = BO_ERR_BAD_TYPE;
         if (rc != BO_OK) break;
         for (int64_t r = r0; r <= r1; r++) {
             if (rowgid[r] < 0) continue;

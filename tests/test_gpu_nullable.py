@@ -164,3 +164,73 @@ def test_nullable_f64_mixed_domain_fails_loud():
     with pytest.raises(RuntimeError, match="decode error 5"):
         s2.finalize()
     s2.close()
+
+
+def test_nullable_with_rowvarying_groupby():
+    """Null-bearing int64 fields under per-row (RLE) group-by: the group
+    run fold counts only valid rows."""
+    import oracle as o
+    from helpers import oracle_blocks
+    rng = random.Random(36)
+    b = PartBuilder()
+    for sid in range(6):
+        n = 3000
+        ts = [T0 + i * MS for i in range(n)]
+        b.add_block_i64_nullable(sid + 1, ts, [1] * n,
+                                 _nullable_vals(rng, n, 0.3))
+        tags = []
+        while len(tags) < n:
+            run = min(rng.randint(1, 100), n - len(tags))
+            tags.extend([ENVS[rng.randrange(4)]] * run)
+        b.set_block_tag(tags)
+    payload, blocks = oracle_blocks(b)
+    orc = o.scan_agg_bytag(payload, blocks, VT_INT64, 0, ENVS)
+    s = Session(0)
+    s.upload_part(b)
+    s.configure_by_tag(VT_INT64, [AGG_SUM, AGG_COUNT, AGG_MIN, AGG_MAX], 0,
+                       ENVS)
+    s.consume()
+    gs = s.finalize()
+    s.close()
+    assert sum(oc.count for oc in orc) > 0
+    for g, oc in zip(gs, orc):
+        assert g.count == oc.count
+        assert g.sum_i == oc.sum_i
+        if oc.count:
+            assert g.min_i == oc.min_i and g.max_i == oc.max_i
+
+
+def test_nullable_f64_with_rowvarying_groupby():
+    import math
+    import oracle as o
+    from helpers import oracle_blocks
+    from banyandb_amd import VT_FLOAT64, FLOAT_RAW_EXP
+    rng = random.Random(37)
+    b = PartBuilder()
+    for sid in range(5):
+        n = 2000
+        ts = [T0 + i * MS for i in range(n)]
+        vals = [None if rng.random() < 0.2
+                else rng.uniform(-1e6, 1e6) for _ in range(n)]
+        b.add_block_f64_nullable(sid + 1, ts, [1] * n, vals)
+        tags = []
+        while len(tags) < n:
+            run = min(rng.randint(1, 80), n - len(tags))
+            tags.extend([ENVS[rng.randrange(4)]] * run)
+        b.set_block_tag(tags)
+    payload, blocks = oracle_blocks(b)
+    orc = o.scan_agg_bytag(payload, blocks, VT_FLOAT64, 0, ENVS)
+    s = Session(0)
+    s.upload_part(b)
+    s.configure_by_tag(VT_FLOAT64, [AGG_SUM, AGG_COUNT, AGG_MIN, AGG_MAX], 0,
+                       ENVS, float_exp=FLOAT_RAW_EXP)
+    s.consume()
+    gs = s.finalize()
+    s.close()
+    assert sum(oc.count for oc in orc) > 0
+    for g, oc in zip(gs, orc):
+        assert g.count == oc.count
+        if oc.count:
+            assert g.min_f == oc.min_f and g.max_f == oc.max_f
+            assert math.isclose(g.sum_f, oc.sum_f, rel_tol=1e-9,
+                                abs_tol=1e-6)
