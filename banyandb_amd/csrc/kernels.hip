@@ -1171,6 +1171,9 @@ __device__ void scan_stream(const uint8_t *stream, int64_t n_deltas, bool dod,
     uint64_t carry_u = 0;
     uint32_t carry_n = 0;
     const bool no_walk = pw0 == nullptr && pw1 == nullptr && pw2 == nullptr;
+    // bound counting is live only on the timestamp-clamp scan; value
+    // folds pass INT64_MAX/MIN and skip the two 64-bit compares per value
+    const bool want_bounds = lo_bound != INT64_MAX || hi_bound != INT64_MIN;
     uint32_t w_cur = 0, w_nx1 = 0;
     int w_valid = 0;   // how many upcoming quad windows are already loaded
     while (j <= jmax) {
@@ -1238,8 +1241,10 @@ __device__ void scan_stream(const uint8_t *stream, int64_t n_deltas, bool dod,
                         l_mn = sv < l_mn ? sv : l_mn;                        \
                         l_mx = sv > l_mx ? sv : l_mx;                        \
                     }                                                        \
-                    if (sv < lo_bound) l_nlo++;                              \
-                    if (sv > hi_bound) l_nhi++;                              \
+                    if (want_bounds) {                                       \
+                        if (sv < lo_bound) l_nlo++;                          \
+                        if (sv > hi_bound) l_nhi++;                          \
+                    }                                                        \
                     idx++;                                                   \
                 }
                 BYDB_QVAL(t0, dA)
