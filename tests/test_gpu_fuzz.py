@@ -87,7 +87,8 @@ def build_scenario(rng, n_groups=1):
 
 def check_scalar(rng, b, is_float, tag_kind, sess_dev=0):
     vtype = VT_FLOAT64 if is_float else VT_INT64
-    exp = b.blocks()[0].exp if is_float else 0
+    # mixed per-block exponents rescale against the part MINIMUM exponent
+    exp = min(d.exp for d in b.blocks()) if is_float else 0
     # random clamp
     mode = rng.randrange(3)
     kw = {}
@@ -118,7 +119,7 @@ def check_scalar(rng, b, is_float, tag_kind, sess_dev=0):
 
 def check_bytag(rng, b, is_float):
     vtype = VT_FLOAT64 if is_float else VT_INT64
-    exp = b.blocks()[0].exp if is_float else 0
+    exp = min(d.exp for d in b.blocks()) if is_float else 0
     domain = list(TAGS[: rng.randint(2, len(TAGS))])
     payload, blocks = oracle_blocks(b)
     orc = o.scan_agg_bytag(payload, blocks, vtype, 0, domain)
@@ -156,7 +157,7 @@ def test_fuzz_grouped_by_code(seed):
     rng = random.Random(0x1F0 + seed)
     b, is_float, _ = build_scenario(rng, n_groups=3)
     vtype = VT_FLOAT64 if is_float else VT_INT64
-    exp = b.blocks()[0].exp if is_float else 0
+    exp = min(d.exp for d in b.blocks()) if is_float else 0
     orc = oracle_scan(b, vtype, n_groups=3)
     s = Session(0)
     s.upload_part(b)
