@@ -59,3 +59,20 @@ def test_session_fails_loudly_without_gpu():
         raise AssertionError("Session() must fail without a GPU")
     except RuntimeError as e:
         assert "no CPU fallback" in str(e) or "failed" in str(e)
+
+
+def test_header_compiles_as_c99():
+    """The drop-in header must stay consumable by cgo: compile and link
+    the pure-C consumer (tools/cabi_check.c) with gcc -std=c99."""
+    import subprocess, os, tempfile
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    with tempfile.TemporaryDirectory() as td:
+        out = os.path.join(td, "cabi_check")
+        subprocess.run(
+            ["gcc", "-std=c99", "-Wall", "-Werror", "-I",
+             os.path.join(repo, "include"),
+             os.path.join(repo, "tools", "cabi_check.c"),
+             "-L", os.path.join(repo, "banyandb_amd"), "-lbydb_gpu",
+             "-Wl,-rpath," + os.path.join(repo, "banyandb_amd"),
+             "-o", out],
+            check=True)
