@@ -1289,7 +1289,7 @@ __device__ void scan_stream(const uint8_t *stream, int64_t n_deltas, bool dod,
         if (pw0 == nullptr && pw1 == nullptr && pw2 == nullptr &&
             (cont & (cont << 1)) == 0 &&
             (int64_t)nterm <= jmax - j + 1 &&
-            (carry_n == 0 || (emask & 1))) {
+            (carry_n == 0 || (carry_n == 1 && (emask & 1)))) {
             uint32_t prev = (uint32_t)__shfl_up((int)b, 1);
             bool is2 = lane > 0 && (prev & 0x80u);
             uint32_t u = is2 ? ((prev & 0x7fu) | ((uint32_t)b << 7)) : b;
