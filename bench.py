@@ -228,6 +228,9 @@ def main():
                 allreduce_partials(dist, part_t, n_groups,
                                    need_minmax=need_minmax,
                                    need_float=W["float"])
+                # the collective runs on torch's stream; the next step's
+                # reset() runs on the session stream — order them
+                torch.cuda.synchronize(device)
             else:
                 t = partials_from_structs(parts)
                 allreduce_partials(dist, t, n_groups,
@@ -239,6 +242,7 @@ def main():
                     allreduce_partials(dist, part_t2, n_groups,
                                        need_minmax=need_minmax,
                                        need_float=True)
+                    torch.cuda.synchronize(device)
                 else:
                     t2 = partials_from_structs(parts2)
                     allreduce_partials(dist, t2, n_groups,
