@@ -1797,13 +1797,11 @@ __device__ void fold_range(const uint8_t *fstream, uint8_t fenc, int64_t first,
                 scan_stream(fstream, n - 1, false, first, 0, a, b, INT64_MAX,
                             INT64_MIN, lane, &ff, derr, bi, nullptr, nullptr,
                             nullptr);
-                uint64_t lsum2 = ff.sum;
                 int64_t lmn = ff.mn, lmx = ff.mx;
                 if (lane == 0 && a <= 0 && 0 <= b) {
                     lmn = first < lmn ? first : lmn;
                     lmx = first > lmx ? first : lmx;
                 }
-                (void)lsum2;
                 bmn = wave_reduce_min(lmn);
                 bmx = wave_reduce_max(lmx);
             }
