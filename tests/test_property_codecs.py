@@ -124,3 +124,44 @@ def test_nullable_block_parity(vals):
     assert g.min_i == min(flat)
     assert g.max_i == max(flat)
     assert g.sum_i == sum(flat)
+
+
+@settings(max_examples=100, deadline=None)
+@given(st.lists(st.integers(min_value=0, max_value=10**6),
+                min_size=1, max_size=300),
+       st.lists(st.integers(min_value=0, max_value=50), min_size=1,
+                max_size=300))
+def test_timestamps_with_versions_roundtrip(strides, vers_raw):
+    """timestamps+versions codec (block.go:386-443, WithVersion encode
+    variants 5-8): ascending timestamps with arbitrary version lists
+    round-trip through one payload."""
+    n = min(len(strides), len(vers_raw))
+    ts = []
+    t = T0
+    for k in range(n):
+        ts.append(t)
+        t += MS + strides[k] * MS
+    versions = [1 + v for v in vers_raw[:n]]
+    meta = o.timestamps_encode(ts, versions)
+    assert 5 <= meta["enc"] <= 8  # always a WithVersion variant
+    assert meta["ts_min"] == ts[0] and meta["ts_max"] == ts[-1]
+    ts2, vers2 = o.timestamps_decode(meta, n)
+    assert ts2 == ts
+    assert vers2 == versions
+
+
+@settings(max_examples=80, deadline=None)
+@given(st.lists(st.integers(min_value=-10**9, max_value=10**9),
+                min_size=1, max_size=200),
+       st.integers(min_value=-(2**62), max_value=2**62),
+       st.integers(min_value=-(2**62), max_value=2**62))
+def test_find_range_matches_python(vals, lo, hi):
+    """FindRange (timestamp/range.go:143-170 inclusive clamp) on an
+    ascending list equals the straightforward Python filter."""
+    ts = sorted(vals)
+    s0, e0, found = o.find_range(ts, lo, hi)
+    idx = [i for i, t in enumerate(ts) if lo <= t <= hi]
+    if not idx:
+        assert not found
+    else:
+        assert found and (s0, e0) == (idx[0], idx[-1])
