@@ -165,3 +165,27 @@ def test_find_range_matches_python(vals, lo, hi):
         assert not found
     else:
         assert found and (s0, e0) == (idx[0], idx[-1])
+
+
+@settings(max_examples=50, deadline=None)
+@given(st.lists(st.integers(min_value=-(2**62), max_value=2**62),
+                min_size=1, max_size=50),
+       st.lists(st.one_of(st.none(),
+                          st.text(min_size=0, max_size=8).map(
+                              lambda t: t.encode()))
+                , min_size=1, max_size=50))
+def test_frame_roundtrip_property(ints, raw_tags):
+    """Wire-frame codec (frame/encode.go layout): arbitrary int64 + bytes
+    columns with nulls round-trip."""
+    from banyandb_amd import frame as fr
+    n = min(len(ints), len(raw_tags))
+    ints = ints[:n]
+    tags = raw_tags[:n]
+    fb = fr.FrameBuilder(n)
+    fb.add_i64(6, "value", "", ints)
+    fb.add_bytes(2, "tag", "default", tags)
+    blob = fb.finish()
+    rd = fr.FrameReader(blob)
+    assert rd.nrows == n and rd.ncols == 2
+    assert rd.col_i64(0) == ints
+    assert rd.col_var(1) == tags
