@@ -3014,9 +3014,10 @@ extern "C" int bydb_consume_multi(bydb_session *s, int64_t min_ts,
         int rthreads = 256;
         int rblocks = (int)((s->n_blocks + rthreads - 1) / rthreads);
         for (int i = 0; i < n_preds; i++) {
-            HIP_TRY(s, hipMemcpyAsync(s->d_pred_bytes + off, preds_in[i],
-                                      pred_lens[i], hipMemcpyHostToDevice,
-                                      s->stream));
+            if (pred_lens[i])   // disabled slots may pass a NULL pointer
+                HIP_TRY(s, hipMemcpyAsync(s->d_pred_bytes + off, preds_in[i],
+                                          pred_lens[i], hipMemcpyHostToDevice,
+                                          s->stream));
             offs[i] = off;
             lens3[i] = pred_lens[i];
             off += pred_lens[i];
