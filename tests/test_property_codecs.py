@@ -189,3 +189,33 @@ def test_frame_roundtrip_property(ints, raw_tags):
     assert rd.nrows == n and rd.ncols == 2
     assert rd.col_i64(0) == ints
     assert rd.col_var(1) == tags
+
+
+@settings(max_examples=60, deadline=None)
+@given(st.lists(st.lists(SMALL, min_size=1, max_size=60), min_size=2,
+                max_size=5))
+def test_reduce_combine_equals_single_fold(shards):
+    """AggModeMap per-shard partials + Reduce combine == folding the
+    concatenation once (aggregation_reduce.go:120-138 semantics)."""
+    import banyandb_amd as ba
+    parts = []
+    for vals in shards:
+        p = ba.Partial()
+        p.sum_i = sum(vals) % 2**64
+        p.sum_i = p.sum_i - 2**64 if p.sum_i >= 2**63 else p.sum_i
+        p.count = len(vals)
+        p.min_i = min(vals)
+        p.max_i = max(vals)
+        p.sum_f = 0.0
+        parts.append(p)
+    res = ba.reduce_partials(parts, len(shards), 1, ba.VT_INT64)[0]
+    flat = [v for vals in shards for v in vals]
+    s = sum(flat)
+    s_w = (s + 2**63) % 2**64 - 2**63
+    assert res.count == len(flat)
+    assert res.sum_i == s_w
+    assert res.min_i == min(flat)
+    assert res.max_i == max(flat)
+    # mean: sum/count with the >=1 clamp (the clamp makes Go's truncating
+    # division and Python's floor division agree on every case)
+    assert res.mean_i == max(1, s_w // len(flat))
