@@ -30,6 +30,7 @@
 #include <cstdio>
 #include <cstring>
 #include <dlfcn.h>
+#include <map>
 #include <string>
 #include <sys/stat.h>
 #include <vector>
@@ -377,20 +378,24 @@ extern "C" int bydb_part_read_dir(bydb_part_builder *b, const char *path) {
         !read_file(dir + "/timestamps.bin", ts_bin) ||
         !read_file(dir + "/fv.bin", fv_bin))
         return BYDB_ERR;
-    // tag families: discover any "<fam>.tf"/".tfm" next to the core files
-    // (single family supported, matching the writer)
-    std::vector<uint8_t> tf_bin, tfm_bin;
-    std::string fam;
-    {
-        // writer uses one family; try "default" then scan metadata-free
-        // common names by probing the directory via the .tfm extension
-        // (portable approach: caller passes family via env or we try
-        // "default"): read_dir keeps it simple — "default".
-        if (read_file(dir + "/default.tfm", tfm_bin)) {
-            fam = "default";
-            read_file(dir + "/default.tf", tf_bin);
-        }
-    }
+    // tag families are loaded lazily by the NAME each blockMetadata record
+    // carries (block_metadata.go:129-147): a part written with any family
+    // name round-trips, and a block referencing a family whose files are
+    // missing is a loud error, never a silent tag drop.
+    struct FamFiles {
+        std::vector<uint8_t> tfm, tf;
+    };
+    std::map<std::string, FamFiles> fams;
+    auto load_family = [&](const std::string &name) -> FamFiles * {
+        auto it = fams.find(name);
+        if (it != fams.end()) return it->second.tfm.empty() ? nullptr
+                                                            : &it->second;
+        FamFiles &ff = fams[name];
+        if (!read_file(dir + "/" + name + ".tfm", ff.tfm) ||
+            !read_file(dir + "/" + name + ".tf", ff.tf) || ff.tfm.empty())
+            return nullptr;
+        return &ff;
+    };
     // decompress meta.bin -> primaryBlockMetadata records
     unsigned long long msz = z_getsize(meta_bin.data(), meta_bin.size());
     std::vector<uint8_t> meta_records(msz);
@@ -431,14 +436,19 @@ extern "C" int bydb_part_read_dir(bydb_part_builder *b, const char *path) {
             d.version_first = (int64_t)rd_u64be(p + pos);
             pos += 8;
             d.version_enc = p[pos++];
-            // tag families
+            // tag families (single family per block supported, matching
+            // the writer; more is a loud error, not a partial read)
             uint64_t nfam;
             if (!rd_varu(p, len, &pos, &nfam)) return BYDB_ERR_BAD_DATA;
+            if (nfam > 1) return BYDB_ERR_BAD_DATA;
             uint64_t tfm_off = 0, tfm_size = 0;
+            std::string blk_fam;
             for (uint64_t f = 0; f < nfam; f++) {
                 uint64_t nl;
                 if (!rd_varu(p, len, &pos, &nl)) return BYDB_ERR_BAD_DATA;
-                pos += nl;  // family name
+                if (pos + nl > len) return BYDB_ERR_BAD_DATA;
+                blk_fam.assign((const char *)p + pos, nl);
+                pos += nl;
                 if (!rd_varu(p, len, &pos, &tfm_off)) return BYDB_ERR_BAD_DATA;
                 if (!rd_varu(p, len, &pos, &tfm_size)) return BYDB_ERR_BAD_DATA;
             }
@@ -487,27 +497,31 @@ extern "C" int bydb_part_read_dir(bydb_part_builder *b, const char *path) {
             d.field_off = bydb_part_builder_payload_len(b);
             d.field_len = fv_size - hp;
             bydb_part_builder_append_raw(b, fp + hp, d.field_len);
-            // -- tag columns from the family's cfm record
-            if (tfm_size > 0 && tfm_off + tfm_size <= tfm_bin.size()) {
-                const uint8_t *cp = tfm_bin.data() + tfm_off;
+            // -- tag columns from the named family's cfm record
+            if (tfm_size > 0) {
+                FamFiles *ff = load_family(blk_fam);
+                if (!ff || tfm_off + tfm_size > ff->tfm.size())
+                    return BYDB_ERR_BAD_DATA;
+                const uint8_t *cp = ff->tfm.data() + tfm_off;
                 size_t cl = tfm_size, cpos = 0;
                 uint64_t ncols2;
-                if (rd_varu(cp, cl, &cpos, &ncols2)) {
-                    for (uint64_t c = 0; c < ncols2 && c < 3; c++) {
-                        uint64_t nl;
-                        if (!rd_varu(cp, cl, &cpos, &nl)) break;
-                        cpos += nl;
-                        cpos += 1;  // valueType
-                        uint64_t toff, tsize;
-                        if (!rd_varu(cp, cl, &cpos, &toff)) break;
-                        if (!rd_varu(cp, cl, &cpos, &tsize)) break;
-                        uint64_t dst_off = bydb_part_builder_payload_len(b);
-                        bydb_part_builder_append_raw(
-                            b, tf_bin.data() + toff, tsize);
-                        if (c == 0) { d.tag_off = dst_off; d.tag_len = tsize; }
-                        else if (c == 1) { d.tag2_off = dst_off; d.tag2_len = tsize; }
-                        else { d.tag3_off = dst_off; d.tag3_len = tsize; }
-                    }
+                if (!rd_varu(cp, cl, &cpos, &ncols2))
+                    return BYDB_ERR_BAD_DATA;
+                for (uint64_t c = 0; c < ncols2 && c < 3; c++) {
+                    uint64_t nl;
+                    if (!rd_varu(cp, cl, &cpos, &nl)) return BYDB_ERR_BAD_DATA;
+                    cpos += nl;
+                    cpos += 1;  // valueType
+                    uint64_t toff, tsize;
+                    if (!rd_varu(cp, cl, &cpos, &toff)) return BYDB_ERR_BAD_DATA;
+                    if (!rd_varu(cp, cl, &cpos, &tsize)) return BYDB_ERR_BAD_DATA;
+                    if (toff + tsize > ff->tf.size()) return BYDB_ERR_BAD_DATA;
+                    uint64_t dst_off = bydb_part_builder_payload_len(b);
+                    bydb_part_builder_append_raw(
+                        b, ff->tf.data() + toff, tsize);
+                    if (c == 0) { d.tag_off = dst_off; d.tag_len = tsize; }
+                    else if (c == 1) { d.tag2_off = dst_off; d.tag2_len = tsize; }
+                    else { d.tag3_off = dst_off; d.tag3_len = tsize; }
                 }
             }
             bydb_part_builder_append_desc(b, &d);

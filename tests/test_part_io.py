@@ -124,3 +124,36 @@ def test_part_dir_roundtrip_nullable_and_plain_tag():
     assert got_nopred.min_i == ref_nopred.min_i
     assert got.count == ref.count
     assert got.sum_i == ref.sum_i
+
+
+def test_part_dir_roundtrip_custom_family_name():
+    """A part written with a non-"default" tag family name round-trips
+    with its tags intact: the reader opens <fam>.tfm/.tf by the family
+    name each blockMetadata record carries (block_metadata.go:129-147),
+    not a hard-coded probe."""
+    b = build_part()
+    orc0p = oracle_scan(b, ba.VT_INT64, pred=b"dev")[0]
+    with tempfile.TemporaryDirectory() as td:
+        p = os.path.join(td, "0000000000000002")
+        b.write_dir(p, tag_family="searchable", tag_names=["env", "region"])
+        assert os.path.exists(os.path.join(p, "searchable.tfm"))
+        assert not os.path.exists(os.path.join(p, "default.tfm"))
+        b2 = ba.PartBuilder()
+        b2.read_dir(p)
+        orc1p = oracle_scan(b2, ba.VT_INT64, pred=b"dev")[0]
+        assert orc1p.count > 0
+        assert (orc1p.count, orc1p.sum_i) == (orc0p.count, orc0p.sum_i)
+
+
+def test_part_dir_missing_family_files_is_loud():
+    """A block that references a tag family whose .tfm/.tf files are
+    absent must error (BYDB_ERR_BAD_DATA), never silently drop the tag
+    columns."""
+    b = build_part()
+    with tempfile.TemporaryDirectory() as td:
+        p = os.path.join(td, "0000000000000003")
+        b.write_dir(p, tag_family="searchable", tag_names=["env", "region"])
+        os.remove(os.path.join(p, "searchable.tfm"))
+        b2 = ba.PartBuilder()
+        with pytest.raises(RuntimeError):
+            b2.read_dir(p)
