@@ -15,6 +15,7 @@
 // oracle in tests/test_product_encoder.py.
 #include "../../include/bydb_gpu.h"
 
+#include <algorithm>
 #include <cmath>
 #include <cstdio>
 #include <cstdlib>
@@ -596,6 +597,13 @@ struct bydb_part_builder {
     std::vector<int64_t> scratch_i64;
     std::vector<double> scratch_f64;
     std::vector<int64_t> scratch_ts, scratch_ver;
+    // duplicate-(sid,ts) fold bookkeeping for the LAST added block
+    // (part.go:192-198): when rows were dropped, last_keep holds the
+    // surviving source indices so set_block_tag can fold its per-row tag
+    // values the same way (the reference's dps.skip drops the
+    // tagFamilies row with the datapoint)
+    std::vector<int64_t> last_keep;
+    int64_t last_orig_n = 0;
 };
 
 extern "C" bydb_part_builder *bydb_part_builder_create(void) {
@@ -685,7 +693,43 @@ static int add_block_common(bydb_part_builder *b, uint64_t series_id,
     d.field_off = b->base_off + fstart;
     d.field_len = b->payload.size() - fstart;
     b->blocks.push_back(d);
+    b->last_keep.clear();
+    b->last_orig_n = n;
     return BYDB_OK;
+}
+
+// Duplicate-(sid,ts) fold at part build, exactly as the reference's
+// mustInitFromDataPoints does (part.go:178 sort + :192-198 skip loop with
+// datapoints.go:189-197 Less): rows order by (ts asc, version desc) and
+// only the FIRST row of each timestamp — the highest version — survives.
+// Returns false when the input is already strictly-ascending unique (the
+// common case; no work).  keep holds the surviving source indices in
+// write order.  (Not reproduced: the reference's tsPrev==0 initialization
+// would also drop a first row whose timestamp is exactly 0 — a
+// 1970-epoch corner with no real inputs.)
+static bool dedup_block_rows(const int64_t *ts, const int64_t *versions,
+                             int64_t n, std::vector<int64_t> &keep) {
+    bool sorted_unique = true;
+    for (int64_t i = 1; i < n; i++)
+        if (ts[i] <= ts[i - 1]) { sorted_unique = false; break; }
+    if (sorted_unique) return false;
+    keep.resize((size_t)n);
+    for (int64_t i = 0; i < n; i++) keep[(size_t)i] = i;
+    std::stable_sort(keep.begin(), keep.end(), [&](int64_t a, int64_t c) {
+        if (ts[a] != ts[c]) return ts[a] < ts[c];
+        return versions[a] > versions[c];
+    });
+    size_t w = 0;
+    int64_t prev_ts = 0;
+    for (size_t i = 0; i < keep.size(); i++) {
+        int64_t src = keep[i];
+        if (w == 0 || ts[src] != prev_ts) {
+            keep[w++] = src;
+            prev_ts = ts[src];
+        }
+    }
+    keep.resize(w);
+    return true;
 }
 
 // Null-bearing float64 column: 8-B big-endian IEEE-754 cells
@@ -703,6 +747,28 @@ extern "C" int bydb_part_builder_add_block_f64_nullable(
     if (n < 1 || n > 8192) {
         b->err = "block row count out of range";
         return BYDB_ERR_BAD_ARG;
+    }
+    {
+        std::vector<int64_t> keep;
+        if (dedup_block_rows(ts, versions, n, keep)) {
+            std::vector<int64_t> ts2, ver2;
+            std::vector<double> vv2;
+            std::vector<uint8_t> va2;
+            for (int64_t src : keep) {
+                ts2.push_back(ts[src]);
+                ver2.push_back(versions[src]);
+                vv2.push_back(vals[src]);
+                va2.push_back(valid[src]);
+            }
+            int rc = bydb_part_builder_add_block_f64_nullable(
+                b, series_id, ts2.data(), ver2.data(), vv2.data(),
+                va2.data(), (int64_t)keep.size(), group_code);
+            if (rc == BYDB_OK) {
+                b->last_keep = keep;
+                b->last_orig_n = n;
+            }
+            return rc;
+        }
     }
     // ts/version/desc handling matches add_block_common; the field stream
     // is [ENC_PLAIN][bytes block of float cells]
@@ -753,6 +819,27 @@ extern "C" int bydb_part_builder_add_block_i64_nullable(
         b->err = "valid mask required";
         return BYDB_ERR_BAD_ARG;
     }
+    {
+        std::vector<int64_t> keep;
+        if (dedup_block_rows(ts, versions, n, keep)) {
+            std::vector<int64_t> ts2, ver2, vv2;
+            std::vector<uint8_t> va2;
+            for (int64_t src : keep) {
+                ts2.push_back(ts[src]);
+                ver2.push_back(versions[src]);
+                vv2.push_back(vals[src]);
+                va2.push_back(valid[src]);
+            }
+            int rc = bydb_part_builder_add_block_i64_nullable(
+                b, series_id, ts2.data(), ver2.data(), vv2.data(),
+                va2.data(), (int64_t)keep.size(), group_code);
+            if (rc == BYDB_OK) {
+                b->last_keep = keep;
+                b->last_orig_n = n;
+            }
+            return rc;
+        }
+    }
     return add_block_common(b, series_id, ts, versions, vals, 0,
                             BYDB_VT_INT64, n, group_code, valid);
 }
@@ -763,6 +850,25 @@ extern "C" int bydb_part_builder_add_block_i64(bydb_part_builder *b,
                                                const int64_t *versions,
                                                const int64_t *vals, int64_t n,
                                                uint32_t group_code) {
+    {
+        std::vector<int64_t> keep;
+        if (dedup_block_rows(ts, versions, n, keep)) {
+            std::vector<int64_t> ts2, ver2, vv2;
+            for (int64_t src : keep) {
+                ts2.push_back(ts[src]);
+                ver2.push_back(versions[src]);
+                vv2.push_back(vals[src]);
+            }
+            int rc = bydb_part_builder_add_block_i64(
+                b, series_id, ts2.data(), ver2.data(), vv2.data(),
+                (int64_t)keep.size(), group_code);
+            if (rc == BYDB_OK) {
+                b->last_keep = keep;
+                b->last_orig_n = n;
+            }
+            return rc;
+        }
+    }
     return add_block_common(b, series_id, ts, versions, vals, 0, BYDB_VT_INT64,
                             n, group_code);
 }
@@ -773,6 +879,26 @@ extern "C" int bydb_part_builder_add_block_f64(bydb_part_builder *b,
                                                const int64_t *versions,
                                                const double *vals, int64_t n,
                                                uint32_t group_code) {
+    {
+        std::vector<int64_t> keep;
+        if (dedup_block_rows(ts, versions, n, keep)) {
+            std::vector<int64_t> ts2, ver2;
+            std::vector<double> vv2;
+            for (int64_t src : keep) {
+                ts2.push_back(ts[src]);
+                ver2.push_back(versions[src]);
+                vv2.push_back(vals[src]);
+            }
+            int rc = bydb_part_builder_add_block_f64(
+                b, series_id, ts2.data(), ver2.data(), vv2.data(),
+                (int64_t)keep.size(), group_code);
+            if (rc == BYDB_OK) {
+                b->last_keep = keep;
+                b->last_orig_n = n;
+            }
+            return rc;
+        }
+    }
     b->scratch_i64.resize((size_t)n);
     int16_t exp;
     if (!floats_to_decimal_list(vals, n, b->scratch_i64.data(), &exp)) {
@@ -792,6 +918,31 @@ extern "C" int bydb_part_builder_set_block_tag(bydb_part_builder *b,
     }
     bydb_block_desc &d = b->blocks.back();
     if ((int64_t)d.count != n) {
+        // If the last block folded duplicate-(sid,ts) rows (part.go:
+        // 192-198), fold this per-row tag list the same way: the
+        // reference's dps.skip drops the tagFamilies row with the
+        // datapoint, so the caller may pass the ORIGINAL row list.
+        if (n == b->last_orig_n &&
+            (int64_t)b->last_keep.size() == (int64_t)d.count) {
+            std::vector<int64_t> offs((size_t)n);
+            int64_t o = 0;
+            for (int64_t i = 0; i < n; i++) {
+                offs[(size_t)i] = o;
+                if (lens[i] > 0) o += lens[i];
+            }
+            std::vector<uint8_t> data2;
+            std::vector<int64_t> lens2;
+            for (int64_t src : b->last_keep) {
+                lens2.push_back(lens[src]);
+                if (lens[src] > 0)
+                    data2.insert(data2.end(), data + offs[(size_t)src],
+                                 data + offs[(size_t)src] + lens[src]);
+            }
+            static const uint8_t kEmpty = 0;
+            return bydb_part_builder_set_block_tag(
+                b, data2.empty() ? &kEmpty : data2.data(), lens2.data(),
+                (int64_t)lens2.size());
+        }
         b->err = "tag row count mismatch";
         return BYDB_ERR_BAD_ARG;
     }
