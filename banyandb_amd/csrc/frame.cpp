@@ -421,3 +421,219 @@ extern "C" int bydb_top_groups(const bydb_result *results, int64_t n_groups,
     *out_n = n;
     return BYDB_OK;
 }
+
+// ---- AggModeReduce over raw map frames with replica dedup ----
+// Restates BatchAggregation's reduce path (pkg/query/vectorized/measure/
+// aggregation_reduce.go): dedup key = LE64(shard_id) ++ group_key
+// (markDedupSeen :83-102; no shard column -> shard 0, i.e. dedup on the
+// group key alone — the same fallback the reference takes when
+// shardIDIdx == -1); the first row per (shard, key) pair Combines
+// (combinePartial :104-138), null value rows skip, MEAN's count sidecar
+// is the column named "<value_name>__agg_count" found by walking forward
+// from the value column (locateMeanCounts :46-60 + meanCountSuffix,
+// aggregation.go:96); Val() finalizes with meanReduce's >=1 clamp
+// (pkg/query/aggregation/function.go:62-71) and min/max's sentinel-aware
+// Combine (:224-228).  Groups emit in first-seen order across frames.
+#include <unordered_map>
+#include <unordered_set>
+
+extern "C" int bydb_reduce_frames(
+    const uint8_t *const *frames, const uint64_t *frame_lens, int n_frames,
+    int shard_col, const int32_t *key_cols, int n_key_cols,
+    const bydb_reduce_spec *specs, int n_specs, int64_t out_cap,
+    int64_t *out_i64, double *out_f64, uint8_t *key_buf,
+    uint64_t key_buf_cap, uint64_t *key_offs, int64_t *out_n_groups) {
+    if (n_frames < 0 || n_key_cols < 1 || n_specs < 1 || !specs || !key_cols)
+        return BYDB_ERR_BAD_ARG;
+    struct Slot {
+        int64_t iv = 0, icnt = 0;
+        double fv = 0, fcnt = 0;
+        bool seen = false;
+    };
+    struct Group {
+        std::vector<Slot> slots;
+    };
+    std::vector<Group> groups;
+    std::vector<std::string> group_keys;          // first-seen order
+    std::unordered_map<std::string, int64_t> group_idx;
+    std::unordered_set<std::string> dedup_seen;   // LE64(shard) ++ key
+    std::vector<bool> spec_float(n_specs, false);
+    bool spec_float_known = false;
+
+    for (int fi = 0; fi < n_frames; fi++) {
+        bydb_frame_reader *r = bydb_frame_open(frames[fi], frame_lens[fi]);
+        if (!r->err.empty()) { bydb_frame_close(r); return BYDB_ERR_BAD_DATA; }
+        const uint64_t nrows = r->nrows;
+        auto colok = [&](int32_t c) { return c >= 0 && (uint64_t)c < r->ncols; };
+        if ((shard_col >= 0 && !colok(shard_col)))
+            { bydb_frame_close(r); return BYDB_ERR_BAD_ARG; }
+        for (int k = 0; k < n_key_cols; k++)
+            if (!colok(key_cols[k])) { bydb_frame_close(r); return BYDB_ERR_BAD_ARG; }
+        // value columns + MEAN count sidecars (by name walk-forward)
+        std::vector<int32_t> count_col(n_specs, -1);
+        for (int si = 0; si < n_specs; si++) {
+            if (!colok(specs[si].input_col) ||
+                (r->cols[specs[si].input_col].type != 1 &&
+                 r->cols[specs[si].input_col].type != 2))
+                { bydb_frame_close(r); return BYDB_ERR_BAD_ARG; }
+            bool isf = r->cols[specs[si].input_col].type == 2;
+            if (!spec_float_known) spec_float[si] = isf;
+            else if (spec_float[si] != isf)
+                { bydb_frame_close(r); return BYDB_ERR_BAD_DATA; }
+            if (specs[si].func == BYDB_AGG_MEAN) {
+                std::string want = r->cols[specs[si].input_col].name +
+                                   "__agg_count";
+                for (uint64_t j = specs[si].input_col + 1; j < r->ncols; j++)
+                    if (r->cols[j].name == want) { count_col[si] = (int32_t)j; break; }
+            }
+        }
+        spec_float_known = true;
+        // per-row numeric readers (i64 raw bits reinterpreted per type)
+        auto num_i = [&](int32_t ci, uint64_t row) -> int64_t {
+            const uint8_t *p = r->buf.data() + r->cols[ci].data_off + row * 8;
+            int64_t v; memcpy(&v, p, 8);
+            if (r->cols[ci].type == 2) { double d; memcpy(&d, p, 8); v = (int64_t)d; }
+            return v;
+        };
+        auto num_f = [&](int32_t ci, uint64_t row) -> double {
+            const uint8_t *p = r->buf.data() + r->cols[ci].data_off + row * 8;
+            if (r->cols[ci].type == 2) { double d; memcpy(&d, p, 8); return d; }
+            int64_t v; memcpy(&v, p, 8); return (double)v;
+        };
+        // var-width cell offsets for key columns
+        for (uint64_t row = 0; row < nrows; row++) {
+            // group key: per component, 0x00 for null else 0x01 ++
+            // uvarint(len) ++ bytes (strings) / 8-B LE (numerics) —
+            // appendKeyComponent semantics (groupby.go:287-364)
+            std::string key;
+            bool bad = false;
+            for (int k = 0; k < n_key_cols && !bad; k++) {
+                int32_t ci = key_cols[k];
+                if (bydb_frame_col_null(r, ci, row)) { key.push_back('\0'); continue; }
+                key.push_back('\1');
+                auto &c = r->cols[ci];
+                if (c.type == 1 || c.type == 2) {
+                    const uint8_t *p = r->buf.data() + c.data_off + row * 8;
+                    key.append((const char *)p, 8);
+                } else {
+                    // walk the var cells to this row (frames are small —
+                    // group-cardinality rows)
+                    const uint8_t *p = r->buf.data();
+                    size_t pos = c.data_off;
+                    uint64_t l = 0;
+                    for (uint64_t j = 0; j <= row; j++) {
+                        if (!rd_uvarint(p, r->buf.size(), &pos, &l)) { bad = true; break; }
+                        if (j < row) pos += l;
+                    }
+                    if (!bad) {
+                        uint8_t lenb[10]; size_t ln = 0; uint64_t u = l;
+                        while (u > 0x7f) { lenb[ln++] = (uint8_t)(0x80 | (u & 0x7f)); u >>= 7; }
+                        lenb[ln++] = (uint8_t)u;
+                        key.append((const char *)lenb, ln);
+                        key.append((const char *)p + pos, l);
+                    }
+                }
+            }
+            if (bad) { bydb_frame_close(r); return BYDB_ERR_BAD_DATA; }
+            // replica dedup on (shard_id, key)
+            int64_t shard = 0;
+            if (shard_col >= 0 && !bydb_frame_col_null(r, shard_col, row))
+                shard = num_i(shard_col, row);
+            std::string dk;
+            dk.reserve(8 + key.size());
+            for (int bshift = 0; bshift < 8; bshift++)
+                dk.push_back((char)(uint8_t)((uint64_t)shard >> (8 * bshift)));
+            dk += key;
+            if (!dedup_seen.insert(dk).second) continue;   // replica: drop
+            // group slot (first-seen order)
+            auto it = group_idx.find(key);
+            int64_t g;
+            if (it == group_idx.end()) {
+                g = (int64_t)groups.size();
+                group_idx.emplace(key, g);
+                groups.push_back(Group{std::vector<Slot>((size_t)n_specs)});
+                group_keys.push_back(key);
+            } else {
+                g = it->second;
+            }
+            // Combine (aggregation_reduce.go:104-138)
+            for (int si = 0; si < n_specs; si++) {
+                int32_t ci = specs[si].input_col;
+                if (bydb_frame_col_null(r, ci, row)) continue;  // null skip
+                Slot &sl = groups[(size_t)g].slots[(size_t)si];
+                if (spec_float[si]) {
+                    double v = num_f(ci, row);
+                    switch (specs[si].func) {
+                    case BYDB_AGG_SUM: case BYDB_AGG_COUNT: sl.fv += v; break;
+                    case BYDB_AGG_MIN: if (!sl.seen || v < sl.fv) sl.fv = v; break;
+                    case BYDB_AGG_MAX: if (!sl.seen || v > sl.fv) sl.fv = v; break;
+                    case BYDB_AGG_MEAN:
+                        sl.fv += v;
+                        if (count_col[si] >= 0) sl.fcnt += num_f(count_col[si], row);
+                        break;
+                    default: bydb_frame_close(r); return BYDB_ERR_BAD_ARG;
+                    }
+                } else {
+                    int64_t v = num_i(ci, row);
+                    switch (specs[si].func) {
+                    case BYDB_AGG_SUM: case BYDB_AGG_COUNT:
+                        sl.iv = (int64_t)((uint64_t)sl.iv + (uint64_t)v); break;
+                    case BYDB_AGG_MIN: if (!sl.seen || v < sl.iv) sl.iv = v; break;
+                    case BYDB_AGG_MAX: if (!sl.seen || v > sl.iv) sl.iv = v; break;
+                    case BYDB_AGG_MEAN:
+                        sl.iv = (int64_t)((uint64_t)sl.iv + (uint64_t)v);
+                        if (count_col[si] >= 0) sl.icnt += num_i(count_col[si], row);
+                        break;
+                    default: bydb_frame_close(r); return BYDB_ERR_BAD_ARG;
+                    }
+                }
+                sl.seen = true;
+            }
+        }
+        bydb_frame_close(r);
+    }
+    // finalize (Val)
+    int64_t ng = (int64_t)groups.size();
+    if (ng > out_cap) return BYDB_ERR_OOM;
+    uint64_t ko = 0;
+    for (int64_t g = 0; g < ng; g++) {
+        if (key_offs) key_offs[g] = ko;
+        if (key_buf) {
+            if (ko + group_keys[(size_t)g].size() > key_buf_cap)
+                return BYDB_ERR_OOM;
+            memcpy(key_buf + ko, group_keys[(size_t)g].data(),
+                   group_keys[(size_t)g].size());
+        }
+        ko += group_keys[(size_t)g].size();
+        for (int si = 0; si < n_specs; si++) {
+            Slot &sl = groups[(size_t)g].slots[(size_t)si];
+            int64_t oi = 0; double of = 0;
+            switch (specs[si].func) {
+            case BYDB_AGG_SUM: case BYDB_AGG_COUNT:
+                oi = sl.iv; of = sl.fv; break;
+            case BYDB_AGG_MIN:
+                oi = sl.seen ? sl.iv : INT64_MAX;
+                of = sl.seen ? sl.fv : 1.7976931348623157e308; break;
+            case BYDB_AGG_MAX:
+                oi = sl.seen ? sl.iv : INT64_MIN;
+                of = sl.seen ? sl.fv : -1.7976931348623157e308; break;
+            case BYDB_AGG_MEAN:
+                // meanReduce.Val (function.go:62-71): count 0 -> 0,
+                // else sum/count clamped to >= 1
+                if (spec_float[si]) {
+                    of = sl.fcnt == 0 ? 0
+                         : (sl.fv / sl.fcnt < 1 ? 1 : sl.fv / sl.fcnt);
+                } else {
+                    oi = sl.icnt == 0 ? 0
+                         : (sl.iv / sl.icnt < 1 ? 1 : sl.iv / sl.icnt);
+                }
+                break;
+            }
+            if (out_i64) out_i64[g * n_specs + si] = oi;
+            if (out_f64) out_f64[g * n_specs + si] = of;
+        }
+    }
+    if (key_offs) key_offs[ng] = ko;
+    *out_n_groups = ng;
+    return BYDB_OK;
+}

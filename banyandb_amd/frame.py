@@ -173,3 +173,55 @@ class FrameReader:
         if getattr(self, "_h", None):
             _l.bydb_frame_close(self._h)
             self._h = None
+
+
+class ReduceSpec(C.Structure):
+    # mirrors bydb_reduce_spec (include/bydb_gpu.h)
+    _fields_ = [("input_col", C.c_int32), ("func", C.c_int32)]
+
+
+_l.bydb_reduce_frames.restype = C.c_int
+_l.bydb_reduce_frames.argtypes = [
+    C.POINTER(u8p), C.POINTER(C.c_uint64), C.c_int,
+    C.c_int, C.POINTER(C.c_int32), C.c_int,
+    C.POINTER(ReduceSpec), C.c_int,
+    C.c_int64, i64p, f64p,
+    u8p, C.c_uint64, C.POINTER(C.c_uint64), i64p]
+
+
+def reduce_frames(frames, specs, key_cols, shard_col=-1, out_cap=65536,
+                  key_buf_cap=1 << 22):
+    """Product-side AggModeReduce with replica dedup
+    (aggregation_reduce.go:83-138) over raw map frames.
+
+    frames: list of frame bytes; specs: list of (input_col, func);
+    key_cols: group-key column indices; shard_col: RoleShardID column
+    (or -1 to dedup on the group key alone).  Returns a list of
+    (key_bytes, [per-spec (int_final, float_final)]) in first-seen group
+    order."""
+    n = len(frames)
+    bufs = [(C.c_uint8 * max(len(f), 1)).from_buffer_copy(f or b"\0")
+            for f in frames]
+    fptrs = (u8p * n)(*[C.cast(b, u8p) for b in bufs])
+    flens = (C.c_uint64 * n)(*[len(f) for f in frames])
+    sarr = (ReduceSpec * len(specs))(*[ReduceSpec(c, f) for c, f in specs])
+    karr = (C.c_int32 * len(key_cols))(*key_cols)
+    out_i = (C.c_int64 * (out_cap * len(specs)))()
+    out_f = (C.c_double * (out_cap * len(specs)))()
+    kbuf = (C.c_uint8 * key_buf_cap)()
+    koffs = (C.c_uint64 * (out_cap + 1))()
+    ng = C.c_int64()
+    rc = _l.bydb_reduce_frames(fptrs, flens, n, shard_col, karr,
+                               len(key_cols), sarr, len(specs), out_cap,
+                               out_i, out_f, kbuf, key_buf_cap, koffs,
+                               C.byref(ng))
+    if rc != 0:
+        raise RuntimeError(f"bydb_reduce_frames rc={rc}")
+    raw = bytes(bytearray(kbuf))
+    out = []
+    for g in range(ng.value):
+        key = raw[koffs[g]: koffs[g + 1]]
+        vals = [(out_i[g * len(specs) + s], out_f[g * len(specs) + s])
+                for s in range(len(specs))]
+        out.append((key, vals))
+    return out

@@ -189,11 +189,13 @@ int bydb_finalize_partials(bydb_session *s, bydb_partial *out, int64_t n_groups)
 /* Reset accumulators to the fold identity (Map.Reset, function.go). */
 int bydb_reset(bydb_session *s);
 
-/* Combine external partials (AggModeReduce semantics,
- * aggregation_reduce.go:120-138) into results on the host. */
-int bydb_reduce_partials(const bydb_partial *parts, int64_t n_parts_per_group,
-                         int64_t n_groups, int field_vtype, uint32_t func_mask,
-                         bydb_result *out);
+/* Combine external partials (AggModeReduce Combine semantics,
+ * aggregation_reduce.go:120-138) into results on the host.  parts is
+ * [n_parts_per_group][n_groups]; float_exp selects the float64 restore
+ * domain (BYDB_FLOAT_RAW_EXP = raw IEEE-754 cells). */
+int bydb_reduce_partials2(const bydb_partial *parts, int64_t n_parts_per_group,
+                          int64_t n_groups, int field_vtype, int16_t float_exp,
+                          bydb_result *out);
 
 /* Timing of the last consume's kernels in milliseconds (HIP events on the
  * session stream) — feeds bench.py's roofline.achieved. */
@@ -263,6 +265,138 @@ int bydb_gen_series_f64(bydb_part_builder *b, uint64_t series_index,
                         int64_t n_dp, int64_t t0, int64_t stride_ns,
                         double base, double ramp, uint64_t seed,
                         uint32_t group_code);
+/* bulk generators: n_series consecutive series (first_index ...) encoded
+ * on `threads` host threads; base = series_index * base_step.
+ * group_mod > 0 assigns group_code = series_index %% group_mod. */
+int bydb_gen_series_bulk_i64(bydb_part_builder *b, uint64_t first_index,
+                             int64_t n_series, int64_t n_dp, int64_t t0,
+                             int64_t stride_ns, int64_t base_step, int64_t ramp,
+                             uint64_t seed, uint32_t group_mod, int threads);
+int bydb_gen_series_bulk_f64(bydb_part_builder *b, uint64_t first_index,
+                             int64_t n_series, int64_t n_dp, int64_t t0,
+                             int64_t stride_ns, double base_step, double ramp,
+                             uint64_t seed, uint32_t group_mod, int threads);
+/* entity-tag table for the generators: every generated block gets
+ * values[series_index %% n_values] as a constant tag column in `slot`.
+ * data = concatenated values; lens[i] < 0 marks a nil value. */
+int bydb_part_builder_set_tag_table(bydb_part_builder *b, int slot,
+                                    const uint8_t *data, const int64_t *lens,
+                                    int64_t n_values);
+/* raw helpers used by the part reader */
+int bydb_part_builder_append_raw(bydb_part_builder *b, const uint8_t *data,
+                                 uint64_t len);
+int bydb_part_builder_append_desc(bydb_part_builder *b,
+                                  const bydb_block_desc *d);
+
+/* ---- per-row group-by on dictionary tag columns ----
+ * computeKey/appendKeyComponent semantics (vectorized/measure/
+ * aggregation.go:523 + groupby.go:287-364) with a host-supplied group
+ * DOMAIN: dense group id = index into the domain value list (the host
+ * layer controls materialisation order; the reference's is first-seen).
+ * Nil or out-of-domain rows drop.  dom_blob = concatenated values,
+ * dom_offs[n_values+1] prefix offsets. */
+int bydb_agg_configure_by_tag(bydb_session *s, int field_vtype,
+                              uint32_t func_mask, int tag_slot,
+                              const uint8_t *dom_blob,
+                              const uint64_t *dom_offs, uint32_t n_values,
+                              int mode);
+/* composite key over up to 3 tag slots: gid = g0 + n0*g1 + n0*n1*g2 */
+int bydb_agg_configure_by_tags(bydb_session *s, int field_vtype,
+                               uint32_t func_mask, const int *slots,
+                               int n_slots, const uint8_t *const *dom_blobs,
+                               const uint64_t *const *dom_offs,
+                               const uint32_t *n_values, int mode);
+
+/* ---- on-disk part directory (Appendix A; banyand/measure/part.go:37-56)
+ * write_dir emits metadata.json / meta.bin / primary.bin / timestamps.bin
+ * / fv.bin / <family>.tfm/.tf in the reference's exact layout; read_dir
+ * loads a part directory (ours or the reference's) into a builder. */
+int bydb_part_write_dir(bydb_part_builder *b, const char *path,
+                        const char *field_name, const char *tag_family,
+                        const char *const *tag_names, int n_tags);
+int bydb_part_read_dir(bydb_part_builder *b, const char *path);
+
+/* ---- columnar wire frame codec (egress / Map->Reduce transport) ----
+ * Byte-stable layout per pkg/query/vectorized/frame (frame.go:26-95,
+ * encode.go:42-170): magic "\0VFR", version, uvarint nrows/ncols, per
+ * column [role, type, name, family, validity bitmap (1=null, LE
+ * bit-packed), data (numeric = N x 8B LE; var = uvarint len + bytes)]. */
+typedef struct bydb_frame_builder bydb_frame_builder;
+typedef struct bydb_frame_reader bydb_frame_reader;
+bydb_frame_builder *bydb_frame_builder_create(uint64_t nrows);
+void bydb_frame_builder_destroy(bydb_frame_builder *b);
+const char *bydb_frame_builder_error(bydb_frame_builder *b);
+int bydb_frame_add_i64(bydb_frame_builder *b, uint8_t role, const char *name,
+                       const char *family, const int64_t *vals,
+                       const uint8_t *nulls);
+int bydb_frame_add_f64(bydb_frame_builder *b, uint8_t role, const char *name,
+                       const char *family, const double *vals,
+                       const uint8_t *nulls);
+int bydb_frame_add_str(bydb_frame_builder *b, uint8_t role, const char *name,
+                       const char *family, const uint8_t *data,
+                       const int64_t *lens);
+int bydb_frame_add_bytes(bydb_frame_builder *b, uint8_t role, const char *name,
+                         const char *family, const uint8_t *data,
+                         const int64_t *lens);
+/* minimal TagValue/FieldValue proto cells (raw_emit.go egress shape);
+ * field_value selects FieldValue (type 6) over TagValue (type 5) */
+int bydb_frame_add_tagvalue_str(bydb_frame_builder *b, uint8_t role,
+                                const char *name, const char *family,
+                                const uint8_t *data, const int64_t *lens,
+                                int field_value);
+int bydb_frame_add_tagvalue_int(bydb_frame_builder *b, uint8_t role,
+                                const char *name, const char *family,
+                                const int64_t *vals, const uint8_t *nulls,
+                                int field_value);
+int bydb_frame_finish(bydb_frame_builder *b);
+uint64_t bydb_frame_len(bydb_frame_builder *b);
+const uint8_t *bydb_frame_data(bydb_frame_builder *b);
+bydb_frame_reader *bydb_frame_open(const uint8_t *data, uint64_t len);
+void bydb_frame_close(bydb_frame_reader *r);
+const char *bydb_frame_reader_error(bydb_frame_reader *r);
+uint64_t bydb_frame_nrows(bydb_frame_reader *r);
+uint64_t bydb_frame_ncols(bydb_frame_reader *r);
+int bydb_frame_col_info(bydb_frame_reader *r, uint64_t ci, uint8_t *role,
+                        uint8_t *type, char *name, uint64_t name_cap,
+                        char *family, uint64_t family_cap);
+int bydb_frame_col_null(bydb_frame_reader *r, uint64_t ci, uint64_t row);
+int bydb_frame_col_i64(bydb_frame_reader *r, uint64_t ci, int64_t *out);
+int bydb_frame_col_var(bydb_frame_reader *r, uint64_t ci, uint8_t *data_out,
+                       uint64_t data_cap, int64_t *lens_out,
+                       uint64_t *data_len);
+
+/* ---- BatchTop ordering over group results ----
+ * top.go:31-121 + ApplyTopToReduce (reduce.go:290): asc keeps the lowest
+ * k (smallest first), desc the highest k; ties keep insertion order;
+ * empty groups are nulls and sort lowest.  value_sel: 0 sum_i, 1 count,
+ * 2 min_i, 3 max_i, 4 mean_i, 5 sum_f, 6 min_f, 7 max_f, 8 mean_f. */
+int bydb_top_groups(const bydb_result *results, int64_t n_groups,
+                    int value_sel, int64_t k, int asc, int64_t *out_idx,
+                    int64_t *out_n);
+
+/* ---- AggModeReduce over raw map frames, with replica dedup ----
+ * Restates the liaison reduce (aggregation_reduce.go:83-138 +
+ * ReduceRawFrames, reduce.go:78): rows keyed by the key columns
+ * (appendKeyComponent semantics); the FIRST row per (shard_id, group_key)
+ * pair Combines, replica duplicates drop; MEAN count sidecars are located
+ * by the "__agg_count" name walk; Val() applies meanReduce's >=1 clamp.
+ * shard_col < 0 dedups on the group key alone (the reference's
+ * shardIDIdx == -1 fallback).  Outputs in first-seen group order:
+ * out_i64/out_f64 are [n_groups][n_specs] finals (int64 or float64
+ * meaningful per the value column's frame type); key_buf/key_offs
+ * (optional) receive the packed group keys.  Returns BYDB_OK and
+ * *out_n_groups, or BYDB_ERR_OOM when out_cap/key_buf_cap are small. */
+typedef struct {
+    int32_t input_col;   /* frame column index of the partial value */
+    int32_t func;        /* BYDB_AGG_* */
+} bydb_reduce_spec;
+int bydb_reduce_frames(const uint8_t *const *frames,
+                       const uint64_t *frame_lens, int n_frames,
+                       int shard_col, const int32_t *key_cols, int n_key_cols,
+                       const bydb_reduce_spec *specs, int n_specs,
+                       int64_t out_cap, int64_t *out_i64, double *out_f64,
+                       uint8_t *key_buf, uint64_t key_buf_cap,
+                       uint64_t *key_offs, int64_t *out_n_groups);
 
 #ifdef __cplusplus
 }

@@ -26,9 +26,6 @@ def test_so_exports_all_header_symbols():
     lib = ctypes.CDLL(SO)
     missing = []
     for sym in declared_symbols():
-        if sym in ("bydb_reduce_partials",):
-            # superseded by bydb_reduce_partials2 (explicit float exponent)
-            sym = "bydb_reduce_partials2"
         try:
             getattr(lib, sym)
         except AttributeError:

@@ -1,11 +1,13 @@
 """End-to-end AggModeMap -> wire frame -> AggModeReduce flow on CPU:
 partials from two shards (oracle-computed) are emitted as raw frames
 (shard_id + tags + value + MEAN count sidecar, the AggModeMap output
-shape — aggregation.go:600-614), decoded on the 'liaison' side, replica-
-deduped and combined (aggregation_reduce.go:83-138), and the final values
-must equal a direct AggModeAll pass over the union."""
+shape — aggregation.go:600-614), then reduced by the PRODUCT entry point
+bydb_reduce_frames (replica dedup + Combine + Val,
+aggregation_reduce.go:83-138), and the final values must equal a direct
+AggModeAll pass over the union."""
 import banyandb_amd as ba
-from banyandb_amd.frame import FrameBuilder, FrameReader, ROLE_TAG, ROLE_FIELD
+from banyandb_amd.frame import (FrameBuilder, reduce_frames, ROLE_TAG,
+                                ROLE_FIELD, ROLE_SHARD)
 from helpers import oracle_scan
 
 T0 = 1_700_000_000_000_000_000
@@ -24,7 +26,7 @@ def shard(rank, seed=0xB4DB):
 def emit_map_frame(shard_id, parts):
     """AggModeMap emit: shard_id first, tag, value, value__agg_count."""
     fb = FrameBuilder(len(parts))
-    fb.add_i64(4, "shard_id", "", [shard_id] * len(parts))
+    fb.add_i64(ROLE_SHARD, "shard_id", "", [shard_id] * len(parts))
     fb.add_str(ROLE_TAG, "service_id", "meta",
                [f"g{g}".encode() for g in range(len(parts))])
     fb.add_i64(ROLE_FIELD, "value", "", [p.sum_i for p in parts])
@@ -33,31 +35,21 @@ def emit_map_frame(shard_id, parts):
 
 
 def test_map_frames_reduce_to_all():
-    # Map phase on each shard (oracle = the reference semantics)
+    # Map phase on each shard (oracle = the reference semantics); each
+    # shard's frame arrives TWICE (a replica) — dedup must drop the copy.
     frames = []
     for rank in range(2):
         res = oracle_scan(shard(rank), ba.VT_INT64, n_groups=N_GROUPS)
         frames.append(emit_map_frame(rank, res))
-        # replica: the same shard emitted twice (dedup must drop it)
         frames.append(emit_map_frame(rank, res))
 
-    # Reduce phase: decode frames, dedup on (shard_id, group_key), Combine
-    seen = set()
-    acc = {}
-    for data in frames:
-        r = FrameReader(data)
-        shard_ids = r.col_i64(0)
-        keys = r.col_var(1)
-        sums = r.col_i64(2)
-        counts = r.col_i64(3)
-        for i in range(r.nrows):
-            dk = (shard_ids[i], keys[i])
-            if dk in seen:        # replica duplicate -> dropped
-                continue
-            seen.add(dk)
-            st = acc.setdefault(keys[i], [0, 0])
-            st[0] = (st[0] + sums[i]) % 2 ** 64
-            st[1] += counts[i]
+    # Reduce phase: the product C-ABI does dedup + Combine + Val.
+    # specs: SUM over col 2, MEAN over col 2 (count sidecar col 3 located
+    # by name), plus plain sum over the count column for the row count.
+    out = reduce_frames(frames,
+                        specs=[(2, ba.AGG_SUM), (2, ba.AGG_MEAN),
+                               (3, ba.AGG_SUM)],
+                        key_cols=[1], shard_col=0)
 
     # AggModeAll over the union
     union = ba.PartBuilder()
@@ -67,7 +59,55 @@ def test_map_frames_reduce_to_all():
                                  s * 1000, 1, 0xB4DB ^ (rank << 32),
                                  group_code=s % N_GROUPS)
     want = oracle_scan(union, ba.VT_INT64, n_groups=N_GROUPS)
-    for g in range(N_GROUPS):
-        st = acc[f"g{g}".encode()]
-        assert st[1] == want[g].count
-        assert st[0] % 2 ** 64 == want[g].sum_i % 2 ** 64
+    assert len(out) == N_GROUPS
+    for g, (key, vals) in enumerate(out):
+        # first-seen order follows the frame row order g0..g3; the key is
+        # the packed appendKeyComponent form: present marker + uvarint len
+        assert key == b"\x01\x02" + f"g{g}".encode()
+        s_sum, s_mean, s_cnt = vals
+        assert s_cnt[0] == want[g].count
+        assert s_sum[0] % 2 ** 64 == want[g].sum_i % 2 ** 64
+        mean = want[g].sum_i // want[g].count
+        assert s_mean[0] == max(mean, 1)
+
+
+def test_reduce_frames_dedup_and_shard_semantics():
+    """Same group from DIFFERENT shards combines; same (shard, group)
+    drops.  shard_col=-1 falls back to key-only dedup (the reference's
+    shardIDIdx == -1 fallback)."""
+    def frame(shard_id, sums):
+        fb = FrameBuilder(len(sums))
+        fb.add_i64(ROLE_SHARD, "shard_id", "", [shard_id] * len(sums))
+        fb.add_str(ROLE_TAG, "g", "", [b"k%d" % i for i in range(len(sums))])
+        fb.add_i64(ROLE_FIELD, "value", "", sums)
+        return fb.finish()
+
+    fa, fb_, fa2 = frame(0, [10, 20]), frame(1, [1, 2]), frame(0, [10, 20])
+    out = reduce_frames([fa, fb_, fa2], specs=[(2, ba.AGG_SUM)],
+                        key_cols=[1], shard_col=0)
+    assert [v[0][0] for _, v in out] == [11, 22]
+
+    # key-only dedup: the second shard's rows now count as replicas
+    out2 = reduce_frames([fa, fb_], specs=[(2, ba.AGG_SUM)],
+                         key_cols=[1], shard_col=-1)
+    assert [v[0][0] for _, v in out2] == [10, 20]
+
+
+def test_reduce_frames_float_min_max_and_nulls():
+    """Float specs combine in the float64 domain; MIN/MAX are
+    sentinel-aware (function.go:224-228); null value rows skip
+    (combinePartial's IsNull check)."""
+    def frame(shard_id, mins, nulls=None):
+        fb = FrameBuilder(len(mins))
+        fb.add_i64(ROLE_SHARD, "shard_id", "", [shard_id] * len(mins))
+        fb.add_str(ROLE_TAG, "g", "", [b"k%d" % i for i in range(len(mins))])
+        fb.add_f64(ROLE_FIELD, "value", "", mins, nulls=nulls)
+        return fb.finish()
+
+    f0 = frame(0, [5.5, -2.0])
+    f1 = frame(1, [3.25, 0.0], nulls=[0, 1])  # group k1's row is null
+    out = reduce_frames([f0, f1], specs=[(2, ba.AGG_MIN), (2, ba.AGG_MAX)],
+                        key_cols=[1], shard_col=0)
+    (k0, v0), (k1, v1) = out
+    assert v0[0][1] == 3.25 and v0[1][1] == 5.5
+    assert v1[0][1] == -2.0 and v1[1][1] == -2.0  # null row skipped
