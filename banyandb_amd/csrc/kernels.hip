@@ -1174,21 +1174,26 @@ __device__ void scan_stream(const uint8_t *stream, int64_t n_deltas, bool dod,
     // bound counting is live only on the timestamp-clamp scan; value
     // folds pass INT64_MAX/MIN and skip the two 64-bit compares per value
     const bool want_bounds = lo_bound != INT64_MAX || hi_bound != INT64_MIN;
-    uint32_t w_cur = 0, w_nx1 = 0;
+    uint32_t w_cur = 0, w_nx1 = 0, w_nx2 = 0;
     int w_valid = 0;   // how many upcoming quad windows are already loaded
     while (j <= jmax) {
         if (!dod && no_walk && carry_n <= 1) {
             // 256-B quad window: FOUR bytes per lane (one u32 load), each
             // lane decoding up to four 1-2-byte varints; one 32-bit scan
-            // covers 256 B
+            // covers 256 B.  Loads lead FOUR windows (1 KiB — the part's
+            // zeroed slack bounds the over-read) so the next windows'
+            // fetches are in flight while this one scans.
             if (w_valid < 1)
                 __builtin_memcpy(&w_cur, stream + pos + 4 * (uint64_t)lane,
                                  4);
             if (w_valid < 2)
                 __builtin_memcpy(&w_nx1,
                                  stream + pos + 256 + 4 * (uint64_t)lane, 4);
-            uint32_t w_nxt;    // two windows of load lead
-            __builtin_memcpy(&w_nxt, stream + pos + 512 + 4 * (uint64_t)lane,
+            if (w_valid < 3)
+                __builtin_memcpy(&w_nx2,
+                                 stream + pos + 512 + 4 * (uint64_t)lane, 4);
+            uint32_t w_nxt;
+            __builtin_memcpy(&w_nxt, stream + pos + 768 + 4 * (uint64_t)lane,
                              4);
             uint32_t b0 = w_cur & 0xffu, b1 = (w_cur >> 8) & 0xffu,
                      b2 = (w_cur >> 16) & 0xffu, b3 = w_cur >> 24;
@@ -1225,33 +1230,55 @@ __device__ void scan_stream(const uint8_t *stream, int64_t n_deltas, bool dod,
                 int32_t sl = dA + dB + dC + dD;
                 int32_t S = wave_incl_scan32(sl, lane);
                 int32_t cum = S - sl;
-                int rb = __popcll(e0 & lanemask_lt(lane)) +
-                         __popcll(e1 & lanemask_lt(lane)) +
-                         __popcll(e2 & lanemask_lt(lane)) +
-                         __popcll(e3 & lanemask_lt(lane));
-                int64_t idx = j + rb;
-#define BYDB_QVAL(tk, dk)                                                    \
-                if (tk) {                                                    \
-                    cum += dk;                                               \
-                    int64_t sv =                                             \
-                        (int64_t)(v_carry + (uint64_t)(int64_t)cum);         \
-                    if (idx >= r0 && idx <= r1) {                            \
+                if (!want_bounds && j >= r0 && j + nt - 1 <= r1) {
+                    // whole window selected (the dominant shape once the
+                    // block-level clamp resolved): no per-value index
+                    // bookkeeping — every terminator folds
+#define BYDB_QVF(tk, dk)                                                     \
+                    if (tk) {                                                \
+                        cum += dk;                                           \
+                        int64_t sv =                                         \
+                            (int64_t)(v_carry + (uint64_t)(int64_t)cum);     \
                         l_sum += (uint64_t)sv;                               \
-                        l_nsel++;                                            \
                         l_mn = sv < l_mn ? sv : l_mn;                        \
                         l_mx = sv > l_mx ? sv : l_mx;                        \
-                    }                                                        \
-                    if (want_bounds) {                                       \
-                        if (sv < lo_bound) l_nlo++;                          \
-                        if (sv > hi_bound) l_nhi++;                          \
-                    }                                                        \
-                    idx++;                                                   \
-                }
-                BYDB_QVAL(t0, dA)
-                BYDB_QVAL(t1, dB)
-                BYDB_QVAL(t2, dC)
-                BYDB_QVAL(t3, dD)
+                    }
+                    BYDB_QVF(t0, dA)
+                    BYDB_QVF(t1, dB)
+                    BYDB_QVF(t2, dC)
+                    BYDB_QVF(t3, dD)
+#undef BYDB_QVF
+                    l_nsel += (uint64_t)((int)t0 + (int)t1 + (int)t2 +
+                                         (int)t3);
+                } else {
+                    int rb = __popcll(e0 & lanemask_lt(lane)) +
+                             __popcll(e1 & lanemask_lt(lane)) +
+                             __popcll(e2 & lanemask_lt(lane)) +
+                             __popcll(e3 & lanemask_lt(lane));
+                    int64_t idx = j + rb;
+#define BYDB_QVAL(tk, dk)                                                    \
+                    if (tk) {                                                \
+                        cum += dk;                                           \
+                        int64_t sv =                                         \
+                            (int64_t)(v_carry + (uint64_t)(int64_t)cum);     \
+                        if (idx >= r0 && idx <= r1) {                        \
+                            l_sum += (uint64_t)sv;                           \
+                            l_nsel++;                                        \
+                            l_mn = sv < l_mn ? sv : l_mn;                    \
+                            l_mx = sv > l_mx ? sv : l_mx;                    \
+                        }                                                    \
+                        if (want_bounds) {                                   \
+                            if (sv < lo_bound) l_nlo++;                      \
+                            if (sv > hi_bound) l_nhi++;                      \
+                        }                                                    \
+                        idx++;                                               \
+                    }
+                    BYDB_QVAL(t0, dA)
+                    BYDB_QVAL(t1, dB)
+                    BYDB_QVAL(t2, dC)
+                    BYDB_QVAL(t3, dD)
 #undef BYDB_QVAL
+                }
                 v_carry +=
                     (uint64_t)(int64_t)__builtin_amdgcn_readlane(S, 63);
                 if (e3 >> 63) {
@@ -1266,8 +1293,9 @@ __device__ void scan_stream(const uint8_t *stream, int64_t n_deltas, bool dod,
                 j += nt;
                 pos += 256;
                 w_cur = w_nx1;
-                w_nx1 = w_nxt;
-                w_valid = 2;
+                w_nx1 = w_nx2;
+                w_nx2 = w_nxt;
+                w_valid = 3;
                 continue;
             }
             w_valid = 0;
