@@ -1174,27 +1174,25 @@ __device__ void scan_stream(const uint8_t *stream, int64_t n_deltas, bool dod,
     // bound counting is live only on the timestamp-clamp scan; value
     // folds pass INT64_MAX/MIN and skip the two 64-bit compares per value
     const bool want_bounds = lo_bound != INT64_MAX || hi_bound != INT64_MIN;
-    uint32_t w_cur = 0, w_nx1 = 0, w_nx2 = 0;
-    int w_valid = 0;   // how many upcoming quad windows are already loaded
+#define QLEAD 8
+    uint32_t wq[QLEAD];  // wq[q] = this lane's u32 of the window at pos+256q
+    int w_valid = 0;     // how many upcoming quad windows are already loaded
     while (j <= jmax) {
         if (!dod && no_walk && carry_n <= 1) {
             // 256-B quad window: FOUR bytes per lane (one u32 load), each
             // lane decoding up to four 1-2-byte varints; one 32-bit scan
-            // covers 256 B.  Loads lead FOUR windows (1 KiB — the part's
-            // zeroed slack bounds the over-read) so the next windows'
-            // fetches are in flight while this one scans.
-            if (w_valid < 1)
-                __builtin_memcpy(&w_cur, stream + pos + 4 * (uint64_t)lane,
-                                 4);
-            if (w_valid < 2)
-                __builtin_memcpy(&w_nx1,
-                                 stream + pos + 256 + 4 * (uint64_t)lane, 4);
-            if (w_valid < 3)
-                __builtin_memcpy(&w_nx2,
-                                 stream + pos + 512 + 4 * (uint64_t)lane, 4);
-            uint32_t w_nxt;
-            __builtin_memcpy(&w_nxt, stream + pos + 768 + 4 * (uint64_t)lane,
-                             4);
+            // covers 256 B.  Loads lead EIGHT windows (2 KiB — the part's
+            // zeroed slack bounds the over-read): the scan is
+            // latency-parked, not issue-bound, so the pipeline depth is
+            // sized to cover one HBM round trip of window fetches.
+#pragma unroll
+            for (int q = 0; q < QLEAD; q++)
+                if (q >= w_valid)
+                    __builtin_memcpy(&wq[q],
+                                     stream + pos + 256 * (uint64_t)q +
+                                         4 * (uint64_t)lane, 4);
+            w_valid = QLEAD;
+            const uint32_t w_cur = wq[0];
             uint32_t b0 = w_cur & 0xffu, b1 = (w_cur >> 8) & 0xffu,
                      b2 = (w_cur >> 16) & 0xffu, b3 = w_cur >> 24;
             uint64_t e0 = __ballot(b0 < 0x80);
@@ -1292,10 +1290,9 @@ __device__ void scan_stream(const uint8_t *stream, int64_t n_deltas, bool dod,
                 }
                 j += nt;
                 pos += 256;
-                w_cur = w_nx1;
-                w_nx1 = w_nx2;
-                w_nx2 = w_nxt;
-                w_valid = 3;
+#pragma unroll
+                for (int q = 0; q + 1 < QLEAD; q++) wq[q] = wq[q + 1];
+                w_valid = QLEAD - 1;
                 continue;
             }
             w_valid = 0;
@@ -2730,10 +2727,11 @@ extern "C" int bydb_part_reserve(bydb_session *s, uint64_t payload_bytes,
     HIP_TRY(s, hipSetDevice(s->device));
     if (s->d_payload) { (void)hipFree(s->d_payload); s->d_payload = nullptr; }
     if (s->d_blocks) { (void)hipFree(s->d_blocks); s->d_blocks = nullptr; }
-    // +1KiB slack: the 256-byte window loop prefetches one window ahead
-    // and may read past the last stream's end
-    HIP_TRY(s, hipMalloc(&s->d_payload, payload_bytes + 1024));
-    HIP_TRY(s, hipMemset(s->d_payload + payload_bytes, 0, 1024));
+    // +4KiB slack: the 256-byte quad-window loop runs an 8-window-deep
+    // load pipeline (2 KiB of lead) and may read past the last stream's
+    // end
+    HIP_TRY(s, hipMalloc(&s->d_payload, payload_bytes + 4096));
+    HIP_TRY(s, hipMemset(s->d_payload + payload_bytes, 0, 4096));
     HIP_TRY(s, hipMalloc(&s->d_blocks, sizeof(bydb_block_desc) * (size_t)n_blocks));
     s->payload_cap = payload_bytes;
     s->blocks_cap = n_blocks;
