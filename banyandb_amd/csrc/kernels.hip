@@ -1174,17 +1174,16 @@ __device__ void scan_stream(const uint8_t *stream, int64_t n_deltas, bool dod,
     // bound counting is live only on the timestamp-clamp scan; value
     // folds pass INT64_MAX/MIN and skip the two 64-bit compares per value
     const bool want_bounds = lo_bound != INT64_MAX || hi_bound != INT64_MIN;
-#define QLEAD 8
+#define QLEAD 4
     uint32_t wq[QLEAD];  // wq[q] = this lane's u32 of the window at pos+256q
     int w_valid = 0;     // how many upcoming quad windows are already loaded
     while (j <= jmax) {
         if (!dod && no_walk && carry_n <= 1) {
             // 256-B quad window: FOUR bytes per lane (one u32 load), each
             // lane decoding up to four 1-2-byte varints; one 32-bit scan
-            // covers 256 B.  Loads lead EIGHT windows (2 KiB — the part's
-            // zeroed slack bounds the over-read): the scan is
-            // latency-parked, not issue-bound, so the pipeline depth is
-            // sized to cover one HBM round trip of window fetches.
+            // covers 256 B.  Loads lead QLEAD windows (an 8-deep lead
+            // was measured SLOWER — register pressure beats the extra
+            // latency cover at 128 VGPRs).
 #pragma unroll
             for (int q = 0; q < QLEAD; q++)
                 if (q >= w_valid)
@@ -1687,6 +1686,8 @@ enum {
     KF_FLOAT = 2,        // float64 field: also accumulate mantissa sum as double
                          // (bits 16-31 of flags carry the configured
                          // decimal exponent for the per-block guard)
+    KF_WALK_ONLY = 4,    // process ONLY PF_WALK blocks (the heavy pass of
+                         // the light/heavy predicate split below)
 };
 
 __device__ __forceinline__ void flush_partial(bydb_partial *partials,
@@ -1875,11 +1876,21 @@ __device__ void fold_range(const uint8_t *fstream, uint8_t fenc, int64_t first,
 // reconstruction, no predicates) sheds the scan/walker machinery and its
 // register pressure; the full one keeps everything.  (EN_* are constant
 // guards — dead branches are eliminated per instantiation.)
-template <bool EN_VALUES, bool EN_PREDS, bool EN_GROUPS>
-// Occupancy: the value-scan instantiations are scattered-load
-// latency-bound (63% of wave cycles parked), so they request 6 waves/SIMD
-// even at the cost of tighter register allocation; the closed-form-only
-// instantiation already fits 6.
+//
+// EN_WALK — the light/heavy predicate split.  Row-varying predicate
+// walkers (PredWalk x3, ~300 B of per-lane state) force scratch spills
+// into the value-scan hot loop even when every block resolves to a
+// uniform verdict (entity tags — the dominant case).  With predicates
+// the host launches TWO passes on the same stream: a light kernel
+// (EN_WALK=false) that folds every PF_CLEAR/PF_SKIP block and skips
+// PF_WALK ones, carrying no walker state at all, then a heavy kernel
+// (EN_WALK=true, flags & KF_WALK_ONLY) that folds only the PF_WALK
+// blocks.  When no block needs a walker the heavy pass is a ~flag-byte
+// sweep.
+template <bool EN_VALUES, bool EN_PREDS, bool EN_GROUPS, bool EN_WALK>
+// Occupancy: 4 waves/SIMD for the value-scan instantiations — requesting
+// 6 forces scratch spills into the window loop and measured ~1.6x SLOWER;
+// the closed-form-only instantiation fits 6 without spills.
 __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) void k_scan_agg_t(
     const uint8_t *__restrict__ payload, const uint8_t *__restrict__ sidecar,
     const bydb_block_desc *__restrict__ blocks,
@@ -1931,6 +1942,17 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
         }
         const bool seg_eligible = segs && seg0_off != SEG_INELIGIBLE;
         if (seg != 0 && !seg_eligible) continue;
+        // light/heavy split routing (see the template comment): the heavy
+        // pass owns exactly the PF_WALK blocks, the light pass everything
+        // else — checked before any clamp work so the off-pass blocks
+        // cost one flag byte
+        if (EN_PREDS && preds != nullptr) {
+            if (flags & KF_WALK_ONLY) {
+                if (pf_pre != PF_WALK) continue;
+            } else if (!EN_WALK && pf_pre == PF_WALK) {
+                continue;
+            }
+        }
         const int64_t n = (int64_t)bd->count;
         if (seg != 0 && (int64_t)seg * SEG_ROWS + 1 > n - 1) continue;
         const int64_t ts_min = bd->ts_min, ts_max = bd->ts_max;
@@ -2030,7 +2052,7 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
                 dev_set_err(derr, DERR_BAD_ENC, (uint64_t)bi);
                 continue;
             }
-            if (pf == PF_WALK) {
+            if (EN_WALK && pf == PF_WALK) {
                 bool skip_block = false;
 #pragma unroll
                 for (int sl = 0; sl < 3; sl++) {
@@ -3158,17 +3180,35 @@ extern "C" int bydb_consume_multi(bydb_session *s, int64_t min_ts,
                 const uint64_t *, const uint8_t *, const SegEntry *,
                 const GroupBlock *, const uint16_t *, int, int64_t, int64_t,
                 int64_t, int64_t, bydb_partial *, DevErr *);
+    void (*kfn_walk)(const uint8_t *, const uint8_t *,
+                     const bydb_block_desc *, int64_t, int64_t, int64_t, int,
+                     const PredBlock *, int, const uint64_t *,
+                     const uint8_t *, const SegEntry *, const GroupBlock *,
+                     const uint16_t *, int, int64_t, int64_t, int64_t,
+                     int64_t, bydb_partial *, DevErr *) = nullptr;
     if (en_values) {
-        if (en_preds) kfn = en_groups ? k_scan_agg_t<true, true, true>
-                                      : k_scan_agg_t<true, true, false>;
-        else kfn = en_groups ? k_scan_agg_t<true, false, true>
-                             : k_scan_agg_t<true, false, false>;
+        if (en_preds) {
+            kfn = en_groups ? k_scan_agg_t<true, true, true, false>
+                            : k_scan_agg_t<true, true, false, false>;
+            kfn_walk = en_groups ? k_scan_agg_t<true, true, true, true>
+                                 : k_scan_agg_t<true, true, false, true>;
+        } else {
+            kfn = en_groups ? k_scan_agg_t<true, false, true, false>
+                            : k_scan_agg_t<true, false, false, false>;
+        }
     } else {
-        if (en_preds) kfn = en_groups ? k_scan_agg_t<false, true, true>
-                                      : k_scan_agg_t<false, true, false>;
-        else kfn = en_groups ? k_scan_agg_t<false, false, true>
-                             : k_scan_agg_t<false, false, false>;
+        if (en_preds) {
+            kfn = en_groups ? k_scan_agg_t<false, true, true, false>
+                            : k_scan_agg_t<false, true, false, false>;
+            kfn_walk = en_groups ? k_scan_agg_t<false, true, true, true>
+                                 : k_scan_agg_t<false, true, false, true>;
+        } else {
+            kfn = en_groups ? k_scan_agg_t<false, false, true, false>
+                            : k_scan_agg_t<false, false, false, false>;
+        }
     }
+    // light pass: every block whose predicate verdict is uniform (no
+    // walker state, no scratch spills in the scan loop)
     hipLaunchKernelGGL(kfn, dim3(grid), dim3(threads), 0, s->stream,
                        s->d_payload, s->d_sidecar, s->d_blocks, s->n_blocks,
                        min_ts, max_ts, flags, preds, n_preds, s->d_pred_bm,
@@ -3176,6 +3216,18 @@ extern "C" int bydb_consume_multi(bydb_session *s, int64_t min_ts,
                        s->gmul[0], s->gmul[1], s->gmul[2],
                        (int64_t)s->n_groups, s->d_acc, s->d_err);
     HIP_TRY(s, hipGetLastError());
+    // heavy pass: only the PF_WALK blocks (row-varying predicates).  When
+    // none exist this is a flag-byte sweep (~tens of us)
+    if (kfn_walk != nullptr) {
+        hipLaunchKernelGGL(kfn_walk, dim3(grid), dim3(threads), 0, s->stream,
+                           s->d_payload, s->d_sidecar, s->d_blocks,
+                           s->n_blocks, min_ts, max_ts, flags | KF_WALK_ONLY,
+                           preds, n_preds, s->d_pred_bm, s->d_pred_flags,
+                           segs, groups, s->d_gmap, s->n_gslots, s->gmul[0],
+                           s->gmul[1], s->gmul[2], (int64_t)s->n_groups,
+                           s->d_acc, s->d_err);
+        HIP_TRY(s, hipGetLastError());
+    }
     HIP_TRY(s, hipEventRecord(s->ev_stop, s->stream));
     s->consumed = true;
     return BYDB_OK;
