@@ -1176,22 +1176,38 @@ __device__ void scan_stream(const uint8_t *stream, int64_t n_deltas, bool dod,
     const bool want_bounds = lo_bound != INT64_MAX || hi_bound != INT64_MIN;
 #define QLEAD 4
     uint32_t wq[QLEAD];  // wq[q] = this lane's u32 of the window at pos+256q
-    int w_valid = 0;     // how many upcoming quad windows are already loaded
+    bool done = false;
     while (j <= jmax) {
+        // The quad path is a SELF-CONTAINED inner loop: with the 64-B
+        // fallback in the same loop body, the compiler hoists the
+        // fallback's byte load above the quad branch and fences it with
+        // s_waitcnt vmcnt(0) — draining the whole QLEAD window pipeline
+        // every iteration (measured: the scan ran at full memory latency
+        // per 256-B window).  Inside this loop the only vector loads are
+        // the window dwords, so the waits stay at vmcnt(QLEAD-1).
         if (!dod && no_walk && carry_n <= 1) {
-            // 256-B quad window: FOUR bytes per lane (one u32 load), each
-            // lane decoding up to four 1-2-byte varints; one 32-bit scan
-            // covers 256 B.  Loads lead QLEAD windows (an 8-deep lead
-            // was measured SLOWER — register pressure beats the extra
-            // latency cover at 128 VGPRs).
+            // pipeline fill: QLEAD windows in flight, then exactly ONE
+            // load per steady-state iteration (at the bottom, QLEAD-1
+            // windows ahead of its use)
 #pragma unroll
             for (int q = 0; q < QLEAD; q++)
-                if (q >= w_valid)
-                    __builtin_memcpy(&wq[q],
-                                     stream + pos + 256 * (uint64_t)q +
-                                         4 * (uint64_t)lane, 4);
-            w_valid = QLEAD;
-            const uint32_t w_cur = wq[0];
+                __builtin_memcpy(&wq[q],
+                                 stream + pos + 256 * (uint64_t)q +
+                                     4 * (uint64_t)lane, 4);
+            // Ring by PHASE, not by shifting: moving wq[q+1] into wq[q]
+            // would make every iteration wait on the NEWEST outstanding
+            // load (vmcnt(0) to read the register being shifted),
+            // re-serialising the pipeline.  The unrolled phase loop keeps
+            // indices static, so reading wq[ph] waits only for the load
+            // issued QLEAD-1 iterations ago.
+            bool quad_exit = false;
+            while (!quad_exit) {
+#pragma unroll
+            for (int ph = 0; ph < QLEAD; ph++) {
+            // 256-B quad window: FOUR bytes per lane (one u32 load), each
+            // lane decoding up to four 1-2-byte varints; one 32-bit scan
+            // covers 256 B.
+            const uint32_t w_cur = wq[ph];
             uint32_t b0 = w_cur & 0xffu, b1 = (w_cur >> 8) & 0xffu,
                      b2 = (w_cur >> 16) & 0xffu, b3 = w_cur >> 24;
             uint64_t e0 = __ballot(b0 < 0x80);
@@ -1289,14 +1305,18 @@ __device__ void scan_stream(const uint8_t *stream, int64_t n_deltas, bool dod,
                 }
                 j += nt;
                 pos += 256;
-#pragma unroll
-                for (int q = 0; q + 1 < QLEAD; q++) wq[q] = wq[q + 1];
-                w_valid = QLEAD - 1;
+                if (j > jmax) { done = true; quad_exit = true; break; }
+                // refill this phase's slot with the window QLEAD-1 ahead
+                __builtin_memcpy(&wq[ph],
+                                 stream + pos + 256 * (uint64_t)(QLEAD - 1) +
+                                     4 * (uint64_t)lane, 4);
                 continue;
             }
-            w_valid = 0;
-        } else {
-            w_valid = 0;
+            quad_exit = true;  // pattern failed: 64-B fallback below
+            break;
+            }        // phase loop
+            }        // inner quad loop
+            if (done) break;
         }
         uint8_t b = stream[pos + (uint64_t)lane];
         uint64_t emask = __ballot(b < 0x80);
