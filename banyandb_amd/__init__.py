@@ -500,3 +500,20 @@ def reduce_partials(parts_flat, n_parts_per_group, n_groups, field_vtype,
     if rc != 0:
         raise RuntimeError(f"reduce_partials rc={rc}")
     return list(out)
+
+
+def i64_tag_cell(v: int) -> bytes:
+    """Stored bytes of an int64 tag value: the reference's order-preserving
+    sign-flip 8-byte BE cell (convert/number.go:33-46 Int64ToBytes) — the
+    raw value bytes measure tag columns hold for ValueTypeInt64
+    (batch_decode.go:48-53 decodes with convert.BytesToInt64).  Use these
+    as tag values / group domains for numeric group keys."""
+    u = (v | (1 << 63)) if v >= 0 else ((1 << 63) - (-v)) % (1 << 64)
+    return u.to_bytes(8, "big")
+
+
+def f64_tag_cell(v: float) -> bytes:
+    """Stored bytes of a float64 tag value: IEEE-754 big-endian
+    (convert/number.go:128-132 Float64ToBytes)."""
+    import struct
+    return struct.pack(">d", v)

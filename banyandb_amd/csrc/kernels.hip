@@ -845,8 +845,22 @@ __global__ void k_resolve_groups(const uint8_t *__restrict__ payload,
     const uint8_t *base = in_sidecar ? sidecar : payload;
     const uint8_t *p = base + (toff & ~TAG_SIDECAR_BIT);
     const uint8_t *end = p + tlen;
-    // group-by on a plain (>256-distinct) column is unsupported in v1
-    if (*p != BYDB_ENC_DICTIONARY) { gb.err = 1; out[bi] = gb; return; }
+    if (*p != BYDB_ENC_DICTIONARY) {
+        if (*p == BYDB_ENC_PLAIN) {
+            // plain (>256-distinct) column: per-row gids are resolved by
+            // the follow-up wave kernel (k_resolve_plain_groups) into
+            // (gid,count) run pairs; width 0xFF + nentries 0 marks
+            // "pending" and the scan errors loudly if it stays that way
+            gb.width = 0xFF;
+            gb.uniform_gid = GID_VARYING;
+            gb.nentries = 0;
+            out[bi] = gb;
+            return;
+        }
+        gb.err = 1;
+        out[bi] = gb;
+        return;
+    }
     p++;
     uint64_t count = 0;
     unsigned sh = 0;
@@ -898,6 +912,129 @@ __global__ void k_resolve_groups(const uint8_t *__restrict__ payload,
     }
     gb.uniform_gid = GID_VARYING;
     out[bi] = gb;
+}
+
+// Per-row group-by on a PLAIN (>256-distinct) tag column — computeKey
+// groups on any key column (groupby.go:287-364); the reference's plain
+// fallback (column.go:266-278) stores a bytes block, host-normalized into
+// the sidecar at part registration.  One wave per block: each row's value
+// maps to its domain gid (equality by bytes, like appendKeyComponent);
+// consecutive equal gids compress into (gid,count) u16 run pairs in the
+// run arena, so the group merge walks them with the same cursor shape as
+// dictionary RLE.  Single-run blocks collapse to a uniform gid and keep
+// every fast fold path.  Nil and out-of-domain rows become gid 0xFFFF
+// (dropped rows).
+#define GRUN_ENTRIES_PER_STREAM (2 * (8192 + 1))
+__global__ __launch_bounds__(WAVE) void k_resolve_plain_groups(
+    const uint8_t *__restrict__ payload, const uint8_t *__restrict__ sidecar,
+    const bydb_block_desc *__restrict__ blocks, int64_t n_blocks, int slot,
+    GroupDomain dom, GroupBlock *__restrict__ out,
+    uint16_t *__restrict__ gruns, uint64_t grun_cap_streams,
+    uint32_t *__restrict__ ctr) {
+    const int lane = threadIdx.x;
+    for (int64_t bi = blockIdx.x; bi < n_blocks; bi += gridDim.x) {
+        GroupBlock gb = out[bi];
+        if (gb.width != 0xFF || gb.err) continue;
+        const bydb_block_desc *bd = &blocks[bi];
+        uint64_t toff = slot == 0 ? bd->tag_off
+                                  : slot == 1 ? bd->tag2_off : bd->tag3_off;
+        const uint8_t *p = (toff & TAG_SIDECAR_BIT)
+                               ? sidecar + (toff & ~TAG_SIDECAR_BIT)
+                               : payload + toff;
+        p++;  // ENC_PLAIN
+        uint64_t n = (uint64_t)p[0] | ((uint64_t)p[1] << 8) |
+                     ((uint64_t)p[2] << 16) | ((uint64_t)p[3] << 24);
+        p += 4;
+        uint8_t wt = *p++;
+        uint32_t wbytes = wt == 0 ? 1 : wt == 1 ? 2 : wt == 2 ? 4 : 8;
+        const uint8_t *lens = p;
+        const uint8_t *vals = lens + n * wbytes;
+        if (n != (uint64_t)bd->count) {
+            if (lane == 0) out[bi].err = 1;
+            continue;
+        }
+        uint32_t slot_id = 0;
+        if (lane == 0) slot_id = atomicAdd(ctr, 1u);
+        slot_id = (uint32_t)__shfl((int)slot_id, 0);
+        if (slot_id >= grun_cap_streams) {
+            if (lane == 0) out[bi].err = 1;   // arena exhausted: loud
+            continue;
+        }
+        uint16_t *rp = gruns + (uint64_t)slot_id * GRUN_ENTRIES_PER_STREAM;
+        uint64_t carry_bytes = 0;
+        uint32_t n_pairs = 0;
+        int32_t cur_gid = -2;   // no open run yet
+        uint32_t cur_cnt = 0;
+        for (uint64_t base = 0; base < n; base += WAVE) {
+            uint64_t row = base + (uint64_t)lane;
+            uint64_t lp1 = 0;
+            if (row < n)
+                for (uint32_t b = 0; b < wbytes; b++)
+                    lp1 = (lp1 << 8) | lens[row * wbytes + b];
+            uint64_t vlen = lp1 ? lp1 - 1 : 0;
+            uint64_t incl = wave_incl_scan(vlen, lane);
+            uint64_t myoff = carry_bytes + incl - vlen;
+            int32_t g = -1;     // nil or out-of-domain: dropped row
+            if (row < n && lp1) {
+                uint32_t gg = domain_lookup(&dom, vals + myoff, vlen);
+                g = gg == GID_NONE ? -1 : (int32_t)gg;
+            }
+            int32_t prev = __shfl_up(g, 1);
+            if (lane == 0) prev = cur_gid;
+            bool valid = row < n;
+            bool is_start = valid && g != prev;
+            uint64_t starts = __ballot(is_start);
+            uint32_t chunk_rows =
+                (uint32_t)(n - base < WAVE ? n - base : (uint64_t)WAVE);
+            if (starts == 0) {
+                cur_cnt += chunk_rows;
+            } else {
+                int F = (int)__builtin_ctzll(starts);
+                bool head = (cur_cnt + (uint32_t)F) > 0;
+                if (lane == 0 && head) {
+                    rp[2 * n_pairs] =
+                        cur_gid < 0 ? 0xFFFFu : (uint16_t)cur_gid;
+                    rp[2 * n_pairs + 1] = (uint16_t)(cur_cnt + (uint32_t)F);
+                }
+                int last = 63 - __clzll(starts);
+                int nst = __popcll(starts);
+                if (is_start && lane != last) {
+                    int rank = __popcll(starts & lanemask_lt(lane));
+                    uint64_t hi = starts >> (lane + 1);   // lane < 63 here
+                    int nxt = lane + 1 + (int)__builtin_ctzll(hi);
+                    uint32_t idx = n_pairs + (head ? 1u : 0u) + (uint32_t)rank;
+                    rp[2 * idx] = g < 0 ? 0xFFFFu : (uint16_t)g;
+                    rp[2 * idx + 1] = (uint16_t)(nxt - lane);
+                }
+                n_pairs += (head ? 1u : 0u) + (uint32_t)(nst - 1);
+                cur_gid = __shfl(g, last);
+                cur_cnt = chunk_rows - (uint32_t)last;
+            }
+            carry_bytes += readlane64(incl, WAVE - 1);
+        }
+        if (cur_cnt > 0) {
+            if (lane == 0) {
+                rp[2 * n_pairs] = cur_gid < 0 ? 0xFFFFu : (uint16_t)cur_gid;
+                rp[2 * n_pairs + 1] = (uint16_t)cur_cnt;
+            }
+            n_pairs++;
+        }
+        if (lane == 0) {
+            GroupBlock *o = &out[bi];
+            if (n_pairs == 1) {
+                // whole block one run: uniform fast path, no runs walked
+                uint16_t g16 = rp[0];
+                o->uniform_gid = g16 == 0xFFFFu ? GID_NONE : (uint32_t)g16;
+                o->width = 0;
+                o->nentries = 0;
+            } else {
+                o->uniform_gid = GID_VARYING;
+                o->rle_bit_off =
+                    (uint64_t)slot_id * GRUN_ENTRIES_PER_STREAM;
+                o->nentries = 2 * n_pairs;
+            }
+        }
+    }
 }
 
 // Resolve prepass: one thread per block parses the dictionary header and
@@ -1920,7 +2057,8 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
     const uint8_t *__restrict__ pred_flags,
     const SegEntry *__restrict__ segs_in,
     const GroupBlock *__restrict__ groups_in,
-    const uint16_t *__restrict__ gmap_in, int n_gslots, int64_t gm0,
+    const uint16_t *__restrict__ gmap_in,
+    const uint16_t *__restrict__ gruns_in, int n_gslots, int64_t gm0,
     int64_t gm1, int64_t gm2, int64_t n_groups,
     bydb_partial *__restrict__ partials, DevErr *derr) {
     const int64_t gmul[3] = {gm0, gm1, gm2};
@@ -2140,12 +2278,15 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
                      (bd->field_off & TAG_SIDECAR_BIT))
                         ? sidecar + (bd->field_off & ~TAG_SIDECAR_BIT)
                         : nullptr;
-                // per-slot run cursors (uniform slots are one infinite run)
+                // per-slot run cursors (uniform slots are one infinite
+                // run); plain slots (width 0xFF) walk (gid,count) u16
+                // pairs from the run arena instead of bit-packed RLE
                 const uint8_t *gsrc[3];
                 uint64_t bit[3];
                 uint32_t ent[3];
                 int64_t run_hi_s[3];
                 int64_t gid_s[3];
+                bool bad_plain = false;
                 for (int sl = 0; sl < n_gslots; sl++) {
                     gsrc[sl] = (gb[sl].rle_bit_off & TAG_SIDECAR_BIT)
                                    ? sidecar : payload;
@@ -2155,16 +2296,31 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
                         ent[sl] = gb[sl].nentries;  // exhausted
                         bit[sl] = 0;
                     } else {
+                        if (gb[sl].width == 0xFF && gb[sl].nentries == 0)
+                            bad_plain = true;  // resolve pass never ran
                         bit[sl] = gb[sl].rle_bit_off & ~TAG_SIDECAR_BIT;
                         ent[sl] = 0;
                         run_hi_s[sl] = 0;
                         gid_s[sl] = -1;
                     }
                 }
+                if (bad_plain) {
+                    dev_set_err(derr, DERR_BAD_ENC, (uint64_t)bi);
+                    continue;
+                }
                 auto advance = [&](int sl) {
                     if (ent[sl] + 1 >= gb[sl].nentries) {
                         run_hi_s[sl] = n;
                         gid_s[sl] = -1;  // RLE exhausted: no group
+                        return;
+                    }
+                    if (gb[sl].width == 0xFF) {
+                        const uint16_t *rp = gruns_in + gb[sl].rle_bit_off;
+                        uint16_t g16 = rp[ent[sl]];
+                        uint16_t cnt16 = rp[ent[sl] + 1];
+                        ent[sl] += 2;
+                        gid_s[sl] = g16 == 0xFFFFu ? -1 : (int64_t)g16;
+                        run_hi_s[sl] += (int64_t)cnt16;
                         return;
                     }
                     uint64_t code = rd_bits_be(gsrc[sl], bit[sl], gb[sl].width);
@@ -2672,6 +2828,11 @@ struct bydb_session {
     uint16_t *d_gmap = nullptr;        // [slot][block][256]
     int64_t groups_cap = 0;
     bool groups_built = false;
+    // plain-tag group-by: (gid,count) run pairs per normalized plain
+    // stream on a grouped slot (k_resolve_plain_groups)
+    uint16_t *d_gruns = nullptr;
+    uint64_t gruns_cap_streams = 0;
+    uint32_t *d_grun_ctr = nullptr;
     uint8_t *d_dom_blob[3] = {};
     uint64_t *d_dom_offs[3] = {};
     uint64_t *d_dom_hashes[3] = {};
@@ -2743,6 +2904,8 @@ extern "C" void bydb_session_destroy(bydb_session *s) {
     if (s->d_segs) (void)hipFree(s->d_segs);
     if (s->d_groups) (void)hipFree(s->d_groups);
     if (s->d_gmap) (void)hipFree(s->d_gmap);
+    if (s->d_gruns) (void)hipFree(s->d_gruns);
+    if (s->d_grun_ctr) (void)hipFree(s->d_grun_ctr);
     for (int sl = 0; sl < 3; sl++) {
         if (s->d_dom_blob[sl]) (void)hipFree(s->d_dom_blob[sl]);
         if (s->d_dom_offs[sl]) (void)hipFree(s->d_dom_offs[sl]);
@@ -3167,6 +3330,27 @@ extern "C" int bydb_consume_multi(bydb_session *s, int64_t min_ts,
         if (!s->groups_built) {
             int rthreads = 256;
             int rblocks = (int)((s->n_blocks + rthreads - 1) / rthreads);
+            // plain (>256-distinct) tag columns on grouped slots resolve
+            // into (gid,count) run pairs; size the arena by the count of
+            // normalized plain streams (an upper bound)
+            if (s->n_plain_host > 0) {
+                uint64_t want = s->n_plain_host;
+                uint64_t bytes =
+                    want * GRUN_ENTRIES_PER_STREAM * sizeof(uint16_t);
+                if (bytes > (4ull << 30)) {
+                    s->err = "plain-tag group-by run arena exceeds 4 GiB";
+                    return BYDB_ERR_OOM;
+                }
+                if (want > s->gruns_cap_streams) {
+                    if (s->d_gruns) (void)hipFree(s->d_gruns);
+                    HIP_TRY(s, hipMalloc(&s->d_gruns, bytes));
+                    s->gruns_cap_streams = want;
+                }
+                if (!s->d_grun_ctr)
+                    HIP_TRY(s, hipMalloc(&s->d_grun_ctr, sizeof(uint32_t)));
+                HIP_TRY(s, hipMemsetAsync(s->d_grun_ctr, 0, sizeof(uint32_t),
+                                          s->stream));
+            }
             for (int i = 0; i < s->n_gslots; i++) {
                 GroupDomain dom;
                 dom.blob = s->d_dom_blob[i];
@@ -3182,6 +3366,17 @@ extern "C" int bydb_consume_multi(bydb_session *s, int64_t min_ts,
                     s->d_groups + (int64_t)i * s->n_blocks,
                     s->d_gmap, (uint32_t)((int64_t)i * 256 * s->n_blocks));
                 HIP_TRY(s, hipGetLastError());
+                if (s->n_plain_host > 0) {
+                    int pgrid =
+                        (int)(s->n_blocks < 65535 ? s->n_blocks : 65535);
+                    hipLaunchKernelGGL(
+                        k_resolve_plain_groups, dim3(pgrid), dim3(WAVE), 0,
+                        s->stream, s->d_payload, s->d_sidecar, s->d_blocks,
+                        s->n_blocks, s->gslots[i], dom,
+                        s->d_groups + (int64_t)i * s->n_blocks, s->d_gruns,
+                        s->gruns_cap_streams, s->d_grun_ctr);
+                    HIP_TRY(s, hipGetLastError());
+                }
             }
             s->groups_built = true;
         }
@@ -3198,14 +3393,16 @@ extern "C" int bydb_consume_multi(bydb_session *s, int64_t min_ts,
     void (*kfn)(const uint8_t *, const uint8_t *, const bydb_block_desc *,
                 int64_t, int64_t, int64_t, int, const PredBlock *, int,
                 const uint64_t *, const uint8_t *, const SegEntry *,
-                const GroupBlock *, const uint16_t *, int, int64_t, int64_t,
-                int64_t, int64_t, bydb_partial *, DevErr *);
+                const GroupBlock *, const uint16_t *, const uint16_t *, int,
+                int64_t, int64_t, int64_t, int64_t, bydb_partial *,
+                DevErr *);
     void (*kfn_walk)(const uint8_t *, const uint8_t *,
                      const bydb_block_desc *, int64_t, int64_t, int64_t, int,
                      const PredBlock *, int, const uint64_t *,
                      const uint8_t *, const SegEntry *, const GroupBlock *,
-                     const uint16_t *, int, int64_t, int64_t, int64_t,
-                     int64_t, bydb_partial *, DevErr *) = nullptr;
+                     const uint16_t *, const uint16_t *, int, int64_t,
+                     int64_t, int64_t, int64_t, bydb_partial *,
+                     DevErr *) = nullptr;
     if (en_values) {
         if (en_preds) {
             kfn = en_groups ? k_scan_agg_t<true, true, true, false>
@@ -3232,8 +3429,8 @@ extern "C" int bydb_consume_multi(bydb_session *s, int64_t min_ts,
     hipLaunchKernelGGL(kfn, dim3(grid), dim3(threads), 0, s->stream,
                        s->d_payload, s->d_sidecar, s->d_blocks, s->n_blocks,
                        min_ts, max_ts, flags, preds, n_preds, s->d_pred_bm,
-                       s->d_pred_flags, segs, groups, s->d_gmap, s->n_gslots,
-                       s->gmul[0], s->gmul[1], s->gmul[2],
+                       s->d_pred_flags, segs, groups, s->d_gmap, s->d_gruns,
+                       s->n_gslots, s->gmul[0], s->gmul[1], s->gmul[2],
                        (int64_t)s->n_groups, s->d_acc, s->d_err);
     HIP_TRY(s, hipGetLastError());
     // heavy pass: only the PF_WALK blocks (row-varying predicates).  When
@@ -3243,9 +3440,9 @@ extern "C" int bydb_consume_multi(bydb_session *s, int64_t min_ts,
                            s->d_payload, s->d_sidecar, s->d_blocks,
                            s->n_blocks, min_ts, max_ts, flags | KF_WALK_ONLY,
                            preds, n_preds, s->d_pred_bm, s->d_pred_flags,
-                           segs, groups, s->d_gmap, s->n_gslots, s->gmul[0],
-                           s->gmul[1], s->gmul[2], (int64_t)s->n_groups,
-                           s->d_acc, s->d_err);
+                           segs, groups, s->d_gmap, s->d_gruns, s->n_gslots,
+                           s->gmul[0], s->gmul[1], s->gmul[2],
+                           (int64_t)s->n_groups, s->d_acc, s->d_err);
         HIP_TRY(s, hipGetLastError());
     }
     HIP_TRY(s, hipEventRecord(s->ev_stop, s->stream));

@@ -1359,10 +1359,20 @@ int bo_scan_agg_bytags(const uint8_t *payload, const bo_block_desc *blocks,
             uint64_t toff = slot == 0 ? b->tag_off : slot == 1 ? b->tag2_off : b->tag3_off;
             uint64_t tlen = slot == 0 ? b->tag_len : slot == 1 ? b->tag2_len : b->tag3_len;
             if (tlen == 0) { dropped = 1; break; }
-            if (payload[toff] != BO_ENC_DICTIONARY) { rc = BO_ERR_BAD_TYPE; break; }
+            /* computeKey groups on any key column (groupby.go:287-364):
+             * dictionary columns and the plain (>256-distinct) fallback
+             * (column.go:266-278) both decode to per-row values here */
             size_t tdl = 0;
-            rc = bo_dictionary_decode(tagdata, (size_t)1 << 24, taglen,
-                                      payload + toff + 1, tlen - 1, n, &tdl);
+            if (payload[toff] == BO_ENC_DICTIONARY)
+                rc = bo_dictionary_decode(tagdata, (size_t)1 << 24, taglen,
+                                          payload + toff + 1, tlen - 1, n,
+                                          &tdl);
+            else if (payload[toff] == BO_ENC_PLAIN)
+                rc = bo_bytes_block_decode(tagdata, (size_t)1 << 24, taglen,
+                                           payload + toff + 1, tlen - 1, n,
+                                           &tdl);
+            else
+                rc = BO_ERR_BAD_TYPE;
             if (rc != BO_OK) break;
             const uint8_t *dv = dom_blobs[si];
             const int64_t *dl = dom_lens[si];
