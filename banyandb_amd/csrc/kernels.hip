@@ -1861,15 +1861,21 @@ __device__ __forceinline__ void flush_partial(bydb_partial *partials,
     if (wsumf != 0.0) atomicAdd(&p->sum_f, wsumf);
 }
 
-// Fold rows [a, b] of a block's field column (no predicates): returns the
-// wrapping sum, selected count and optional min/max.  Shared by the
-// per-run grouped fold; mirrors the main kernel's dispatch.
+// Fold rows [a, b] of a block's field column: returns the wrapping sum,
+// selected count and optional min/max.  Shared by the per-run grouped
+// fold; mirrors the main kernel's dispatch.  bw0..bw2: optional
+// BITMAP-mode predicate walkers (plain >256-card tag columns,
+// k_resolve_plain) — stateless O(1) per-row lookups, so they filter any
+// sub-range; RLE walkers never reach here (the run merge intersects
+// their runs into [a, b] instead).
 __device__ void fold_range(const uint8_t *fstream, uint8_t fenc, int64_t first,
                            uint64_t field_len, int64_t n, int64_t a, int64_t b,
                            bool need_values, int lane, uint64_t *out_sum,
                            int64_t *out_mn, int64_t *out_mx, bool *out_have_mm,
                            uint64_t *out_cnt, const uint8_t *plain_norm,
-                           bool raw_f64, DevErr *derr, uint64_t bi) {
+                           bool raw_f64, DevErr *derr, uint64_t bi,
+                           PredWalk *bw0 = nullptr, PredWalk *bw1 = nullptr,
+                           PredWalk *bw2 = nullptr) {
     uint64_t nsel = (uint64_t)(b - a + 1);
     uint64_t bsum = 0;
     int64_t bmn = INT64_MAX, bmx = INT64_MIN;
@@ -1895,6 +1901,9 @@ __device__ void fold_range(const uint8_t *fstream, uint8_t fenc, int64_t first,
         for (int64_t base = a; base <= b; base += WAVE) {
             int64_t row = base + lane;
             bool ok = row <= b && ((vbm[row >> 6] >> (row & 63)) & 1);
+            if (bw0) ok = pred_match_rows(bw0, row, ok) && ok;
+            if (bw1) ok = pred_match_rows(bw1, row, ok) && ok;
+            if (bw2) ok = pred_match_rows(bw2, row, ok) && ok;
             if (ok) {
                 uint64_t u = __builtin_bswap64(cells[row]);
                 if (raw_f64) {
@@ -1932,6 +1941,86 @@ __device__ void fold_range(const uint8_t *fstream, uint8_t fenc, int64_t first,
         *out_mn = bmn;
         *out_mx = bmx;
         *out_have_mm = have;
+        return;
+    }
+    if (bw0 || bw1 || bw2) {
+        // bitmap-filtered fold over the non-Plain encodings: arithmetic
+        // progressions via the predicated row walk, varint streams via
+        // the walker-aware scan (the closed forms cannot filter rows)
+        if (fenc == BYDB_ENC_CONST || fenc == BYDB_ENC_DELTA_CONST) {
+            int64_t dd = 0;
+            if (fenc == BYDB_ENC_DELTA_CONST) {
+                int vl;
+                dd = decode_one_varint(fstream, &vl);
+            }
+            uint64_t psum, pcnt;
+            fold_arith_pred(first, dd, a, b, lane, bw0, bw1, bw2, &psum,
+                            &pcnt, &bmn, &bmx);
+            *out_sum = psum;   // wave_reduce_add already wave-uniform
+            *out_cnt = pcnt;
+            *out_mn = bmn;
+            *out_mx = bmx;
+            *out_have_mm = pcnt > 0;
+            return;
+        }
+        if (fenc == BYDB_ENC_DELTA || fenc == BYDB_ENC_DELTA_OF_DELTA) {
+            bool dod = fenc == BYDB_ENC_DELTA_OF_DELTA;
+            const uint8_t *st = fstream;
+            int64_t d1 = 0;
+            if (dod) {
+                int vl;
+                d1 = decode_one_varint(st, &vl);
+                st += vl;
+            }
+            int64_t row1 = (int64_t)((uint64_t)first + (uint64_t)d1);
+            uint64_t lsum = 0, lcnt = 0;
+            int64_t lmn = INT64_MAX, lmx = INT64_MIN;
+            bool need0 = lane == 0 && a <= 0 && 0 <= b;
+            bool m0 = true;
+            if (bw0) m0 = pred_match_rows(bw0, 0, need0 && m0) && m0;
+            if (bw1) m0 = pred_match_rows(bw1, 0, need0 && m0) && m0;
+            if (bw2) m0 = pred_match_rows(bw2, 0, need0 && m0) && m0;
+            if (need0 && m0) {
+                lsum += (uint64_t)first;
+                lcnt++;
+                lmn = first;
+                lmx = first;
+            }
+            if (dod) {
+                bool need1 = lane == 0 && a <= 1 && 1 <= b;
+                bool m1 = true;
+                if (bw0) m1 = pred_match_rows(bw0, 1, need1 && m1) && m1;
+                if (bw1) m1 = pred_match_rows(bw1, 1, need1 && m1) && m1;
+                if (bw2) m1 = pred_match_rows(bw2, 1, need1 && m1) && m1;
+                if (need1 && m1) {
+                    lsum += (uint64_t)row1;
+                    lcnt++;
+                    lmn = row1 < lmn ? row1 : lmn;
+                    lmx = row1 > lmx ? row1 : lmx;
+                }
+            }
+            ScanFold ff;
+            scan_stream(st, n - 1, dod, dod ? row1 : first, d1, a, b,
+                        INT64_MAX, INT64_MIN, lane, &ff, derr, bi, bw0, bw1,
+                        bw2);
+            lsum += ff.sum;
+            lcnt += ff.nsel;
+            lmn = ff.mn < lmn ? ff.mn : lmn;
+            lmx = ff.mx > lmx ? ff.mx : lmx;
+            uint64_t wsum2 = wave_reduce_add(lsum);
+            *out_cnt = wave_reduce_add(lcnt);
+            *out_mn = wave_reduce_min(lmn);
+            *out_mx = wave_reduce_max(lmx);
+            *out_have_mm = *out_cnt > 0;
+            *out_sum = wsum2;
+            return;
+        }
+        dev_set_err(derr, DERR_BAD_ENC, bi);
+        *out_sum = 0;
+        *out_mn = bmn;
+        *out_mx = bmx;
+        *out_have_mm = false;
+        *out_cnt = 0;
         return;
     }
     if (fenc == BYDB_ENC_CONST) {
@@ -2263,13 +2352,16 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
                 for (int sl = 0; sl < n_gslots; sl++)
                     block_group += (int64_t)gb[sl].uniform_gid * gmul[sl];
             } else {
-                // row-varying predicates join the run merge below as
-                // additional RLE cursors; bitmap-mode (plain-tag) walkers
-                // have no runs and stay a loud v1 limit here
-                if (pred_on && ((wp0 && wp0->bm) || (wp1 && wp1->bm) ||
-                                (wp2 && wp2->bm))) {
-                    dev_set_err(derr, DERR_BAD_ENC, (uint64_t)bi);
-                    continue;
+                // row-varying predicates join the run merge below:
+                // RLE walkers as additional run cursors; BITMAP-mode
+                // (plain-tag) walkers have no runs, so they filter
+                // per-row inside fold_range instead (stateless O(1)
+                // lookups — heavy-pass (EN_WALK) instantiations only)
+                PredWalk *bw0 = nullptr, *bw1 = nullptr, *bw2 = nullptr;
+                if (EN_WALK) {
+                    if (wp0 && wp0->bm) { bw0 = wp0; wp0 = nullptr; }
+                    if (wp1 && wp1->bm) { bw1 = wp1; wp1 = nullptr; }
+                    if (wp2 && wp2->bm) { bw2 = wp2; wp2 = nullptr; }
                 }
                 if (seg != 0) continue;  // whole block folded at seg 0
                 const uint8_t *gstream = payload + bd->field_off;
@@ -2365,7 +2457,8 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
                                    bd->field_len, n, aa, bb2,
                                    EN_VALUES && (flags & KF_NEED_VALUES), lane,
                                    &rsum, &rmn, &rmx, &rhave, &rcnt,
-                                   plain_norm, raw_f64, derr, (uint64_t)bi);
+                                   plain_norm, raw_f64, derr, (uint64_t)bi,
+                                   bw0, bw1, bw2);
                         double rfs = 1.0;
                         if ((flags & KF_FLOAT) && fdiff) {
                             int64_t mfac = c_pow10i[fdiff];

@@ -139,20 +139,28 @@ def test_plain_tag_float_column():
     assert math.isclose(g.sum_f, orc.sum_f, rel_tol=1e-9)
 
 
-def test_plain_groupby_fails_loud():
-    """Group-by on a plain column is out of scope in v1 — must be a loud
-    device error, never a silent wrong grouping."""
+def test_plain_groupby_partial_domain():
+    """Group-by on a plain column (landed round 2 —
+    k_resolve_plain_groups): a two-value domain folds exactly the rows
+    carrying those values; everything else drops.  Parity vs the
+    oracle's bytags over the same domain."""
+    import oracle as o
+    from helpers import oracle_blocks
     rng = random.Random(76)
     b = PartBuilder()
     n = 3000
     ts = [T0 + i * MS for i in range(n)]
     b.add_block_i64(1, ts, [1] * n, list(range(n)))
     b.set_block_tag(_users(rng, n))
+    domain = [b"user_001", b"user_002"]
+    payload, blocks = oracle_blocks(b)
+    orc = o.scan_agg_bytags(payload, blocks, VT_INT64, [0], [domain])
     s = Session(0)
     s.upload_part(b)
-    s.configure_by_tag(VT_INT64, [AGG_SUM, AGG_COUNT], 0,
-                       [b"user_001", b"user_002"])
+    s.configure_by_tag(VT_INT64, [AGG_SUM, AGG_COUNT], 0, domain)
     s.consume()
-    with pytest.raises(RuntimeError, match="decode error"):
-        s.finalize()
+    gs = s.finalize()
     s.close()
+    assert sum(oc.count for oc in orc) > 0
+    for g, oc in zip(gs, orc):
+        assert (g.count, g.sum_i) == (oc.count, oc.sum_i)

@@ -271,3 +271,78 @@ class TestGpuPlainGroupBy:
         gs, orc = self.run_both(b, VT_FLOAT64, domain,
                                 float_exp=FLOAT_RAW_EXP)
         cmp_groups(gs, orc, float_mode=True)
+
+
+@pytest.mark.gpu
+class TestBitmapPredInGroupMerge:
+    """Bitmap-mode (plain >256-card tag) predicates combined with
+    row-varying group-by — previously a loud v1 error, now filtered
+    per-row inside fold_range (the reference applies predicates to any
+    column uniformly)."""
+
+    def _part(self, seed, nullable=False):
+        rng = random.Random(seed)
+        ENVS = [b"prod", b"dev", b"qa"]
+        users = [b"user_%03d" % i for i in range(280)]
+        b = PartBuilder()
+        n = 2000
+        for sid in range(4):
+            ts = [T0 + i * MS for i in range(n)]
+            if nullable:
+                b.add_block_i64_nullable(
+                    sid + 1, ts, [1] * n,
+                    [None if rng.random() < 0.15 else
+                     rng.randint(-10**6, 10**6) for _ in range(n)])
+            else:
+                b.add_block_i64(sid + 1, ts, [1] * n,
+                                [rng.randint(-10**6, 10**6)
+                                 for _ in range(n)])
+            tags = []
+            while len(tags) < n:
+                tags.extend([ENVS[rng.randrange(3)]] * rng.randint(1, 40))
+            b.set_block_tag(tags[:n])
+            b.set_block_tag([users[rng.randrange(280)]
+                             for _ in range(n)])  # plain, i.i.d.
+        return b, ENVS, users
+
+    def test_plain_pred_with_dict_group(self):
+        b, ENVS, users = self._part(21)
+        payload, blocks = oracle_blocks(b)
+        orc = o.scan_agg_bytags(payload, blocks, VT_INT64, [0], [ENVS],
+                                preds=[b"", b"user_007", b""])
+        s = Session(0)
+        s.upload_part(b)
+        s.configure_by_tag(VT_INT64, FUNCS, 0, ENVS)
+        s.consume(preds=[b"", b"user_007"])
+        gs = s.finalize()
+        s.close()
+        assert sum(oc.count for oc in orc) > 0
+        cmp_groups(gs, orc)
+
+    def test_plain_pred_with_plain_group(self):
+        """Both the group slot and the predicate slot are plain."""
+        b, ENVS, users = self._part(22)
+        domain = users[:270]
+        payload, blocks = oracle_blocks(b)
+        orc = o.scan_agg_bytags(payload, blocks, VT_INT64, [1], [domain],
+                                preds=[b"prod", b"", b""])
+        s = Session(0)
+        s.upload_part(b)
+        s.configure_by_tag(VT_INT64, FUNCS, 1, domain)
+        s.consume(preds=[b"prod"])
+        gs = s.finalize()
+        s.close()
+        cmp_groups(gs, orc)
+
+    def test_plain_pred_dict_group_nullable_field(self):
+        b, ENVS, users = self._part(23, nullable=True)
+        payload, blocks = oracle_blocks(b)
+        orc = o.scan_agg_bytags(payload, blocks, VT_INT64, [0], [ENVS],
+                                preds=[b"", b"user_012", b""])
+        s = Session(0)
+        s.upload_part(b)
+        s.configure_by_tag(VT_INT64, FUNCS, 0, ENVS)
+        s.consume(preds=[b"", b"user_012"])
+        gs = s.finalize()
+        s.close()
+        cmp_groups(gs, orc)
