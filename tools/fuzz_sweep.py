@@ -95,11 +95,10 @@ def check_scenario2(rng, b, kind, envs, regions):
     if mode == 0 and kind != 2:
         preds = [b"", b"r%d" % rng.randrange(6), b""]
     elif mode == 1 and kind == 2:
+        # bitmap-mode (plain-tag) predicate — combines with scalar AND
+        # row-varying group folds (fold_range per-row filtering)
         preds = [b"user_%03d" % rng.randrange(300), b"", b""]
-    # bitmap-mode (plain-tag) predicates cannot join row-varying group
-    # merges (loud v1 limit) — keep those scenarios scalar
-    force_scalar = kind == 2 and preds is not None
-    if force_scalar or rng.random() < 0.5:
+    if rng.random() < 0.4:
         # scalar (with optional preds)
         orc = o.scan_agg(payload, blocks, vtype, preds=preds)[0]
         s = Session(0)
@@ -110,6 +109,24 @@ def check_scenario2(rng, b, kind, envs, regions):
         g = s.finalize()[0]
         s.close()
         pairs = [(g, orc)]
+    elif kind == 2 and rng.random() < 0.5:
+        # group by the PLAIN high-cardinality tag (slot 0), domain a
+        # random subset; optional region predicate on slot 1
+        dom = [b"user_%03d" % i
+               for i in range(0, 300, rng.choice([1, 2, 3]))]
+        gpreds = None
+        if rng.random() < 0.5:
+            gpreds = [b"", b"r%d" % rng.randrange(6), b""]
+        orc = o.scan_agg_bytags(payload, blocks, vtype, [0], [dom],
+                                preds=gpreds)
+        s = Session(0)
+        s.upload_part(b)
+        s.configure_by_tag(vtype, [AGG_SUM, AGG_COUNT, AGG_MIN, AGG_MAX], 0,
+                           dom, float_exp=fexp)
+        s.consume(preds=[p_ or b"" for p_ in gpreds] if gpreds else None)
+        gs = s.finalize()
+        s.close()
+        pairs = list(zip(gs, orc))
     else:
         # group by regions (slot 1), optional preds on other slots
         orc = o.scan_agg_bytags(payload, blocks, vtype, [1], [regions],
