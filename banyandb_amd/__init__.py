@@ -310,7 +310,11 @@ class PartBuilder:
         n = _lib.bydb_part_builder_payload_len(self._h)
         if n == 0:
             return b""
-        return C.string_at(_lib.bydb_part_builder_payload(self._h), n)
+        ptr = _lib.bydb_part_builder_payload(self._h)
+        # NOT string_at: its size argument is a C int, so payloads past
+        # 2 GiB would truncate (mod 2^32) silently
+        return bytes((C.c_ubyte * n).from_address(
+            C.addressof(ptr.contents)))
 
     @property
     def payload_len(self) -> int:

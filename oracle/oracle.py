@@ -499,3 +499,30 @@ def scan_agg(payload: bytes, blocks, field_vtype, min_ts=INT64_MIN,
     if rc != 0:
         raise ValueError(f"scan_agg failed rc={rc}")
     return list(out)
+
+
+def make_descs(blocks):
+    """Build the oracle-side (BlockDesc * n) array once (zero-copy scans
+    slice it per thread)."""
+    descs = (BlockDesc * len(blocks))()
+    for i, b in enumerate(blocks):
+        for k, v in b.items():
+            setattr(descs[i], k, v)
+    return descs
+
+
+def scan_agg_raw(src_ptr, descs, i0, n_blocks, field_vtype):
+    """Zero-copy whole-range scan for the bench's CPU-baseline leg: the
+    caller owns the payload buffer (the part builder's arena) and a
+    make_descs array; [i0, i0+n_blocks) is this call's slice.  Returns a
+    single AggResult.  The underlying C call releases the GIL, so
+    threads over disjoint slices run on separate cores."""
+    out = (AggResult * 1)()
+    dp = C.cast(C.byref(descs, i0 * C.sizeof(BlockDesc)),
+                C.POINTER(BlockDesc))
+    rc = _L.bo_scan_agg_grouped(src_ptr, dp, n_blocks, field_vtype,
+                                INT64_MIN, INT64_MAX,
+                                C.cast(b"\0", u8p), 0, out, 1)
+    if rc != 0:
+        raise ValueError(f"scan_agg_raw rc={rc}")
+    return out[0]
