@@ -234,6 +234,7 @@ enum {
     DERR_BAD_ENC = 2,       // unsupported encode type in kernel
     DERR_DESC_TS = 3,       // descending timestamps (reference never writes them)
     DERR_GROUP_RANGE = 4,   // block group_code >= configured n_groups
+    DERR_F64_SUM_OVF = 6,   // float64 block mantissa sum would exceed int64
     DERR_EXP_MISMATCH = 5,  // float64 block's decimal exponent differs from
                             // the configured session exponent (per-block
                             // rescale is a later row; silent mixing would
@@ -2229,6 +2230,22 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
                 fdiff = (int)bd->exp - (int)cfge;
                 if (fdiff < 0 || fdiff > 18) {
                     dev_set_err(derr, DERR_EXP_MISMATCH, (uint64_t)bi);
+                    continue;
+                }
+                // mantissa-sum overflow guard: the engine
+                // carries each block's decimal-int mantissa sum in int64
+                // (exact, then one double convert); the reference's
+                // per-row float64 adds never wrap.  A block whose
+                // |firstValue| alone implies the sum exceeds int64
+                // (|mean| > 2^63/8192 ~ 1.1e15 — the realistic overflow
+                // mode is a large baseline) errs loudly instead of
+                // returning a silently wrong sum.  Delta-driven escapes
+                // from a small firstValue remain a documented limit.
+                uint64_t afirst = bd->field_first < 0
+                                      ? (uint64_t)(-(uint64_t)bd->field_first)
+                                      : (uint64_t)bd->field_first;
+                if (afirst > (1ull << 63) / (uint64_t)(bd->count + 1u)) {
+                    dev_set_err(derr, DERR_F64_SUM_OVF, (uint64_t)bi);
                     continue;
                 }
             }

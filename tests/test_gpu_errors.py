@@ -5,7 +5,8 @@ import ctypes as C
 import pytest
 
 import banyandb_amd as ba
-from banyandb_amd import PartBuilder, Session, VT_INT64, AGG_SUM, AGG_COUNT
+from banyandb_amd import (PartBuilder, Session, VT_INT64, VT_FLOAT64,
+                          AGG_SUM, AGG_COUNT)
 
 pytestmark = pytest.mark.gpu
 
@@ -166,3 +167,22 @@ def test_float_mixed_exponent_blocks_rescale():
     assert math.isclose(g.sum_f, orc.sum_f, rel_tol=1e-9)
     assert math.isclose(g.min_f, orc.min_f, rel_tol=1e-12)
     assert math.isclose(g.max_f, orc.max_f, rel_tol=1e-12)
+
+
+def test_f64_mantissa_sum_overflow_is_loud():
+    """A float64 block whose mantissa sum would exceed int64 errs loudly
+    (DERR_F64_SUM_OVF) instead of returning a silently wrapped sum
+    (the reference's per-row float adds never wrap).  Large
+    17-digit mantissas x 8192 rows overflow 2^63."""
+    n = 8192
+    ts = [10 ** 18 + i * 10 ** 6 for i in range(n)]
+    base = 1234567890123456.0   # mantissa 1.23e15 (no trailing zeros)
+    b = PartBuilder()
+    b.add_block_f64(1, ts, [1] * n, [base + i for i in range(n)])
+    s = Session(0)
+    s.upload_part(b)
+    s.configure(VT_FLOAT64, [AGG_SUM, AGG_COUNT], float_exp=0)
+    s.consume()
+    with pytest.raises(RuntimeError, match="device decode error 6"):
+        s.finalize()
+    s.close()
