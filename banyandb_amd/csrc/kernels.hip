@@ -1278,22 +1278,19 @@ struct ScanFold {
     uint64_t sum;       // wrapping sum of selected values
     int64_t mn, mx;     // min/max of selected values
     uint64_t nsel;      // selected row count
-    uint64_t n_lo;      // rows with v < lo_bound (ts clamp use)
-    uint64_t n_hi;      // rows with v > hi_bound
 };
 
 __device__ void scan_stream(const uint8_t *stream, int64_t n_deltas, bool dod,
                             int64_t first_plus /* v at row (dod?1:0) */,
                             int64_t d1_init, int64_t r0, int64_t r1,
-                            int64_t lo_bound, int64_t hi_bound, int lane,
-                            ScanFold *f, DevErr *derr, uint64_t bi,
+                            int lane, ScanFold *f, DevErr *derr, uint64_t bi,
                             PredWalk *pw0, PredWalk *pw1, PredWalk *pw2) {
     uint64_t pos = 0;
     int64_t j = dod ? 2 : 1;
     int64_t jmax = n_deltas;            // always scan the whole stream
     uint64_t v_carry = (uint64_t)first_plus;
     uint64_t d1_carry = (uint64_t)d1_init;
-    uint64_t l_sum = 0, l_nsel = 0, l_nlo = 0, l_nhi = 0;
+    uint64_t l_sum = 0, l_nsel = 0;
     int64_t l_mn = INT64_MAX, l_mx = INT64_MIN;
     // Fixed-size window advance with a cross-window varint carry
     // (carry_u holds the carried low groups, carry_n its byte count), so
@@ -1309,9 +1306,6 @@ __device__ void scan_stream(const uint8_t *stream, int64_t n_deltas, bool dod,
     uint64_t carry_u = 0;
     uint32_t carry_n = 0;
     const bool no_walk = pw0 == nullptr && pw1 == nullptr && pw2 == nullptr;
-    // bound counting is live only on the timestamp-clamp scan; value
-    // folds pass INT64_MAX/MIN and skip the two 64-bit compares per value
-    const bool want_bounds = lo_bound != INT64_MAX || hi_bound != INT64_MIN;
 #define QLEAD 4
     uint32_t wq[QLEAD];  // wq[q] = this lane's u32 of the window at pos+256q
     bool done = false;
@@ -1381,7 +1375,7 @@ __device__ void scan_stream(const uint8_t *stream, int64_t n_deltas, bool dod,
                 int32_t sl = dA + dB + dC + dD;
                 int32_t S = wave_incl_scan32(sl, lane);
                 int32_t cum = S - sl;
-                if (!want_bounds && j >= r0 && j + nt - 1 <= r1) {
+                if (j >= r0 && j + nt - 1 <= r1) {
                     // whole window selected (the dominant shape once the
                     // block-level clamp resolved): no per-value index
                     // bookkeeping — every terminator folds
@@ -1417,10 +1411,6 @@ __device__ void scan_stream(const uint8_t *stream, int64_t n_deltas, bool dod,
                             l_nsel++;                                        \
                             l_mn = sv < l_mn ? sv : l_mn;                    \
                             l_mx = sv > l_mx ? sv : l_mx;                    \
-                        }                                                    \
-                        if (want_bounds) {                                   \
-                            if (sv < lo_bound) l_nlo++;                      \
-                            if (sv > hi_bound) l_nhi++;                      \
                         }                                                    \
                         idx++;                                               \
                     }
@@ -1505,8 +1495,6 @@ __device__ void scan_stream(const uint8_t *stream, int64_t n_deltas, bool dod,
                     l_mn = sv < l_mn ? sv : l_mn;
                     l_mx = sv > l_mx ? sv : l_mx;
                 }
-                if (sv < lo_bound) l_nlo++;
-                if (sv > hi_bound) l_nhi++;
             }
             if (emask >> 63) {
                 carry_n = 0;
@@ -1570,18 +1558,12 @@ __device__ void scan_stream(const uint8_t *stream, int64_t n_deltas, bool dod,
         if (pw0) in_sel = pred_match_rows(pw0, myj, in_sel) && in_sel;
         if (pw1) in_sel = pred_match_rows(pw1, myj, in_sel) && in_sel;
         if (pw2) in_sel = pred_match_rows(pw2, myj, in_sel) && in_sel;
-        {
+        if (in_sel) {
             int64_t sv = (int64_t)val;
-            if (in_sel) {
-                l_sum += val;
-                l_nsel++;
-                l_mn = sv < l_mn ? sv : l_mn;
-                l_mx = sv > l_mx ? sv : l_mx;
-            }
-            if (is_term) {
-                if (sv < lo_bound) l_nlo++;
-                if (sv > hi_bound) l_nhi++;
-            }
+            l_sum += val;
+            l_nsel++;
+            l_mn = sv < l_mn ? sv : l_mn;
+            l_mx = sv > l_mx ? sv : l_mx;
         }
         if (j + nterm > jmax) break;
         // carry out: bytes after the last terminator head the next value
@@ -1609,8 +1591,105 @@ __device__ void scan_stream(const uint8_t *stream, int64_t n_deltas, bool dod,
     f->mn = l_mn;
     f->mx = l_mx;
     f->nsel = l_nsel;
-    f->n_lo = l_nlo;
-    f->n_hi = l_nhi;
+}
+
+// Count rows of a delta / delta-of-delta stream whose value falls below
+// lo_bound (n_lo) or above hi_bound (n_hi) — all the FindRange clamp
+// needs (range.go:143-170; ascending blocks).  This is the ONLY consumer
+// of value bounds, kept OUT of scan_stream so the fold kernels (incl.
+// the closed-form-only headline instantiation, which needs the clamp but
+// not value folds) carry none of the scan's fast-window register state.
+// Correctness path: it runs only on boundary blocks whose ts range
+// partially overlaps the query; generic 64-B ballot windows.  Totals are
+// wave-reduced; all lanes return them.
+__device__ void clamp_count_stream(
+    const uint8_t *stream, int64_t n_deltas,
+                                   bool dod, int64_t first_plus,
+                                   int64_t d1_init, int64_t lo_bound,
+                                   int64_t hi_bound, int lane,
+                                   uint64_t *out_nlo, uint64_t *out_nhi,
+                                   DevErr *derr, uint64_t bi) {
+    uint64_t pos = 0;
+    int64_t j = dod ? 2 : 1;
+    int64_t jmax = n_deltas;
+    uint64_t v_carry = (uint64_t)first_plus;
+    uint64_t d1_carry = (uint64_t)d1_init;
+    uint64_t l_nlo = 0, l_nhi = 0;
+    uint64_t carry_u = 0;
+    uint32_t carry_n = 0;
+    while (j <= jmax) {
+        uint8_t b = stream[pos + (uint64_t)lane];
+        uint64_t emask = __ballot(b < 0x80);
+        if (emask == 0) { dev_set_err(derr, DERR_BAD_STREAM, bi); break; }
+        int rank = __popcll(emask & lanemask_lt(lane));
+        int64_t myj = j + rank;
+        bool is_term = (b < 0x80) && (myj <= jmax);
+        int nterm = __popcll(emask);
+        uint64_t d = 0;
+        if (emask == ~0ull && carry_n == 0) {
+            d = (uint64_t)zz_dec(b);
+        } else {
+            uint64_t below = emask & lanemask_lt(lane);
+            int start = below ? (64 - __clzll(below)) : 0;
+            int mylen = lane - start + 1;
+            if (mylen > 10) mylen = 10;
+            int maxlen = (b < 0x80) ? mylen : 1;
+            for (int off2 = 32; off2 > 0; off2 >>= 1) {
+                int t = __shfl_xor(maxlen, off2);
+                maxlen = t > maxlen ? t : maxlen;
+            }
+            uint64_t u = (uint64_t)(b & 0x7f) << (7 * (mylen - 1));
+            for (int k = 1; k < maxlen; ++k) {
+                uint32_t bk = (uint32_t)__shfl_up((int)b, k);
+                if (k < mylen)
+                    u |= (uint64_t)(bk & 0x7f) << (7 * (mylen - 1 - k));
+            }
+            if (below == 0 && carry_n)
+                u = carry_u | (u << (7 * carry_n));
+            if (is_term) d = (uint64_t)zz_dec(u);
+        }
+        if (!is_term) d = 0;
+        uint64_t val;
+        if (dod) {
+            uint64_t s1 = wave_incl_scan(d, lane);
+            uint64_t d1j = is_term ? (d1_carry + s1) : 0;
+            uint64_t s2 = wave_incl_scan(d1j, lane);
+            val = v_carry + s2;
+            v_carry += readlane64(s2, 63);
+            d1_carry += readlane64(s1, 63);
+        } else {
+            uint64_t sscan = wave_incl_scan(d, lane);
+            val = v_carry + sscan;
+            v_carry += readlane64(sscan, 63);
+        }
+        if (is_term) {
+            int64_t sv = (int64_t)val;
+            if (sv < lo_bound) l_nlo++;
+            if (sv > hi_bound) l_nhi++;
+        }
+        if (j + nterm > jmax) break;
+        int last_lane = 63 - __clzll(emask);
+        int n_tail = 63 - last_lane;
+        if (n_tail) {
+            uint64_t tail = 0;
+            if (lane > last_lane) {
+                int sh2 = lane - last_lane - 1;
+                if (sh2 > 9) sh2 = 9;
+                tail = (uint64_t)(b & 0x7f) << (7 * sh2);
+            }
+            for (int off2 = 32; off2 > 0; off2 >>= 1)
+                tail |= (uint64_t)__shfl_xor((long long)tail, off2);
+            carry_u = tail;
+            carry_n = n_tail > 9 ? 9 : (uint32_t)n_tail;
+        } else {
+            carry_u = 0;
+            carry_n = 0;
+        }
+        j += nterm;
+        pos += 64;
+    }
+    *out_nlo = wave_reduce_add(l_nlo);
+    *out_nhi = wave_reduce_add(l_nhi);
 }
 
 // ---------------- varint segment index ----------------
@@ -2002,8 +2081,7 @@ __device__ void fold_range(const uint8_t *fstream, uint8_t fenc, int64_t first,
             }
             ScanFold ff;
             scan_stream(st, n - 1, dod, dod ? row1 : first, d1, a, b,
-                        INT64_MAX, INT64_MIN, lane, &ff, derr, bi, bw0, bw1,
-                        bw2);
+                        lane, &ff, derr, bi, bw0, bw1, bw2);
             lsum += ff.sum;
             lcnt += ff.nsel;
             lmn = ff.mn < lmn ? ff.mn : lmn;
@@ -2067,8 +2145,8 @@ __device__ void fold_range(const uint8_t *fstream, uint8_t fenc, int64_t first,
                 bmx = vr0 > bmx ? vr0 : bmx;
             } else {
                 ScanFold ff;
-                scan_stream(fstream, n - 1, false, first, 0, a, b, INT64_MAX,
-                            INT64_MIN, lane, &ff, derr, bi, nullptr, nullptr,
+                scan_stream(fstream, n - 1, false, first, 0, a, b,
+                            lane, &ff, derr, bi, nullptr, nullptr,
                             nullptr);
                 int64_t lmn = ff.mn, lmx = ff.mx;
                 if (lane == 0 && a <= 0 && 0 <= b) {
@@ -2092,8 +2170,8 @@ __device__ void fold_range(const uint8_t *fstream, uint8_t fenc, int64_t first,
         if (need_values) {
             int64_t row1 = (int64_t)((uint64_t)first + (uint64_t)d1);
             ScanFold ff;
-            scan_stream(fstream + vl, n - 1, true, row1, d1, a, b, INT64_MAX,
-                        INT64_MIN, lane, &ff, derr, bi, nullptr, nullptr,
+            scan_stream(fstream + vl, n - 1, true, row1, d1, a, b,
+                        lane, &ff, derr, bi, nullptr, nullptr,
                         nullptr);
             int64_t lmn = ff.mn, lmx = ff.mx;
             if (lane == 0) {
@@ -2283,13 +2361,11 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
                     d1 = decode_one_varint(s, &vl);
                     s += vl;
                 }
-                ScanFold tsf;
                 int64_t row1 = (int64_t)((uint64_t)ts_min + (uint64_t)d1);
-                scan_stream(s, n - 1, dod, dod ? row1 : vfirst, d1, 1, 0,
-                            min_ts, max_ts, lane, &tsf, derr, (uint64_t)bi,
-                            nullptr, nullptr, nullptr);
-                uint64_t nlo = wave_reduce_add(tsf.n_lo);
-                uint64_t nhi = wave_reduce_add(tsf.n_hi);
+                uint64_t nlo, nhi;
+                clamp_count_stream(s, n - 1, dod, dod ? row1 : vfirst, d1,
+                                   min_ts, max_ts, lane, &nlo, &nhi, derr,
+                                   (uint64_t)bi);
                 // rows 0 (and 1 for dod) were not in the stream
                 if (ts_min < min_ts) nlo++;
                 if (ts_min > max_ts) nhi++;
@@ -2587,8 +2663,7 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
             }
             ScanFold ff;
             scan_stream(s, n - 1, dod, dod ? row1 : first, d1, r0, r1,
-                        INT64_MAX, INT64_MIN, lane, &ff, derr, (uint64_t)bi,
-                        wp0, wp1, wp2);
+                        lane, &ff, derr, (uint64_t)bi, wp0, wp1, wp2);
             lsum += ff.sum;
             lcnt += ff.nsel;
             lmn = ff.mn < lmn ? ff.mn : lmn;
@@ -2686,8 +2761,8 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
                                        : (int64_t)seg * SEG_ROWS;
                     scan_stream(fstream + e.byte_off, b - rel0, dod,
                                 e.v_start, e.d1_start, a - rel0, b - rel0,
-                                INT64_MAX, INT64_MIN, lane, &ff, derr,
-                                (uint64_t)bi, nullptr, nullptr, nullptr);
+                                lane, &ff, derr, (uint64_t)bi, nullptr,
+                                nullptr, nullptr);
                     lsum = ff.sum;
                     lcnt = ff.nsel;
                     lmn = ff.mn;
@@ -2740,8 +2815,8 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
                 int64_t row1 = (int64_t)((uint64_t)first + (uint64_t)d1);
                 ScanFold ff;
                 scan_stream(s, n - 1, dod, dod ? row1 : first, d1, r0, r1,
-                            INT64_MAX, INT64_MIN, lane, &ff, derr, (uint64_t)bi,
-                            nullptr, nullptr, nullptr);
+                            lane, &ff, derr, (uint64_t)bi, nullptr, nullptr,
+                            nullptr);
                 uint64_t lsum = ff.sum;
                 int64_t lmn = ff.mn, lmx = ff.mx;
                 // rows outside the stream: row 0 (value=first) and, for
