@@ -1157,7 +1157,8 @@ __global__ void k_resolve_preds(
     const bydb_block_desc *__restrict__ blocks, int64_t n_blocks,
     const uint8_t *__restrict__ pred_bytes, uint64_t o0, uint64_t l0,
     uint64_t o1, uint64_t l1, uint64_t o2, uint64_t l2, int n_preds,
-    PredBlock *__restrict__ out, uint8_t *__restrict__ flags) {
+    PredBlock *__restrict__ out, uint8_t *__restrict__ flags,
+    uint32_t *__restrict__ walk_count) {
     int64_t bi = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (bi >= n_blocks) return;
     const bydb_block_desc *bd = &blocks[bi];
@@ -1174,6 +1175,11 @@ __global__ void k_resolve_preds(
         if (v == PF_WALK) f = PF_WALK;
     }
     flags[bi] = f;
+    // walk-block census: lets the heavy scan pass exit immediately when
+    // every predicate verdict is uniform (the dominant entity-tag case)
+    uint64_t wmask = __ballot(f == PF_WALK);
+    if ((threadIdx.x & 63) == 0 && wmask)
+        atomicAdd(walk_count, (uint32_t)__popcll(wmask));
 }
 
 // Second resolve pass for plain (non-dictionary) tag columns: one wave per
@@ -2212,7 +2218,13 @@ __device__ void fold_range(const uint8_t *fstream, uint8_t fenc, int64_t first,
 // (EN_WALK=true, flags & KF_WALK_ONLY) that folds only the PF_WALK
 // blocks.  When no block needs a walker the heavy pass is a ~flag-byte
 // sweep.
-template <bool EN_VALUES, bool EN_PREDS, bool EN_GROUPS, bool EN_WALK>
+// EN_CLAMP — compile the FindRange row clamp only when the query's
+// [min_ts, max_ts] does NOT already cover the whole resident part (the
+// host tracks the part's ts extent and selects the instantiation).  The
+// clamp's scan machinery otherwise inflates register pressure enough to
+// spill scratch reloads into the dense fold loop (~7% on the headline).
+template <bool EN_VALUES, bool EN_PREDS, bool EN_GROUPS, bool EN_WALK,
+          bool EN_CLAMP>
 // Occupancy: 4 waves/SIMD for the value-scan instantiations — requesting
 // 6 forces scratch spills into the window loop and measured ~1.6x SLOWER;
 // the closed-form-only instantiation fits 6 without spills.
@@ -2228,7 +2240,12 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
     const uint16_t *__restrict__ gmap_in,
     const uint16_t *__restrict__ gruns_in, int n_gslots, int64_t gm0,
     int64_t gm1, int64_t gm2, int64_t n_groups,
-    bydb_partial *__restrict__ partials, DevErr *derr) {
+    bydb_partial *__restrict__ partials, DevErr *derr,
+    const uint32_t *__restrict__ walk_count) {
+    // heavy pass with no PF_WALK blocks anywhere: one scalar load, done
+    if ((flags & KF_WALK_ONLY) && walk_count != nullptr &&
+        *walk_count == 0)
+        return;
     const int64_t gmul[3] = {gm0, gm1, gm2};
     const PredBlock *preds = EN_PREDS ? preds_in : nullptr;
     const SegEntry *segs = EN_VALUES ? segs_in : nullptr;
@@ -2332,7 +2349,7 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
         // ---- row clamp (timestamp.FindRange, range.go:143-170) ----
         int64_t r0 = 0, r1 = n - 1;
         if (ts_min > ts_max) { dev_set_err(derr, DERR_DESC_TS, (uint64_t)bi); continue; }
-        if (!(min_ts <= ts_min && max_ts >= ts_max)) {
+        if (EN_CLAMP && !(min_ts <= ts_min && max_ts >= ts_max)) {
             if (ts_min > max_ts || ts_max < min_ts) continue;  // no overlap
             uint8_t tenc = bd->ts_enc_with_version;
             // common type (encoding.GetCommonType)
@@ -2989,6 +3006,7 @@ struct bydb_session {
     // part residency
     uint8_t *d_payload = nullptr;
     uint64_t payload_cap = 0, payload_len = 0;
+    int64_t part_ts_min = INT64_MAX, part_ts_max = INT64_MIN;
     bydb_block_desc *d_blocks = nullptr;
     int64_t blocks_cap = 0, n_blocks = 0;
     // aggregation config
@@ -3036,6 +3054,7 @@ struct bydb_session {
     uint64_t pred_bm_cap = 0;  // words
     uint8_t *d_pred_flags = nullptr;
     int64_t pred_flags_cap = 0;
+    uint32_t *d_walk_count = nullptr;
     float last_ms = 0.0f;
     bool consumed = false;
     int16_t float_exp = 0;  // shared decimal exponent for float64 restore
@@ -3102,6 +3121,7 @@ extern "C" void bydb_session_destroy(bydb_session *s) {
     if (s->d_plain_ctr) (void)hipFree(s->d_plain_ctr);
     if (s->d_pred_bm) (void)hipFree(s->d_pred_bm);
     if (s->d_pred_flags) (void)hipFree(s->d_pred_flags);
+    if (s->d_walk_count) (void)hipFree(s->d_walk_count);
     if (s->ev_start) (void)hipEventDestroy(s->ev_start);
     if (s->ev_stop) (void)hipEventDestroy(s->ev_stop);
     if (s->stream) (void)hipStreamDestroy(s->stream);
@@ -3127,6 +3147,8 @@ extern "C" int bydb_part_reserve(bydb_session *s, uint64_t payload_bytes,
     s->blocks_cap = n_blocks;
     s->payload_len = 0;
     s->n_blocks = 0;
+    s->part_ts_min = INT64_MAX;
+    s->part_ts_max = INT64_MIN;
     s->sidecar_len = 0;
     s->n_plain_host = 0;
     s->segs_built = false;
@@ -3155,6 +3177,8 @@ extern "C" int bydb_part_append(bydb_session *s, const uint8_t *payload,
     const uint64_t part_end = s->payload_len + len;
     for (int64_t i = 0; i < n_blocks; i++) {
         const bydb_block_desc *bd = &blocks[i];
+        if (bd->ts_min < s->part_ts_min) s->part_ts_min = bd->ts_min;
+        if (bd->ts_max > s->part_ts_max) s->part_ts_max = bd->ts_max;
         // loud host-side validation: every stream the descriptor names
         // must lie inside the part appended so far (catches builders whose
         // offset base drifted from this session's arena)
@@ -3258,6 +3282,8 @@ extern "C" int bydb_part_append(bydb_session *s, const uint8_t *payload,
 extern "C" int bydb_part_clear(bydb_session *s) {
     s->payload_len = 0;
     s->n_blocks = 0;
+    s->part_ts_min = INT64_MAX;
+    s->part_ts_max = INT64_MIN;
     s->sidecar_len = 0;
     s->n_plain_host = 0;
     return BYDB_OK;
@@ -3414,9 +3440,14 @@ extern "C" int bydb_consume_multi(bydb_session *s, int64_t min_ts,
         }
         if (s->pred_flags_cap < s->n_blocks) {
             if (s->d_pred_flags) (void)hipFree(s->d_pred_flags);
+    if (s->d_walk_count) (void)hipFree(s->d_walk_count);
             HIP_TRY(s, hipMalloc(&s->d_pred_flags, (size_t)s->n_blocks));
             s->pred_flags_cap = s->n_blocks;
         }
+        if (!s->d_walk_count)
+            HIP_TRY(s, hipMalloc(&s->d_walk_count, sizeof(uint32_t)));
+        HIP_TRY(s, hipMemsetAsync(s->d_walk_count, 0, sizeof(uint32_t),
+                                  s->stream));
         uint64_t total = 0;
         for (int i = 0; i < n_preds; i++) total += pred_lens[i];
         if (s->pred_bytes_cap < total) {
@@ -3443,7 +3474,7 @@ extern "C" int bydb_consume_multi(bydb_session *s, int64_t min_ts,
                            s->stream, s->d_payload, s->d_sidecar, s->d_blocks,
                            s->n_blocks, s->d_pred_bytes, offs[0], lens3[0],
                            offs[1], lens3[1], offs[2], lens3[2], n_preds,
-                           s->d_preds, s->d_pred_flags);
+                           s->d_preds, s->d_pred_flags, s->d_walk_count);
         HIP_TRY(s, hipGetLastError());
         // plain (non-dictionary) columns present: the host counted every
         // normalized (block,slot) stream at part_append, so the bitmap
@@ -3580,35 +3611,47 @@ extern "C" int bydb_consume_multi(bydb_session *s, int64_t min_ts,
                 const uint64_t *, const uint8_t *, const SegEntry *,
                 const GroupBlock *, const uint16_t *, const uint16_t *, int,
                 int64_t, int64_t, int64_t, int64_t, bydb_partial *,
-                DevErr *);
+                DevErr *, const uint32_t *);
     void (*kfn_walk)(const uint8_t *, const uint8_t *,
                      const bydb_block_desc *, int64_t, int64_t, int64_t, int,
                      const PredBlock *, int, const uint64_t *,
                      const uint8_t *, const SegEntry *, const GroupBlock *,
                      const uint16_t *, const uint16_t *, int, int64_t,
-                     int64_t, int64_t, int64_t, bydb_partial *,
-                     DevErr *) = nullptr;
-    if (en_values) {
-        if (en_preds) {
-            kfn = en_groups ? k_scan_agg_t<true, true, true, false>
-                            : k_scan_agg_t<true, true, false, false>;
-            kfn_walk = en_groups ? k_scan_agg_t<true, true, true, true>
-                                 : k_scan_agg_t<true, true, false, true>;
-        } else {
-            kfn = en_groups ? k_scan_agg_t<true, false, true, false>
-                            : k_scan_agg_t<true, false, false, false>;
-        }
-    } else {
-        if (en_preds) {
-            kfn = en_groups ? k_scan_agg_t<false, true, true, false>
-                            : k_scan_agg_t<false, true, false, false>;
-            kfn_walk = en_groups ? k_scan_agg_t<false, true, true, true>
-                                 : k_scan_agg_t<false, true, false, true>;
-        } else {
-            kfn = en_groups ? k_scan_agg_t<false, false, true, false>
-                            : k_scan_agg_t<false, false, false, false>;
-        }
+                     int64_t, int64_t, int64_t, bydb_partial *, DevErr *,
+                     const uint32_t *) = nullptr;
+    // the clamp-free instantiations run when the query range covers the
+    // whole resident part (the common analytic scan; per-block FindRange
+    // is then an identity)
+    const bool en_clamp = !(s->n_blocks > 0 && min_ts <= s->part_ts_min &&
+                            max_ts >= s->part_ts_max);
+#define BYDB_KSEL(C)                                                         \
+    if (en_values) {                                                         \
+        if (en_preds) {                                                      \
+            kfn = en_groups ? k_scan_agg_t<true, true, true, false, C>       \
+                            : k_scan_agg_t<true, true, false, false, C>;     \
+            kfn_walk = en_groups ? k_scan_agg_t<true, true, true, true, C>   \
+                                 : k_scan_agg_t<true, true, false, true, C>; \
+        } else {                                                             \
+            kfn = en_groups ? k_scan_agg_t<true, false, true, false, C>      \
+                            : k_scan_agg_t<true, false, false, false, C>;    \
+        }                                                                    \
+    } else {                                                                 \
+        if (en_preds) {                                                      \
+            kfn = en_groups ? k_scan_agg_t<false, true, true, false, C>      \
+                            : k_scan_agg_t<false, true, false, false, C>;    \
+            kfn_walk = en_groups ? k_scan_agg_t<false, true, true, true, C>  \
+                                 : k_scan_agg_t<false, true, false, true, C>;\
+        } else {                                                             \
+            kfn = en_groups ? k_scan_agg_t<false, false, true, false, C>     \
+                            : k_scan_agg_t<false, false, false, false, C>;   \
+        }                                                                    \
     }
+    if (en_clamp) {
+        BYDB_KSEL(true)
+    } else {
+        BYDB_KSEL(false)
+    }
+#undef BYDB_KSEL
     // light pass: every block whose predicate verdict is uniform (no
     // walker state, no scratch spills in the scan loop)
     hipLaunchKernelGGL(kfn, dim3(grid), dim3(threads), 0, s->stream,
@@ -3616,7 +3659,7 @@ extern "C" int bydb_consume_multi(bydb_session *s, int64_t min_ts,
                        min_ts, max_ts, flags, preds, n_preds, s->d_pred_bm,
                        s->d_pred_flags, segs, groups, s->d_gmap, s->d_gruns,
                        s->n_gslots, s->gmul[0], s->gmul[1], s->gmul[2],
-                       (int64_t)s->n_groups, s->d_acc, s->d_err);
+                       (int64_t)s->n_groups, s->d_acc, s->d_err, nullptr);
     HIP_TRY(s, hipGetLastError());
     // heavy pass: only the PF_WALK blocks (row-varying predicates).  When
     // none exist this is a flag-byte sweep (~tens of us)
@@ -3627,7 +3670,8 @@ extern "C" int bydb_consume_multi(bydb_session *s, int64_t min_ts,
                            preds, n_preds, s->d_pred_bm, s->d_pred_flags,
                            segs, groups, s->d_gmap, s->d_gruns, s->n_gslots,
                            s->gmul[0], s->gmul[1], s->gmul[2],
-                           (int64_t)s->n_groups, s->d_acc, s->d_err);
+                           (int64_t)s->n_groups, s->d_acc, s->d_err,
+                           s->d_walk_count);
         HIP_TRY(s, hipGetLastError());
     }
     HIP_TRY(s, hipEventRecord(s->ev_stop, s->stream));
