@@ -145,6 +145,9 @@ def _load():
     lib.bydb_set_float_exp.argtypes = [C.c_void_p, C.c_int16]
     lib.bydb_last_consume_ms.restype = C.c_double
     lib.bydb_last_consume_ms.argtypes = [C.c_void_p]
+    lib.bydb_group_first_seen.restype = C.c_int
+    lib.bydb_group_first_seen.argtypes = [C.c_void_p, C.POINTER(C.c_uint64),
+                                          C.c_int64]
     lib.bydb_reduce_partials2.restype = C.c_int
     lib.bydb_reduce_partials2.argtypes = [C.POINTER(Partial), C.c_int64, C.c_int64,
                                           C.c_int, C.c_int16, C.POINTER(Result)]
@@ -484,6 +487,19 @@ class Session:
 
     def last_consume_ms(self):
         return _lib.bydb_last_consume_ms(self._h)
+
+    def group_first_seen(self):
+        """Per-group first-seen keys; sorting group ids by them gives the
+        reference first-seen materialisation order (2^64-1 = group never
+        entered)."""
+        out = (C.c_uint64 * self.n_groups)()
+        self._ck(_lib.bydb_group_first_seen(self._h, out, self.n_groups))
+        return list(out)
+
+    def groups_in_first_seen_order(self):
+        ks = self.group_first_seen()
+        return [g for g, k in sorted(enumerate(ks), key=lambda t: t[1])
+                if ks[g] != (1 << 64) - 1]
 
     def close(self):
         if self._h:

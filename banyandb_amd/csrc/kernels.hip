@@ -1933,13 +1933,25 @@ enum {
                          // the light/heavy predicate split below)
 };
 
+// wfirst: the (item << 13 | row) key where this wave first ENTERED the
+// group (rows that pass predicates and map to the group, before the
+// null-field check — matching where the reference's computeKey
+// materialises a group, aggregation.go:523).  The global atomicMin over
+// these keys is the group's first appearance in storage order, which is
+// the reference's first-seen materialisation order (blocks iterate in
+// (sid, minTs) order, rows ascend).  ~0 = no entry this flush.
 __device__ __forceinline__ void flush_partial(bydb_partial *partials,
                                               int64_t group, uint64_t wsum,
                                               uint64_t wcnt, int64_t wmin,
                                               int64_t wmax, double wsumf,
-                                              int lane) {
+                                              int lane,
+                                              uint64_t *first_seen,
+                                              uint64_t wfirst) {
     if (group < 0 || lane != 0) return;
     bydb_partial *p = &partials[group];
+    if (first_seen != nullptr && wfirst != ~0ull)
+        atomicMin((unsigned long long *)&first_seen[group],
+                  (unsigned long long)wfirst);
     if (wsum) atomicAdd((unsigned long long *)&p->sum_i, (unsigned long long)wsum);
     if (wcnt) atomicAdd((unsigned long long *)&p->count, (unsigned long long)wcnt);
     if (wmin != INT64_MAX) atomicMin((long long *)&p->min_i, (long long)wmin);
@@ -2241,7 +2253,8 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
     const uint16_t *__restrict__ gruns_in, int n_gslots, int64_t gm0,
     int64_t gm1, int64_t gm2, int64_t n_groups,
     bydb_partial *__restrict__ partials, DevErr *derr,
-    const uint32_t *__restrict__ walk_count) {
+    const uint32_t *__restrict__ walk_count,
+    uint64_t *__restrict__ first_seen) {
     // heavy pass with no PF_WALK blocks anywhere: one scalar load, done
     if ((flags & KF_WALK_ONLY) && walk_count != nullptr &&
         *walk_count == 0)
@@ -2258,6 +2271,7 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
     // per-wave partial (flushed on group change / at the end)
     int64_t cur_group = -1;
     uint64_t wsum = 0, wcnt = 0;
+    uint64_t wfirst = ~0ull;   // first (item,row) entry key of cur_group
     int64_t wmin = INT64_MAX, wmax = INT64_MIN;
     double wsumf = 0.0;
 
@@ -2588,10 +2602,14 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
                         }
                         if (comp != cur_group) {
                             flush_partial(partials, cur_group, wsum, wcnt,
-                                          wmin, wmax, wsumf, lane);
+                                          wmin, wmax, wsumf, lane,
+                                          first_seen, wfirst);
                             cur_group = comp;
                             wsum = 0; wcnt = 0; wmin = INT64_MAX;
                             wmax = INT64_MIN; wsumf = 0;
+                            wfirst = ((uint64_t)wi << 13) | (uint64_t)aa;
+                        } else if (wfirst == ~0ull) {
+                            wfirst = ((uint64_t)wi << 13) | (uint64_t)aa;
                         }
                         wsum += rsum;
                         wcnt += rcnt;
@@ -2959,9 +2977,12 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
         }
         if (block_group != cur_group) {
             flush_partial(partials, cur_group, wsum, wcnt, wmin, wmax, wsumf,
-                          lane);
+                          lane, first_seen, wfirst);
             cur_group = block_group;
             wsum = 0; wcnt = 0; wmin = INT64_MAX; wmax = INT64_MIN; wsumf = 0;
+            wfirst = ((uint64_t)wi << 13) | (uint64_t)r0;
+        } else if (wfirst == ~0ull) {
+            wfirst = ((uint64_t)wi << 13) | (uint64_t)r0;
         }
         wsum += bsum;
         wcnt += nsel_eff;
@@ -2980,7 +3001,8 @@ __global__ __launch_bounds__(256, (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) 
             }
         }
     }
-    flush_partial(partials, cur_group, wsum, wcnt, wmin, wmax, wsumf, lane);
+    flush_partial(partials, cur_group, wsum, wcnt, wmin, wmax, wsumf, lane,
+                  first_seen, wfirst);
 }
 
 // init kernel: set partials to the fold identity (Map.Reset, function.go)
@@ -3055,6 +3077,8 @@ struct bydb_session {
     uint8_t *d_pred_flags = nullptr;
     int64_t pred_flags_cap = 0;
     uint32_t *d_walk_count = nullptr;
+    uint64_t *d_first_seen = nullptr;
+    uint64_t first_seen_cap = 0;
     float last_ms = 0.0f;
     bool consumed = false;
     int16_t float_exp = 0;  // shared decimal exponent for float64 restore
@@ -3122,6 +3146,7 @@ extern "C" void bydb_session_destroy(bydb_session *s) {
     if (s->d_pred_bm) (void)hipFree(s->d_pred_bm);
     if (s->d_pred_flags) (void)hipFree(s->d_pred_flags);
     if (s->d_walk_count) (void)hipFree(s->d_walk_count);
+    if (s->d_first_seen) (void)hipFree(s->d_first_seen);
     if (s->ev_start) (void)hipEventDestroy(s->ev_start);
     if (s->ev_stop) (void)hipEventDestroy(s->ev_stop);
     if (s->stream) (void)hipStreamDestroy(s->stream);
@@ -3303,6 +3328,11 @@ extern "C" int bydb_agg_configure(bydb_session *s, int field_vtype,
         HIP_TRY(s, hipMalloc(&s->d_partials, sizeof(bydb_partial) * n_groups));
         s->partials_cap = n_groups;
     }
+    if (s->first_seen_cap < n_groups) {
+        if (s->d_first_seen) (void)hipFree(s->d_first_seen);
+        HIP_TRY(s, hipMalloc(&s->d_first_seen, sizeof(uint64_t) * n_groups));
+        s->first_seen_cap = n_groups;
+    }
     s->d_acc = s->d_partials;
     s->n_gslots = 0;
     return bydb_reset(s);
@@ -3415,6 +3445,9 @@ extern "C" int bydb_reset(bydb_session *s) {
     hipLaunchKernelGGL(k_reset_partials, dim3(blocks), dim3(threads), 0,
                        s->stream, s->d_acc, (int64_t)s->n_groups);
     HIP_TRY(s, hipGetLastError());
+    if (s->d_first_seen)
+        HIP_TRY(s, hipMemsetAsync(s->d_first_seen, 0xFF,
+                                  sizeof(uint64_t) * s->n_groups, s->stream));
     HIP_TRY(s, hipMemsetAsync(s->d_err, 0, sizeof(DevErr), s->stream));
     s->consumed = false;
     return BYDB_OK;
@@ -3441,6 +3474,7 @@ extern "C" int bydb_consume_multi(bydb_session *s, int64_t min_ts,
         if (s->pred_flags_cap < s->n_blocks) {
             if (s->d_pred_flags) (void)hipFree(s->d_pred_flags);
     if (s->d_walk_count) (void)hipFree(s->d_walk_count);
+    if (s->d_first_seen) (void)hipFree(s->d_first_seen);
             HIP_TRY(s, hipMalloc(&s->d_pred_flags, (size_t)s->n_blocks));
             s->pred_flags_cap = s->n_blocks;
         }
@@ -3611,14 +3645,14 @@ extern "C" int bydb_consume_multi(bydb_session *s, int64_t min_ts,
                 const uint64_t *, const uint8_t *, const SegEntry *,
                 const GroupBlock *, const uint16_t *, const uint16_t *, int,
                 int64_t, int64_t, int64_t, int64_t, bydb_partial *,
-                DevErr *, const uint32_t *);
+                DevErr *, const uint32_t *, uint64_t *);
     void (*kfn_walk)(const uint8_t *, const uint8_t *,
                      const bydb_block_desc *, int64_t, int64_t, int64_t, int,
                      const PredBlock *, int, const uint64_t *,
                      const uint8_t *, const SegEntry *, const GroupBlock *,
                      const uint16_t *, const uint16_t *, int, int64_t,
                      int64_t, int64_t, int64_t, bydb_partial *, DevErr *,
-                     const uint32_t *) = nullptr;
+                     const uint32_t *, uint64_t *) = nullptr;
     // the clamp-free instantiations run when the query range covers the
     // whole resident part (the common analytic scan; per-block FindRange
     // is then an identity)
@@ -3659,7 +3693,8 @@ extern "C" int bydb_consume_multi(bydb_session *s, int64_t min_ts,
                        min_ts, max_ts, flags, preds, n_preds, s->d_pred_bm,
                        s->d_pred_flags, segs, groups, s->d_gmap, s->d_gruns,
                        s->n_gslots, s->gmul[0], s->gmul[1], s->gmul[2],
-                       (int64_t)s->n_groups, s->d_acc, s->d_err, nullptr);
+                       (int64_t)s->n_groups, s->d_acc, s->d_err, nullptr,
+                       s->d_first_seen);
     HIP_TRY(s, hipGetLastError());
     // heavy pass: only the PF_WALK blocks (row-varying predicates).  When
     // none exist this is a flag-byte sweep (~tens of us)
@@ -3671,7 +3706,7 @@ extern "C" int bydb_consume_multi(bydb_session *s, int64_t min_ts,
                            segs, groups, s->d_gmap, s->d_gruns, s->n_gslots,
                            s->gmul[0], s->gmul[1], s->gmul[2],
                            (int64_t)s->n_groups, s->d_acc, s->d_err,
-                           s->d_walk_count);
+                           s->d_walk_count, s->d_first_seen);
         HIP_TRY(s, hipGetLastError());
     }
     HIP_TRY(s, hipEventRecord(s->ev_stop, s->stream));
@@ -3769,6 +3804,29 @@ extern "C" int bydb_set_float_exp(bydb_session *s, int16_t exp) {
 extern "C" int bydb_finalize_partials(bydb_session *s, bydb_partial *out,
                                       int64_t n_groups) {
     return finalize_common(s, out, n_groups);
+}
+
+/* Per-group first-seen keys ((item << 13) | row in storage order; ~0 =
+ * group never entered).  Sorting group ids by these keys reproduces the
+ * reference's first-seen group materialisation order (computeKey
+ * creates groups in row-iteration order, aggregation.go:523), so a Go
+ * twin can emit NextBatch rows in reference order from the host-domain
+ * buffers.  Caveats (documented in DESIGN.md): a group whose only rows
+ * fail a BITMAP-mode (plain-tag) predicate inside a row-varying merge
+ * may still record the run-entry key; within one (item,row) tie the
+ * order is exact. */
+extern "C" int bydb_group_first_seen(bydb_session *s, uint64_t *out,
+                                     int64_t n_groups) {
+    HIP_TRY(s, hipSetDevice(s->device));
+    if ((uint32_t)n_groups != s->n_groups || !s->d_first_seen) {
+        s->err = "n_groups mismatch or not configured";
+        return BYDB_ERR_BAD_ARG;
+    }
+    HIP_TRY(s, hipStreamSynchronize(s->stream));
+    HIP_TRY(s, hipMemcpy(out, s->d_first_seen,
+                         sizeof(uint64_t) * (size_t)n_groups,
+                         hipMemcpyDeviceToHost));
+    return BYDB_OK;
 }
 
 static void partial_to_result(const bydb_partial *p, int field_vtype,

@@ -189,6 +189,14 @@ int bydb_finalize_partials(bydb_session *s, bydb_partial *out, int64_t n_groups)
 /* Reset accumulators to the fold identity (Map.Reset, function.go). */
 int bydb_reset(bydb_session *s);
 
+/* Per-group first-seen keys ((item << 13) | row in storage order; ~0 =
+ * group never entered).  Sorting group ids by these keys reproduces the
+ * reference's first-seen group materialisation order (computeKey creates
+ * groups in row-iteration order, vectorized/measure/aggregation.go:523),
+ * so the Go twin can emit NextBatch rows in reference order from the
+ * host-domain result buffers.  out holds n_groups entries. */
+int bydb_group_first_seen(bydb_session *s, uint64_t *out, int64_t n_groups);
+
 /* Combine external partials (AggModeReduce Combine semantics,
  * aggregation_reduce.go:120-138) into results on the host.  parts is
  * [n_parts_per_group][n_groups]; float_exp selects the float64 restore
