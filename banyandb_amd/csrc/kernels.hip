@@ -3473,8 +3473,6 @@ extern "C" int bydb_consume_multi(bydb_session *s, int64_t min_ts,
         }
         if (s->pred_flags_cap < s->n_blocks) {
             if (s->d_pred_flags) (void)hipFree(s->d_pred_flags);
-    if (s->d_walk_count) (void)hipFree(s->d_walk_count);
-    if (s->d_first_seen) (void)hipFree(s->d_first_seen);
             HIP_TRY(s, hipMalloc(&s->d_pred_flags, (size_t)s->n_blocks));
             s->pred_flags_cap = s->n_blocks;
         }
