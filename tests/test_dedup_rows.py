@@ -72,3 +72,35 @@ def test_dedup_nullable_and_f64():
     r2 = oracle_scan(b2, ba.VT_FLOAT64)[0]
     assert r2.count == 2
     assert r2.sum_f == 7.5 + 2.25
+
+
+def test_dedup_property_random():
+    """Property: for random (ts, version) multisets the surviving rows
+    equal the reference model — sort by (ts asc, version desc) stably,
+    keep the first row per timestamp (part.go:178 + datapoints.go:189-197
+    + the :192-198 skip loop)."""
+    import random as _r
+    for seed in range(40):
+        rng = _r.Random(1000 + seed)
+        n = rng.randint(1, 60)
+        ts = [T0 + rng.randint(0, 9) * MS for _ in range(n)]
+        ver = [rng.randint(0, 4) for _ in range(n)]
+        vals = list(range(n))
+        # reference model
+        order = sorted(range(n), key=lambda i: (ts[i], -ver[i]))
+        seen, keep = set(), []
+        for i in order:
+            if ts[i] not in seen:
+                seen.add(ts[i])
+                keep.append(i)
+        want = sorted((ts[i], vals[i]) for i in keep)
+
+        b = ba.PartBuilder()
+        b.add_block_i64(1, ts, ver, vals)
+        d = b.blocks()[0]
+        assert d.count == len(keep), (seed, d.count, len(keep))
+        r = oracle_scan(b, ba.VT_INT64)[0]
+        assert r.count == len(keep)
+        assert r.sum_i == sum(v for _, v in want), seed
+        assert r.min_i == min(v for _, v in want)
+        assert r.max_i == max(v for _, v in want)

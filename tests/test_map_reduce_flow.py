@@ -111,3 +111,27 @@ def test_reduce_frames_float_min_max_and_nulls():
     (k0, v0), (k1, v1) = out
     assert v0[0][1] == 3.25 and v0[1][1] == 5.5
     assert v1[0][1] == -2.0 and v1[1][1] == -2.0  # null row skipped
+
+
+def test_reduce_frames_edge_cases():
+    """Edge contracts of the product reduce: zero frames -> zero groups;
+    capacity overflow -> loud BYDB_ERR_OOM; numeric key columns."""
+    import pytest as _pytest
+    assert reduce_frames([], specs=[(2, ba.AGG_SUM)], key_cols=[1],
+                         shard_col=0) == []
+
+    def frame(vals, keys):
+        fb = FrameBuilder(len(vals))
+        fb.add_i64(ROLE_SHARD, "shard_id", "", [0] * len(vals))
+        fb.add_i64(ROLE_TAG, "k", "", keys)        # NUMERIC key column
+        fb.add_i64(ROLE_FIELD, "value", "", vals)
+        return fb.finish()
+
+    f = frame([5, 7, 9], [100, -3, 100])
+    out = reduce_frames([f], specs=[(2, ba.AGG_SUM)], key_cols=[1],
+                        shard_col=0)
+    assert [v[0][0] for _, v in out] == [14, 7]   # 100 combines, -3 alone
+
+    with _pytest.raises(RuntimeError):
+        reduce_frames([f], specs=[(2, ba.AGG_SUM)], key_cols=[1],
+                      shard_col=0, out_cap=1)
