@@ -127,10 +127,13 @@ def test_reduce_frames_edge_cases():
         fb.add_i64(ROLE_FIELD, "value", "", vals)
         return fb.finish()
 
+    # two rows with the SAME (shard, key) in one frame: the second is a
+    # replica duplicate and drops (markDedupSeen — a map shard emits one
+    # row per group, so a repeat can only be a replica)
     f = frame([5, 7, 9], [100, -3, 100])
     out = reduce_frames([f], specs=[(2, ba.AGG_SUM)], key_cols=[1],
                         shard_col=0)
-    assert [v[0][0] for _, v in out] == [14, 7]   # 100 combines, -3 alone
+    assert [v[0][0] for _, v in out] == [5, 7]
 
     with _pytest.raises(RuntimeError):
         reduce_frames([f], specs=[(2, ba.AGG_SUM)], key_cols=[1],
