@@ -127,16 +127,51 @@ def check_scenario2(rng, b, kind, envs, regions):
         gs = s.finalize()
         s.close()
         pairs = list(zip(gs, orc))
+    elif rng.random() < 0.3:
+        # COMPOSITE key over (env slot 0, regions slot 1) — when slot 0
+        # is plain (kind==2) this mixes a plain key slot with a
+        # dictionary one; optional ts clamp
+        doms = [[b"prod", b"dev", b"staging", b"qa"], regions]
+        if kind == 2:
+            doms[0] = [b"user_%03d" % i for i in range(0, 300, 2)]
+        lo = T0 + rng.randint(0, 500) * MS
+        hi = T0 + rng.randint(500, 9000) * MS
+        orc = o.scan_agg_bytags(payload, blocks, vtype, [0, 1], doms,
+                                min_ts=lo, max_ts=hi)
+        s = Session(0)
+        s.upload_part(b)
+        s.configure_by_tags(vtype, [AGG_SUM, AGG_COUNT, AGG_MIN, AGG_MAX],
+                            [0, 1], doms, float_exp=fexp)
+        s.consume(min_ts=lo, max_ts=hi)
+        gs = s.finalize()
+        fsk = s.group_first_seen()
+        s.close()
+        for g in range(len(gs)):
+            if gs[g].count > 0:
+                assert fsk[g] != (1 << 64) - 1, "first-seen missing"
+        pairs = list(zip(gs, orc))
     else:
-        # group by regions (slot 1), optional preds on other slots
+        # group by regions (slot 1), optional preds on other slots;
+        # sometimes MODE_MAP + host reduce_partials2 instead of ALL
         orc = o.scan_agg_bytags(payload, blocks, vtype, [1], [regions],
                                 preds=preds)
         s = Session(0)
         s.upload_part(b)
-        s.configure_by_tag(vtype, [AGG_SUM, AGG_COUNT, AGG_MIN, AGG_MAX], 1,
-                           regions, float_exp=fexp)
-        s.consume(preds=preds)
-        gs = s.finalize()
+        if rng.random() < 0.3:
+            from banyandb_amd import MODE_MAP, reduce_partials
+            s.configure_by_tag(vtype,
+                               [AGG_SUM, AGG_COUNT, AGG_MIN, AGG_MAX], 1,
+                               regions, mode=MODE_MAP, float_exp=fexp)
+            s.consume(preds=preds)
+            parts = s.finalize_partials()
+            gs = reduce_partials(parts, 1, len(regions), vtype,
+                                 float_exp=fexp)
+        else:
+            s.configure_by_tag(vtype,
+                               [AGG_SUM, AGG_COUNT, AGG_MIN, AGG_MAX], 1,
+                               regions, float_exp=fexp)
+            s.consume(preds=preds)
+            gs = s.finalize()
         s.close()
         pairs = list(zip(gs, orc))
     for g, oc in pairs:
