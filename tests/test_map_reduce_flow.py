@@ -138,3 +138,56 @@ def test_reduce_frames_edge_cases():
     with _pytest.raises(RuntimeError):
         reduce_frames([f], specs=[(2, ba.AGG_SUM)], key_cols=[1],
                       shard_col=0, out_cap=1)
+
+
+def test_reduce_frames_property_random():
+    """Property: random frames (random shard ids, keys, values, nulls,
+    replicas) reduce exactly like a Python model of
+    aggregation_reduce.go's dedup + Combine + Val."""
+    import random as _r
+    for seed in range(30):
+        rng = _r.Random(7000 + seed)
+        n_frames = rng.randint(1, 5)
+        keyspace = [b"k%d" % i for i in range(rng.randint(1, 6))]
+        frames, rows = [], []
+        for _ in range(n_frames):
+            nr = rng.randint(1, 8)
+            shards = [rng.randint(0, 2) for _ in range(nr)]
+            keys = [keyspace[rng.randrange(len(keyspace))]
+                    for _ in range(nr)]
+            vals = [rng.randint(-10**9, 10**9) for _ in range(nr)]
+            nulls = [1 if rng.random() < 0.2 else 0 for _ in range(nr)]
+            fb = FrameBuilder(nr)
+            fb.add_i64(ROLE_SHARD, "shard_id", "", shards)
+            fb.add_str(ROLE_TAG, "g", "", keys)
+            fb.add_i64(ROLE_FIELD, "value", "", vals, nulls=nulls)
+            frames.append(fb.finish())
+            rows.extend(zip(shards, keys, vals, nulls))
+        # model: first (shard,key) wins; null value rows still dedup but
+        # skip the Combine (IsNull check happens after markDedupSeen)
+        seen, acc, order = set(), {}, []
+        for sh, k, v, nu in rows:
+            if (sh, k) in seen:
+                continue
+            seen.add((sh, k))
+            if k not in acc:
+                acc[k] = [0, None, None]
+                order.append(k)
+            if not nu:
+                st = acc[k]
+                st[0] = (st[0] + v) % 2 ** 64
+                st[1] = v if st[1] is None else min(st[1], v)
+                st[2] = v if st[2] is None else max(st[2], v)
+        out = reduce_frames(frames,
+                            specs=[(2, ba.AGG_SUM), (2, ba.AGG_MIN),
+                                   (2, ba.AGG_MAX)],
+                            key_cols=[1], shard_col=0)
+        assert len(out) == len(order), seed
+        for (key, vals_), k in zip(out, order):
+            st = acc[k]
+            assert key.endswith(k), (seed, key, k)
+            assert vals_[0][0] % 2 ** 64 == st[0], seed
+            if st[1] is not None:
+                assert vals_[1][0] == st[1] and vals_[2][0] == st[2], seed
+            else:
+                assert vals_[1][0] == 2 ** 63 - 1, seed   # min sentinel

@@ -158,3 +158,15 @@ def test_bytes_block_zstd_roundtrip():
     p = o.bytes_block_encode(small)
     assert p[-len(b"abc") - 2] == 0 or True  # plain block marker present
     assert o.bytes_block_decode(p, 4) == small
+
+
+def test_tag_cell_helpers_match_oracle_cell_codec():
+    """i64_tag_cell must equal the oracle's restatement of
+    convert.Int64ToBytes (the stored int64 tag/field cell bytes,
+    number.go:33-46); f64_tag_cell is IEEE-754 BE (number.go:128-132)."""
+    import struct
+    from banyandb_amd import i64_tag_cell, f64_tag_cell
+    for v in [0, 1, -1, 63, -64, 2**40, -(2**40), 2**63 - 1, -(2**63)]:
+        assert i64_tag_cell(v) == o.cell_encode(v), v
+    for f in [0.0, -2.5, 1e300, -1e-300]:
+        assert f64_tag_cell(f) == struct.pack(">d", f)
