@@ -1,5 +1,5 @@
 """Wide randomized parity sweep (GPU): N seeds of the fuzz scenarios.
-Usage: python tools/fuzz_sweep.py [n_seeds]"""
+Usage: python tools/fuzz_sweep.py [n_seeds] [seed_base]"""
 import random
 import sys
 import os
@@ -16,9 +16,10 @@ MS = 10 ** 6
 
 def main():
     n = int(sys.argv[1]) if len(sys.argv) > 1 else 100
+    base = int(sys.argv[2], 0) if len(sys.argv) > 2 else 0xABC000
     fails = 0
     for seed in range(n):
-        rng = random.Random(0xABC000 + seed)
+        rng = random.Random(base + seed)
         try:
             if seed % 2 == 0:
                 b, is_float, tag_kind = build_scenario(rng)
