@@ -1286,11 +1286,16 @@ struct ScanFold {
     uint64_t nsel;      // selected row count
 };
 
+// full: compile-time promise that [r0, r1] covers every streamed row
+// (the clamp-free instantiations pass the template constant !EN_CLAMP),
+// letting the per-value selection branch dead-code out of the quad fold
+// — less register pressure for the 7-wave build.
 __device__ void scan_stream(const uint8_t *stream, int64_t n_deltas, bool dod,
                             int64_t first_plus /* v at row (dod?1:0) */,
                             int64_t d1_init, int64_t r0, int64_t r1,
                             int lane, ScanFold *f, DevErr *derr, uint64_t bi,
-                            PredWalk *pw0, PredWalk *pw1, PredWalk *pw2) {
+                            PredWalk *pw0, PredWalk *pw1, PredWalk *pw2,
+                            bool full = false) {
     uint64_t pos = 0;
     int64_t j = dod ? 2 : 1;
     int64_t jmax = n_deltas;            // always scan the whole stream
@@ -1381,7 +1386,7 @@ __device__ void scan_stream(const uint8_t *stream, int64_t n_deltas, bool dod,
                 int32_t sl = dA + dB + dC + dD;
                 int32_t S = wave_incl_scan32(sl, lane);
                 int32_t cum = S - sl;
-                if (j >= r0 && j + nt - 1 <= r1) {
+                if (full || (j >= r0 && j + nt - 1 <= r1)) {
                     // whole window selected (the dominant shape once the
                     // block-level clamp resolved): no per-value index
                     // bookkeeping — every terminator folds
@@ -1495,7 +1500,7 @@ __device__ void scan_stream(const uint8_t *stream, int64_t n_deltas, bool dod,
             }
             if (b < 0x80) {
                 int64_t sv = (int64_t)sv_u;
-                if (myj >= r0 && myj <= r1) {
+                if (full || (myj >= r0 && myj <= r1)) {
                     l_sum += (uint64_t)sv;
                     l_nsel++;
                     l_mn = sv < l_mn ? sv : l_mn;
@@ -1560,7 +1565,7 @@ __device__ void scan_stream(const uint8_t *stream, int64_t n_deltas, bool dod,
             val = v_carry + sscan;
             v_carry += readlane64(sscan, 63);
         }
-        bool in_sel = is_term && myj >= r0 && myj <= r1;
+        bool in_sel = is_term && (full || (myj >= r0 && myj <= r1));
         if (pw0) in_sel = pred_match_rows(pw0, myj, in_sel) && in_sel;
         if (pw1) in_sel = pred_match_rows(pw1, myj, in_sel) && in_sel;
         if (pw2) in_sel = pred_match_rows(pw2, myj, in_sel) && in_sel;
@@ -2705,7 +2710,8 @@ __global__ __launch_bounds__(256, (EN_VALUES && !EN_CLAMP && !EN_GROUPS) ? 7 : (
             }
             ScanFold ff;
             scan_stream(s, n - 1, dod, dod ? row1 : first, d1, r0, r1,
-                        lane, &ff, derr, (uint64_t)bi, wp0, wp1, wp2);
+                        lane, &ff, derr, (uint64_t)bi, wp0, wp1, wp2,
+                        !EN_CLAMP);
             lsum += ff.sum;
             lcnt += ff.nsel;
             lmn = ff.mn < lmn ? ff.mn : lmn;
@@ -2804,7 +2810,7 @@ __global__ __launch_bounds__(256, (EN_VALUES && !EN_CLAMP && !EN_GROUPS) ? 7 : (
                     scan_stream(fstream + e.byte_off, b - rel0, dod,
                                 e.v_start, e.d1_start, a - rel0, b - rel0,
                                 lane, &ff, derr, (uint64_t)bi, nullptr,
-                                nullptr, nullptr);
+                                nullptr, nullptr, !EN_CLAMP);
                     lsum = ff.sum;
                     lcnt = ff.nsel;
                     lmn = ff.mn;
@@ -2858,7 +2864,7 @@ __global__ __launch_bounds__(256, (EN_VALUES && !EN_CLAMP && !EN_GROUPS) ? 7 : (
                 ScanFold ff;
                 scan_stream(s, n - 1, dod, dod ? row1 : first, d1, r0, r1,
                             lane, &ff, derr, (uint64_t)bi, nullptr, nullptr,
-                            nullptr);
+                            nullptr, !EN_CLAMP);
                 uint64_t lsum = ff.sum;
                 int64_t lmn = ff.mn, lmx = ff.mx;
                 // rows outside the stream: row 0 (value=first) and, for
