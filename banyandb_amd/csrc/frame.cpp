@@ -465,7 +465,9 @@ extern "C" int bydb_reduce_frames(
         if (!r->err.empty()) { bydb_frame_close(r); return BYDB_ERR_BAD_DATA; }
         const uint64_t nrows = r->nrows;
         auto colok = [&](int32_t c) { return c >= 0 && (uint64_t)c < r->ncols; };
-        if ((shard_col >= 0 && !colok(shard_col)))
+        if (shard_col >= 0 &&
+            (!colok(shard_col) || (r->cols[shard_col].type != 1 &&
+                                   r->cols[shard_col].type != 2)))
             { bydb_frame_close(r); return BYDB_ERR_BAD_ARG; }
         for (int k = 0; k < n_key_cols; k++)
             if (!colok(key_cols[k])) { bydb_frame_close(r); return BYDB_ERR_BAD_ARG; }
