@@ -2246,13 +2246,14 @@ template <bool EN_VALUES, bool EN_PREDS, bool EN_GROUPS, bool EN_WALK,
 // 6 forces scratch spills into the window loop and measured ~1.6x SLOWER;
 // the closed-form-only instantiation fits 6 without spills.
 // Occupancy: the clamp-free value-scan instantiations (the analytic
-// full-range f64/min-max path) run at SEVEN waves/SIMD — measured best
-// in a 4/5/6/7/8 sweep (6.40 -> 5.67 ms on configs[2]) once the clamp
-// machinery was compiled out; the extra waves hide the decode's
+// full-range f64/min-max path) run at EIGHT waves/SIMD — swept 4..8 on
+// hardware twice: 7 won before the compile-time full-range flag
+// (6.40 -> 5.67 ms on configs[2]), 8 wins after it shed the selection
+// branch's registers (-> ~5.25 ms); the extra waves hide the decode's
 // dependency bubbles and the VGPR-spill cost stays below the win.  The
 // other instantiations keep 4 (more state: clamp / groups), the
-// closed-form-only ones 6.
-__global__ __launch_bounds__(256, (EN_VALUES && !EN_CLAMP && !EN_GROUPS) ? 7 : (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) void k_scan_agg_t(
+// closed-form-only ones 6 (8 measured WORSE there: 2.84 -> 3.25 ms).
+__global__ __launch_bounds__(256, (EN_VALUES && !EN_CLAMP && !EN_GROUPS) ? 8 : (EN_VALUES || EN_PREDS || EN_GROUPS) ? 4 : 6) void k_scan_agg_t(
     const uint8_t *__restrict__ payload, const uint8_t *__restrict__ sidecar,
     const bydb_block_desc *__restrict__ blocks,
     int64_t n_blocks, int64_t min_ts, int64_t max_ts, int flags,
