@@ -364,20 +364,27 @@ __device__ void dense_region(const uint8_t *stream, int64_t lo, int64_t hi,
 }
 
 // wave-wide inclusive scan of an int32 (per-window lane sums)
+// row-masked DPP: rows outside ROWMASK receive `old` = 0, so `v += ...`
+// is a no-op there — the canonical GCN cross-row scan combine.
+template <int CTRL, int ROWMASK>
+__device__ __forceinline__ uint32_t dpp_mov32_rm(uint32_t v) {
+    return (uint32_t)__builtin_amdgcn_update_dpp(0, (int)v, CTRL, ROWMASK,
+                                                 0xF, true);
+}
+
 __device__ __forceinline__ int32_t wave_incl_scan32(int32_t v, int lane) {
     v += (int32_t)dpp_mov32<0x111>((uint32_t)v);
     v += (int32_t)dpp_mov32<0x112>((uint32_t)v);
     v += (int32_t)dpp_mov32<0x114>((uint32_t)v);
     v += (int32_t)dpp_mov32<0x118>((uint32_t)v);
-    int32_t s15 = __builtin_amdgcn_readlane(v, 15);
-    int32_t s31 = __builtin_amdgcn_readlane(v, 31);
-    int32_t s47 = __builtin_amdgcn_readlane(v, 47);
-    // branchless cross-row fixup (see the 64-bit variant)
-    int r = lane >> 4;
-    int32_t m1 = -(int32_t)(r >= 1);
-    int32_t m2 = -(int32_t)(r >= 2);
-    int32_t m3 = -(int32_t)(r >= 3);
-    return v + (s15 & m1) + (s31 & m2) + (s47 & m3);
+    // cross-row combine via row broadcasts instead of three readlanes +
+    // masked adds: row_bcast15 adds lane 15 into row 1 and lane 47 into
+    // row 3; row_bcast31 then adds lane 31 (= rows 0+1 total) into rows
+    // 2 and 3 — two DPP ops, no SALU round trips
+    v += (int32_t)dpp_mov32_rm<0x142, 0xa>((uint32_t)v);  // ROW_BCAST15
+    v += (int32_t)dpp_mov32_rm<0x143, 0xc>((uint32_t)v);  // ROW_BCAST31
+    (void)lane;
+    return v;
 }
 
 // Dense min/max over an all-1-byte delta stream: fold min/max of the
